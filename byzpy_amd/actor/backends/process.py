@@ -1,0 +1,156 @@
+"""ProcessActorBackend — spawn-context child process with a duplex Pipe.
+
+Reference parity: engine/actor/backends/process.py (child _worker loop
+19-108; parent request serialization under a lock 303-321; tensors cross
+as shm handles via wrap/unwrap_payload).
+"""
+from __future__ import annotations
+
+import asyncio
+import itertools
+import multiprocessing as mp
+import queue as stdlib_queue
+import threading
+import uuid
+from typing import Any, Dict
+
+import cloudpickle
+
+from byzpy_amd.actor.channels import Endpoint
+from byzpy_amd.actor.ipc import unwrap_payload, wrap_payload
+from byzpy_amd.actor.router import channel_router
+
+_ids = itertools.count()
+
+
+def _worker(conn) -> None:
+    obj = None
+    mailboxes: Dict[str, stdlib_queue.Queue] = {}
+    while True:
+        try:
+            msg = conn.recv()
+        except (EOFError, OSError):
+            break
+        op = msg[0]
+        try:
+            if op == "construct":
+                factory, args, kwargs = cloudpickle.loads(msg[1])
+                obj = factory(*args, **kwargs)
+                conn.send(("ok", None))
+            elif op == "call":
+                method, args, kwargs = msg[1]
+                args = unwrap_payload(args)
+                kwargs = unwrap_payload(kwargs)
+                fn = getattr(obj, method)
+                result = fn(*args, **kwargs)
+                if asyncio.iscoroutine(result):
+                    result = asyncio.run(_await(result))
+                conn.send(("ok", wrap_payload(result)))
+            elif op == "chan_open":
+                mailboxes.setdefault(msg[1], stdlib_queue.Queue())
+                conn.send(("ok", None))
+            elif op == "chan_deliver":
+                name, payload = msg[1]
+                mailboxes.setdefault(name, stdlib_queue.Queue()).put(payload)
+                conn.send(("ok", None))
+            elif op == "chan_get":
+                name = msg[1]
+                payload = mailboxes.setdefault(name, stdlib_queue.Queue()).get()
+                conn.send(("ok", payload))
+            elif op == "close":
+                conn.send(("ok", None))
+                break
+            else:
+                conn.send(("err", f"unknown op {op!r}"))
+        except BaseException as e:  # noqa: BLE001
+            try:
+                conn.send(("err", repr(e)))
+            except Exception:
+                break
+    conn.close()
+
+
+async def _await(coro):
+    return await coro
+
+
+class ProcessActorBackend:
+    scheme = "process"
+
+    def __init__(self) -> None:
+        self.actor_id = f"process-{next(_ids)}-{uuid.uuid4().hex[:8]}"
+        self._proc: mp.Process | None = None
+        self._conn = None
+        # serializes Pipe send/recv pairs across event-loop executor threads
+        self._io_lock = threading.Lock()
+
+    async def start(self) -> None:
+        if self._proc is not None:
+            return
+        ctx = mp.get_context("spawn")
+        self._conn, child = ctx.Pipe(duplex=True)
+        self._proc = ctx.Process(target=_worker, args=(child,), daemon=True)
+        self._proc.start()
+        child.close()
+        channel_router.register(self.scheme, self.actor_id, self)
+
+    def _request_sync(self, msg: Any) -> Any:
+        with self._io_lock:
+            self._conn.send(msg)
+            status, payload = self._conn.recv()
+        if status != "ok":
+            raise RuntimeError(f"process actor error: {payload}")
+        return payload
+
+    async def _request(self, msg: Any) -> Any:
+        loop = asyncio.get_running_loop()
+        return await loop.run_in_executor(None, self._request_sync, msg)
+
+    async def construct(self, factory: Any, /, *args: Any, **kwargs: Any) -> None:
+        blob = cloudpickle.dumps((factory, args, kwargs))
+        await self._request(("construct", blob))
+
+    async def call(self, method: str, /, *args: Any, **kwargs: Any) -> Any:
+        payload = await self._request(
+            ("call", (method, wrap_payload(args), wrap_payload(kwargs)))
+        )
+        return unwrap_payload(payload)
+
+    async def close(self) -> None:
+        if self._proc is None:
+            return
+        try:
+            await self._request(("close", None))
+        except Exception:
+            pass
+        self._proc.join(timeout=5)
+        if self._proc.is_alive():
+            self._proc.terminate()
+            self._proc.join(timeout=5)
+        self._proc = None
+        channel_router.unregister(self.scheme, self.actor_id)
+
+    def get_endpoint(self) -> Endpoint:
+        return Endpoint(scheme=self.scheme, address="local", actor_id=self.actor_id)
+
+    # -- channels ----------------------------------------------------------
+    async def chan_open(self, name: str) -> None:
+        await self._request(("chan_open", name))
+
+    async def _deliver(self, name: str, payload: Any) -> None:
+        await self._request(("chan_deliver", (name, payload)))
+
+    async def chan_put(self, endpoint: Endpoint, name: str, payload: Any) -> None:
+        target = channel_router.lookup(endpoint.scheme, endpoint.actor_id)
+        if target is not None:
+            await target._deliver(name, wrap_payload(payload))
+            return
+        if endpoint.scheme == "tcp":
+            from byzpy_amd.actor.transports import tcp
+
+            await tcp.chan_put(endpoint, name, payload)
+            return
+        raise RuntimeError(f"no route to endpoint {endpoint!r}")
+
+    async def chan_get(self, name: str) -> Any:
+        return unwrap_payload(await self._request(("chan_get", name)))

@@ -1,0 +1,83 @@
+"""ParallelScheduler — independent graph nodes run concurrently.
+
+Reference parity: engine/graph/parallel_scheduler.py (in-degree counting
+95-108, ready-batch gather 164-196, shared back-pressure semaphore
+80-86/149-162 via metadata["subtask_semaphore"]).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Any, Dict, List, Optional
+
+from byzpy_amd.graph.graph import ComputationGraph, GraphInput, MessageSource
+from byzpy_amd.graph.scheduler import NodeScheduler
+from byzpy_amd.ops.base import OpContext
+
+
+class ParallelScheduler(NodeScheduler):
+    def __init__(
+        self,
+        graph: ComputationGraph,
+        pool: Any = None,
+        metadata: Optional[dict] = None,
+        *,
+        max_concurrent_nodes: Optional[int] = None,
+        max_pending_subtasks: Optional[int] = None,
+    ) -> None:
+        super().__init__(graph, pool, metadata)
+        self.max_concurrent_nodes = max_concurrent_nodes
+        self.max_pending_subtasks = max_pending_subtasks
+
+    async def run(self, inputs: Optional[Dict[str, Any]] = None) -> Any:
+        inputs = dict(inputs or {})
+        missing = [r for r in self.graph.required_inputs() if r not in inputs]
+        if missing:
+            raise KeyError(f"missing graph inputs: {missing}")
+
+        md = self._base_metadata()
+        pool_size = int(md.get("pool_size", 1))
+        pending_limit = self.max_pending_subtasks or pool_size * 8
+        md["subtask_semaphore"] = asyncio.Semaphore(max(1, pending_limit))
+
+        indeg: Dict[str, int] = {}
+        dependents: Dict[str, List[str]] = {n: [] for n in self.graph.nodes}
+        for name, node in self.graph.nodes.items():
+            deps = [s for s in node.inputs.values() if isinstance(s, str)]
+            indeg[name] = len(deps)
+            for dep in deps:
+                dependents[dep].append(name)
+
+        results: Dict[str, Any] = {}
+        ready = [n for n, k in indeg.items() if k == 0]
+        node_gate = (
+            asyncio.Semaphore(self.max_concurrent_nodes)
+            if self.max_concurrent_nodes
+            else None
+        )
+
+        async def run_node(name: str) -> None:
+            node = self.graph.nodes[name]
+            kwargs = {
+                arg: await self._resolve_input(spec, results, inputs)
+                for arg, spec in node.inputs.items()
+            }
+            ctx = OpContext(pool=self.pool, metadata=dict(md))
+            if node_gate is not None:
+                async with node_gate:
+                    results[name] = await node.op.run(ctx, **kwargs)
+            else:
+                results[name] = await node.op.run(ctx, **kwargs)
+
+        while ready:
+            batch = list(ready)
+            ready.clear()
+            await asyncio.gather(*(run_node(n) for n in batch))
+            for done in batch:
+                for dep in dependents[done]:
+                    indeg[dep] -= 1
+                    if indeg[dep] == 0:
+                        ready.append(dep)
+
+        if len(self.graph.outputs) == 1:
+            return results[self.graph.outputs[0]]
+        return {o: results[o] for o in self.graph.outputs}

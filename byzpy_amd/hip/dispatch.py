@@ -1,0 +1,270 @@
+"""Device dispatch for the robust-aggregation hot ops.
+
+CPU tensors -> byzpy_amd.ops.functional (pure torch, the parity oracle).
+CUDA(ROCm) tensors -> hand-written gfx950 kernels in byzpy_amd/_hip_ops
+(SURVEY.md §2.7 kernel inventory K1-K14). Ops whose GPU path composes
+library GEMMs / batched eig (SMEA combos, CAF matvecs, attacks) run the
+functional torch path on-device — rocBLAS library GEMMs are the sanctioned
+path for plain GEMM shapes.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Sequence
+
+import torch
+
+from byzpy_amd import hip as _hip
+from byzpy_amd.ops import functional as F
+
+_COLSEL_MEDIAN = 0
+_COLSEL_TRIMMED = 1
+_COLSEL_MEAMED = 2
+
+# Largest n the register-resident column-sort kernels support (padded to a
+# power of two; VGPR-bound). Larger n falls back to a documented torch path.
+COLSEL_MAX_N = 128
+
+
+def _gpu(X: torch.Tensor) -> bool:
+    return X.is_cuda
+
+
+# -- coordinate-wise (K1-K3) ------------------------------------------------
+
+
+def median(X: torch.Tensor) -> torch.Tensor:
+    if _gpu(X) and X.shape[0] <= COLSEL_MAX_N:
+        return _hip.require().colsel(X.contiguous(), _COLSEL_MEDIAN, 0)
+    return F.median(X)
+
+
+def trimmed_mean(X: torch.Tensor, f: int) -> torch.Tensor:
+    if _gpu(X) and X.shape[0] <= COLSEL_MAX_N:
+        return _hip.require().colsel(X.contiguous(), _COLSEL_TRIMMED, int(f))
+    return F.trimmed_mean(X, f)
+
+
+def mean_of_medians(X: torch.Tensor, f: int) -> torch.Tensor:
+    if _gpu(X) and X.shape[0] <= COLSEL_MAX_N:
+        return _hip.require().colsel(X.contiguous(), _COLSEL_MEAMED, int(f))
+    return F.mean_of_medians(X, f)
+
+
+# -- pairwise distances / Krum (K4, K5) -------------------------------------
+
+
+def gram(X: torch.Tensor) -> torch.Tensor:
+    """X @ X.T with f32 accumulation; MFMA split-K kernel on device."""
+    if _gpu(X):
+        return _hip.require().gram(X.contiguous())
+    Xf = X.float()
+    return Xf @ Xf.T
+
+
+def pairwise_sq_dists(X: torch.Tensor) -> torch.Tensor:
+    if _gpu(X):
+        G = gram(X)
+        norms = torch.diagonal(G)
+        D2 = norms[:, None] + norms[None, :] - 2.0 * G
+        return D2.clamp_(min=0.0)
+    return F.pairwise_sq_dists(X)
+
+
+def multi_krum_scores(X: torch.Tensor, f: int) -> torch.Tensor:
+    n = X.shape[0]
+    k = n - f - 1
+    if k < 1:
+        raise ValueError(f"need n - f - 1 >= 1, got n={n}, f={f}")
+    if _gpu(X):
+        D2 = pairwise_sq_dists(X)
+        D2 = D2 + torch.diag(
+            torch.full((n,), float("inf"), device=X.device, dtype=D2.dtype)
+        )
+        smallest = torch.topk(D2, k=k, dim=1, largest=False).values
+        return smallest.sum(dim=1)
+    return F.multi_krum_scores(X, f)
+
+
+def multi_krum(X: torch.Tensor, f: int, q: int) -> torch.Tensor:
+    if _gpu(X):
+        scores = multi_krum_scores(X, f)
+        winners = torch.topk(scores, k=q, largest=False).indices
+        return mean_rows(X, winners)
+    return F.multi_krum(X, f, q)
+
+
+def krum(X: torch.Tensor, f: int) -> torch.Tensor:
+    scores = multi_krum_scores(X, f)
+    return X[int(torch.argmin(scores))].clone()
+
+
+def mean_rows(X: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """Mean of the selected rows, f32 accumulation (K10 gather-mean)."""
+    if _gpu(X):
+        return _hip.require().mean_rows(X.contiguous(), idx.to(torch.int32))
+    return X.float()[idx].mean(dim=0).to(X.dtype)
+
+
+# -- row norms / scaling (K8, K12) ------------------------------------------
+
+
+def row_sqnorms(X: torch.Tensor) -> torch.Tensor:
+    if _gpu(X):
+        return _hip.require().row_sqnorms(X.contiguous())
+    Xf = X.float()
+    return (Xf * Xf).sum(dim=1)
+
+
+def row_scale(X: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    if _gpu(X):
+        return _hip.require().row_scale(X.contiguous(), scales.float())
+    return (X.float() * scales.float()[:, None]).to(X.dtype)
+
+
+def clip_rows(X: torch.Tensor, threshold: float) -> torch.Tensor:
+    if _gpu(X):
+        norms = row_sqnorms(X).sqrt_().clamp_min_(1e-20)
+        scale = torch.clamp(threshold / norms, max=1.0)
+        return row_scale(X, scale)
+    return F.clip_rows(X, threshold)
+
+
+def arc_clip(X: torch.Tensor, f: int) -> torch.Tensor:
+    n = X.shape[0]
+    k = int(2 * f / n * (n - f))
+    if k <= 0:
+        return X.clone()
+    if _gpu(X):
+        norms = row_sqnorms(X).sqrt_()
+        order = torch.argsort(norms, descending=True)
+        threshold = norms[order[k]]
+        scale = torch.clamp(threshold / norms.clamp_min(1e-20), max=1.0)
+        return row_scale(X, scale)
+    return F.arc_clip(X, f)
+
+
+def cge(X: torch.Tensor, f: int) -> torch.Tensor:
+    n = X.shape[0]
+    k = n - f
+    if _gpu(X):
+        norms = row_sqnorms(X)
+        idx = torch.argsort(norms, stable=True)[:k]
+        return mean_rows(X, idx)
+    return F.cge(X, f)
+
+
+# -- iterative fixed-point ops (K6, K7) -------------------------------------
+
+
+def geometric_median(
+    X: torch.Tensor,
+    *,
+    tol: float = 1e-6,
+    max_iter: int = 256,
+    eps: float = 1e-12,
+    init: str = "median",
+) -> torch.Tensor:
+    if not _gpu(X):
+        return F.geometric_median(X, tol=tol, max_iter=max_iter, eps=eps, init=init)
+    ext = _hip.require()
+    Xc = X.contiguous()
+    z = (median(Xc) if init == "median" else Xc.float().mean(dim=0)).float()
+    # fused per-iteration kernel pair; convergence polled every `poll` iters
+    # to avoid a host sync per iteration (SURVEY.md §7 hard part 2)
+    poll = 4
+    shift = torch.zeros((), device=X.device, dtype=torch.float32)
+    it = 0
+    while it < max_iter:
+        steps = min(poll, max_iter - it)
+        for _ in range(steps):
+            z = ext.weiszfeld_iter(Xc, z, float(eps), shift)
+        it += steps
+        if float(shift) <= tol:  # one sync per `poll` iterations
+            break
+    return z.to(X.dtype)
+
+
+def centered_clipping(
+    X: torch.Tensor,
+    *,
+    c_tau: float,
+    M: int = 10,
+    eps: float = 1e-12,
+    init: str = "mean",
+) -> torch.Tensor:
+    if not _gpu(X):
+        return F.centered_clipping(X, c_tau=c_tau, M=M, eps=eps, init=init)
+    ext = _hip.require()
+    Xc = X.contiguous()
+    if init == "mean":
+        v = Xc.float().mean(dim=0)
+    elif init == "median":
+        v = median(Xc).float()
+    else:
+        v = torch.zeros(X.shape[1], device=X.device, dtype=torch.float32)
+    for _ in range(M):
+        v = ext.cc_iter(Xc, v, float(c_tau), float(eps))
+    return v.to(X.dtype)
+
+
+# -- pre-aggregators --------------------------------------------------------
+
+
+def bucketing(
+    X: torch.Tensor, bucket_size: int, perm: Optional[Sequence[int]] = None
+) -> torch.Tensor:
+    if _gpu(X):
+        n = X.shape[0]
+        if perm is None:
+            perm_t = torch.randperm(n, device=X.device, dtype=torch.int32)
+        else:
+            perm_t = torch.as_tensor(list(perm), device=X.device, dtype=torch.int32)
+        return _hip.require().bucket_mean(X.contiguous(), perm_t, int(bucket_size))
+    return F.bucketing(X, bucket_size, perm)
+
+
+def nnm(X: torch.Tensor, f: int) -> torch.Tensor:
+    n = X.shape[0]
+    k = n - f
+    if _gpu(X):
+        D2 = pairwise_sq_dists(X)
+        idx = torch.topk(D2, k=k, dim=1, largest=False).indices.to(torch.int32)
+        return _hip.require().group_mean_rows(X.contiguous(), idx)
+    return F.nnm(X, f)
+
+
+# -- remaining ops: torch composition on either device ----------------------
+
+
+def minimum_diameter_averaging(X: torch.Tensor, f: int) -> torch.Tensor:
+    if _gpu(X):
+        D2 = pairwise_sq_dists(X)
+        subset = F.mda_subset(D2, f)
+        idx = torch.tensor(subset, device=X.device, dtype=torch.long)
+        return mean_rows(X, idx)
+    return F.minimum_diameter_averaging(X, f)
+
+
+def monna(X: torch.Tensor, f: int, reference_index: int = 0) -> torch.Tensor:
+    n = X.shape[0]
+    k = n - f
+    if _gpu(X):
+        Xf = X
+        ref = X[reference_index : reference_index + 1]
+        # ||x - r||^2 = ||x||^2 + ||r||^2 - 2 x.r ; reuse the norm kernel
+        norms = row_sqnorms(X)
+        dots = (X.float() @ X.float()[reference_index]).flatten()
+        d2 = norms + norms[reference_index] - 2.0 * dots
+        d2[reference_index] = -1.0
+        idx = torch.argsort(d2, stable=True)[:k]
+        return mean_rows(X, idx)
+    return F.monna(X, f, reference_index)
+
+
+def smea(X: torch.Tensor, f: int) -> torch.Tensor:
+    return F.smea(X, f)
+
+
+def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
+    return F.caf(X, f, power_iters=power_iters)

@@ -1,0 +1,210 @@
+"""Numerical unit tests with hand-computed fixtures (SURVEY.md §4 pattern 1)."""
+import math
+
+import pytest
+import torch
+
+from byzpy_amd.ops import functional as F
+
+
+def T(rows):
+    return torch.tensor(rows, dtype=torch.float32)
+
+
+class TestMedian:
+    def test_odd(self):
+        X = T([[1.0, 5.0], [3.0, -1.0], [2.0, 9.0]])
+        assert torch.allclose(F.median(X), T([2.0, 5.0])[0:2])
+
+    def test_even_true_median(self):
+        X = T([[1.0], [2.0], [3.0], [10.0]])
+        assert torch.allclose(F.median(X), torch.tensor([2.5]))
+
+    def test_robust_to_outlier(self):
+        X = T([[0.0], [0.1], [-0.1], [1e9], [0.05]])
+        assert abs(float(F.median(X))) < 0.2
+
+
+class TestTrimmedMean:
+    def test_hand(self):
+        X = T([[1.0], [2.0], [3.0], [100.0], [-100.0]])
+        # f=1 drops -100 and 100 -> mean(1,2,3) = 2
+        assert torch.allclose(F.trimmed_mean(X, 1), torch.tensor([2.0]))
+
+    def test_f0_is_mean(self):
+        X = torch.randn(6, 9)
+        assert torch.allclose(F.trimmed_mean(X, 0), X.mean(dim=0), atol=1e-6)
+
+    def test_bad_f(self):
+        with pytest.raises(ValueError):
+            F.trimmed_mean(torch.randn(4, 2), 2)
+
+
+class TestMeanOfMedians:
+    def test_hand(self):
+        X = T([[0.0], [1.0], [2.0], [50.0]])
+        # median = 1.5; n-f = 3 closest to 1.5: {0,1,2} -> mean 1
+        assert torch.allclose(F.mean_of_medians(X, 1), torch.tensor([1.0]))
+
+    def test_f0_is_mean(self):
+        X = torch.randn(5, 7)
+        assert torch.allclose(F.mean_of_medians(X, 0), X.mean(dim=0), atol=1e-6)
+
+
+class TestKrum:
+    def test_scores_hand(self):
+        # three clustered + one far point; n=4, f=1 -> k = n-f-1 = 2
+        X = T([[0.0, 0.0], [0.1, 0.0], [0.0, 0.1], [10.0, 10.0]])
+        scores = F.multi_krum_scores(X, 1)
+        assert scores.argmax() == 3
+        assert scores.argmin() in (0, 1, 2)
+
+    def test_krum_picks_cluster_member(self):
+        X = T([[0.0], [0.01], [0.02], [99.0]])
+        out = F.krum(X, 1)
+        assert float(out) < 1.0
+
+    def test_multi_krum_mean(self):
+        X = T([[1.0], [1.1], [0.9], [50.0]])
+        out = F.multi_krum(X, 1, 3)
+        assert abs(float(out) - 1.0) < 0.2
+
+
+class TestGeometricMedian:
+    def test_collinear(self):
+        X = T([[0.0], [1.0], [10.0]])
+        # geometric median of 1-D points = median
+        out = F.geometric_median(X, tol=1e-9, max_iter=500)
+        assert abs(float(out) - 1.0) < 1e-3
+
+    def test_symmetric(self):
+        X = T([[1.0, 0.0], [-1.0, 0.0], [0.0, 1.0], [0.0, -1.0]])
+        out = F.geometric_median(X, tol=1e-9)
+        assert torch.allclose(out, torch.zeros(2), atol=1e-4)
+
+
+class TestMDA:
+    def test_excludes_outlier(self):
+        X = T([[0.0], [0.1], [0.2], [9.0]])
+        out = F.minimum_diameter_averaging(X, 1)
+        assert torch.allclose(out, torch.tensor([0.1]), atol=1e-6)
+
+    def test_subset_exact(self):
+        X = T([[0.0], [1.0], [1.1], [1.2], [5.0]])
+        D2 = F.pairwise_sq_dists(X)
+        assert F.mda_subset(D2, 2) == (1, 2, 3)
+
+
+class TestMoNNA:
+    def test_reference_first(self):
+        X = T([[0.0], [0.1], [5.0], [0.2]])
+        out = F.monna(X, 1, reference_index=0)
+        assert torch.allclose(out, torch.tensor([0.1]), atol=1e-6)  # mean(0,.1,.2)
+
+
+class TestSMEA:
+    def test_low_variance_subset(self):
+        X = T([[0.0], [0.1], [0.05], [10.0]])
+        out = F.smea(X, 1)
+        assert abs(float(out) - 0.05) < 0.05
+
+
+class TestCenteredClipping:
+    def test_large_tau_is_mean(self):
+        X = torch.randn(8, 5)
+        out = F.centered_clipping(X, c_tau=1e9, M=3)
+        assert torch.allclose(out, X.mean(dim=0), atol=1e-5)
+
+    def test_resists_outlier(self):
+        X = T([[0.0], [0.1], [-0.1], [1000.0]])
+        out = F.centered_clipping(X, c_tau=0.5, M=10, init="median")
+        # v moves at most c_tau per iteration, so the outlier drags it < M*c_tau
+        assert abs(float(out)) < 5.0
+
+    def test_zero_init(self):
+        X = T([[1.0], [1.0], [1.0], [1.0]])
+        out = F.centered_clipping(X, c_tau=10.0, M=5, init="zero")
+        assert abs(float(out) - 1.0) < 1e-5
+
+
+class TestCGE:
+    def test_drops_largest_norms(self):
+        X = T([[1.0, 0.0], [0.0, 1.0], [100.0, 100.0]])
+        out = F.cge(X, 1)
+        assert torch.allclose(out, torch.tensor([0.5, 0.5]))
+
+
+class TestCAF:
+    def test_clean_data_near_mean(self):
+        g = torch.Generator().manual_seed(0)
+        X = torch.randn(16, 8, generator=g)
+        out = F.caf(X, 2)
+        assert (out - X.mean(dim=0)).norm() < 2.0
+
+
+class TestPreAggs:
+    def test_clip_rows(self):
+        X = T([[3.0, 4.0], [0.3, 0.4]])
+        out = F.clip_rows(X, 1.0)
+        assert torch.allclose(out[0], T([[0.6, 0.8]])[0], atol=1e-6)
+        assert torch.allclose(out[1], X[1], atol=1e-6)
+
+    def test_arc(self):
+        X = T([[10.0, 0.0], [1.0, 0.0], [0.5, 0.0], [0.1, 0.0]])
+        out = F.arc_clip(X, 1)
+        # k = floor(2*1/4*3) = 1 -> clip top-1 norm row to 2nd-largest (1.0)
+        assert torch.allclose(out[0], T([[1.0, 0.0]])[0], atol=1e-5)
+
+    def test_bucketing_deterministic(self):
+        X = T([[1.0], [2.0], [3.0], [4.0]])
+        out = F.bucketing(X, 2, perm=[0, 1, 2, 3])
+        assert torch.allclose(out, T([[1.5], [3.5]]))
+
+    def test_nnm(self):
+        X = T([[0.0], [0.1], [10.0]])
+        out = F.nnm(X, 1)
+        assert torch.allclose(out[0], torch.tensor([0.05]), atol=1e-6)
+        assert out.shape == X.shape
+
+
+class TestAttacks:
+    def test_empire(self):
+        X = T([[1.0, 2.0], [3.0, 4.0]])
+        assert torch.allclose(F.empire(X, -1.0), T([[-2.0, -3.0]])[0])
+
+    def test_sign_flip(self):
+        g = T([[1.0, -2.0]])[0]
+        assert torch.allclose(F.sign_flip(g, -1.0), T([[-1.0, 2.0]])[0])
+
+    def test_little_shape_and_z(self):
+        X = torch.randn(12, 6)
+        out = F.little(X, 2)
+        assert out.shape == (6,)
+
+    def test_gaussian_seeded(self):
+        X = torch.zeros(3, 8)
+        a = F.gaussian_attack(X, seed=7)
+        b = F.gaussian_attack(X, seed=7)
+        assert torch.equal(a, b)
+
+    def test_inf(self):
+        X = torch.zeros(3, 4)
+        assert torch.isinf(F.inf_attack(X)).all()
+
+    def test_mimic(self):
+        X = T([[1.0], [2.0], [3.0]])
+        assert float(F.mimic(X, 1)) == 2.0
+
+
+class TestLabelFlip:
+    def test_runs(self):
+        from byzpy_amd.attacks import LabelFlipAttack
+
+        torch.manual_seed(0)
+        model = torch.nn.Linear(4, 3)
+        x = torch.randn(5, 4)
+        y = torch.randint(0, 3, (5,))
+        atk = LabelFlipAttack(num_classes=3)
+        g = atk.apply(model=model, batch=(x, y))
+        d = sum(p.numel() for p in model.parameters())
+        assert g.shape == (d,)

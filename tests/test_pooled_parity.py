@@ -1,0 +1,119 @@
+"""Direct-vs-pooled parity (SURVEY.md §4 pattern 2): every aggregator's
+chunked pool path must match its direct path."""
+import asyncio
+
+import pytest
+import torch
+
+from byzpy_amd import run_operator
+from byzpy_amd.aggregators import (
+    CAF,
+    CenteredClipping,
+    ComparativeGradientElimination,
+    CoordinateWiseMedian,
+    CoordinateWiseTrimmedMean,
+    GeometricMedian,
+    Krum,
+    MeanOfMedians,
+    MinimumDiameterAveraging,
+    MoNNA,
+    MultiKrum,
+    SMEA,
+)
+from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
+from byzpy_amd.pre_aggregators import ARC, Bucketing, Clipping, NearestNeighborMixing
+
+
+def _run(coro):
+    return asyncio.run(coro)
+
+
+@pytest.fixture(scope="module")
+def data():
+    g = torch.Generator().manual_seed(42)
+    return torch.randn(12, 257, generator=g)
+
+
+AGGS = [
+    CoordinateWiseMedian(chunk_size=64),
+    CoordinateWiseTrimmedMean(2, chunk_size=64),
+    MeanOfMedians(2, chunk_size=64),
+    MultiKrum(2, 3, chunk_size=4),
+    Krum(2, chunk_size=4),
+    GeometricMedian(chunk_size=4),
+    MinimumDiameterAveraging(2, chunk_size=64),
+    MoNNA(2, chunk_size=4),
+    SMEA(8, chunk_size=64),
+    CenteredClipping(c_tau=0.5, chunk_size=4),
+    ComparativeGradientElimination(2, chunk_size=64),
+    CAF(2),
+]
+
+
+@pytest.mark.parametrize("agg", AGGS, ids=[a.name for a in AGGS])
+def test_thread_pool_matches_direct(agg, data):
+    grads = list(data)
+    direct = agg.aggregate(grads)
+
+    async def pooled():
+        return await run_operator(
+            agg,
+            {"gradients": grads},
+            pool_config=ActorPoolConfig(backend="thread", count=3),
+        )
+
+    out = _run(pooled())
+    assert torch.allclose(out, direct, atol=1e-4), f"{agg.name} pooled != direct"
+
+
+@pytest.mark.parametrize(
+    "agg",
+    [CoordinateWiseMedian(chunk_size=64), MultiKrum(2, 3, chunk_size=4)],
+    ids=["median", "multi-krum"],
+)
+def test_process_pool_matches_direct(agg, data):
+    grads = list(data)
+    direct = agg.aggregate(grads)
+
+    async def pooled():
+        return await run_operator(
+            agg,
+            {"gradients": grads},
+            pool_config=ActorPoolConfig(backend="process", count=2),
+        )
+
+    out = _run(pooled())
+    assert torch.allclose(out, direct, atol=1e-4)
+
+
+PRE_AGGS = [
+    Clipping(1.0, chunk_size=4),
+    Bucketing(3, perm=list(range(12))),
+    NearestNeighborMixing(2, feature_chunk_size=64),
+    ARC(2, chunk_size=4),
+]
+
+
+@pytest.mark.parametrize("pre", PRE_AGGS, ids=[p.name for p in PRE_AGGS])
+def test_preagg_thread_pool_matches_direct(pre, data):
+    vecs = list(data)
+    direct = pre.pre_aggregate(vecs)
+
+    async def pooled():
+        return await run_operator(
+            pre,
+            {"vectors": vecs},
+            pool_config=ActorPoolConfig(backend="thread", count=3),
+        )
+
+    out = _run(pooled())
+    assert len(out) == len(direct)
+    for a, b in zip(out, direct):
+        assert torch.allclose(a, b, atol=1e-4)
+
+
+def test_bf16_inputs_roundtrip(data):
+    grads = [g.bfloat16() for g in data]
+    out = CoordinateWiseMedian().aggregate(grads)
+    assert out.dtype == torch.bfloat16
+    assert out.shape == (257,)
