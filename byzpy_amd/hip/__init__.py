@@ -21,6 +21,8 @@ def extension() -> Optional[Any]:
     if not _tried:
         _tried = True
         try:
+            import torch  # noqa: F401  (libtorch must be loaded first)
+
             _ext = importlib.import_module("byzpy_amd._hip_ops")
         except Exception as e:  # noqa: BLE001
             _err = e

@@ -1,0 +1,238 @@
+// Column-wise order-statistic kernels: per-coordinate
+// median / trimmed mean / mean-of-medians over the n-axis of an (n, d)
+// matrix, one column per thread.
+//
+// Two variants:
+//  - register kernel (n <= 64): the column lives in a register array; the
+//    bitonic network is fully unrolled so every index is compile-time
+//    (guide §5.4 rule 20 — runtime-indexed register arrays spill).
+//    Order statistics at runtime positions are extracted with predicated
+//    unrolled scans for the same reason.
+//  - LDS kernel (64 < n <= 512): one wave per block, each lane owns a
+//    column staged in LDS with a +1 pad row stride (bank-conflict free,
+//    guide §6 G4); runtime bitonic loops.
+//
+// Loads are coalesced: adjacent lanes read adjacent columns, so each
+// row-iteration is one 256 B (f32) / 128 B (bf16) wave transaction.
+#include "common.h"
+
+namespace {
+
+constexpr float PAD = 3.0e38f;  // sorts after every real value
+
+enum Mode { MEDIAN = 0, TRIMMED = 1, MEAMED = 2 };
+
+// ---------------------------------------------------------------------------
+// register variant, n <= P, P in {8, 16, 32, 64}
+// ---------------------------------------------------------------------------
+
+template <int P>
+DEV void bitonic_sort_reg(float (&v)[P]) {
+#pragma unroll
+  for (int k = 2; k <= P; k <<= 1) {
+#pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = (i & k) == 0;
+          const float a = v[i], b = v[l];
+          if (asc ? (a > b) : (a < b)) { v[i] = b; v[l] = a; }
+        }
+      }
+    }
+  }
+}
+
+// key-value variant (sort k ascending, carry v along)
+template <int P>
+DEV void bitonic_sort_kv_reg(float (&key)[P], float (&val)[P]) {
+#pragma unroll
+  for (int k = 2; k <= P; k <<= 1) {
+#pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = (i & k) == 0;
+          const float a = key[i], b = key[l];
+          if (asc ? (a > b) : (a < b)) {
+            key[i] = b; key[l] = a;
+            const float t = val[i]; val[i] = val[l]; val[l] = t;
+          }
+        }
+      }
+    }
+  }
+}
+
+template <int P>
+DEV float extract_at(const float (&v)[P], int pos) {
+  float r = 0.0f;
+#pragma unroll
+  for (int i = 0; i < P; ++i)
+    if (i == pos) r = v[i];
+  return r;
+}
+
+template <int P, int MODE, typename T>
+__global__ void
+// MEAMED keeps two live P-float arrays (values + deviation keys); at P=64
+// that needs ~140 VGPRs, above the occupancy heuristic's 128 cap — ask for
+// 2 waves/SIMD so the allocator may use up to 256 VGPRs instead of spilling.
+__launch_bounds__(256, 2)
+colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
+                                  int n, long d, int f) {
+  const long col0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long col = col0; col < d; col += stride) {
+    float v[P];
+#pragma unroll
+    for (int i = 0; i < P; ++i)
+      v[i] = (i < n) ? to_f<T>(X[(long)i * d + col]) : PAD;
+    bitonic_sort_reg<P>(v);  // in place: meamed only needs sorted order
+
+    const float med_lo = extract_at<P>(v, (n - 1) >> 1);
+    const float med_hi = extract_at<P>(v, n >> 1);
+    const float med = 0.5f * (med_lo + med_hi);
+
+    float result;
+    if (MODE == MEDIAN) {
+      result = med;
+    } else if (MODE == TRIMMED) {
+      float s = 0.0f;
+#pragma unroll
+      for (int i = 0; i < P; ++i)
+        if (i >= f && i < n - f) s += v[i];
+      result = s / (float)(n - 2 * f);
+    } else {  // MEAMED: mean of the n-f values closest to the median
+      float dev[P];
+#pragma unroll
+      for (int i = 0; i < P; ++i)
+        dev[i] = (i < n) ? fabsf(v[i] - med) : PAD;
+      bitonic_sort_kv_reg<P>(dev, v);
+      float s = 0.0f;
+#pragma unroll
+      for (int i = 0; i < P; ++i)
+        if (i < n - f) s += v[i];
+      result = s / (float)(n - f);
+    }
+    out[col] = from_f<T>(result);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LDS variant, 64 < n <= 512; one wave per block, lane-owned columns
+// ---------------------------------------------------------------------------
+
+DEV void bitonic_sort_lds(float* v, int stride, int P) {
+  for (int k = 2; k <= P; k <<= 1)
+    for (int j = k >> 1; j > 0; j >>= 1)
+      for (int i = 0; i < P; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = (i & k) == 0;
+          const float a = v[i * stride], b = v[l * stride];
+          if (asc ? (a > b) : (a < b)) { v[i * stride] = b; v[l * stride] = a; }
+        }
+      }
+}
+
+template <int MODE, typename T>
+__global__ void colsel_lds_kernel(const T* __restrict__ X, T* __restrict__ out,
+                                  int n, long d, int f, int P) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: [P][WAVE] floats, lane l owns column buf[i*WAVE + l]. Uniform i
+  // across the wave puts each lane on bank (lane % 32): conflict-free for
+  // ds_read_b32 (lanes l and l+32 are different lane groups, guide §2).
+  float* buf = reinterpret_cast<float*>(smem);
+  const int lane = threadIdx.x;  // blockDim.x == 64
+  const int stride = WAVE;
+  const long col = (long)blockIdx.x * WAVE + lane;
+  if (col >= d) return;
+  float* mine = buf + lane;
+  for (int i = 0; i < P; ++i)
+    mine[i * stride] = (i < n) ? to_f<T>(X[(long)i * d + col]) : PAD;
+
+  bitonic_sort_lds(mine, stride, P);
+  const float med =
+      0.5f * (mine[((n - 1) >> 1) * stride] + mine[(n >> 1) * stride]);
+  if (MODE == MEDIAN) {
+    out[col] = from_f<T>(med);
+  } else if (MODE == TRIMMED) {
+    float s = 0.0f;
+    for (int i = f; i < n - f; ++i) s += mine[i * stride];
+    out[col] = from_f<T>(s / (float)(n - 2 * f));
+  } else {  // MEAMED: the n-f values closest to med form a contiguous
+    // window of the sorted column — shrink [l, r) from whichever end
+    // deviates more (two-pointer, O(f)).
+    int l = 0, r = n;
+    for (int k = 0; k < f; ++k) {
+      const float dl = med - mine[l * stride];
+      const float dr = mine[(r - 1) * stride] - med;
+      if (dl > dr) ++l; else --r;
+    }
+    float s = 0.0f;
+    for (int i = l; i < r; ++i) s += mine[i * stride];
+    out[col] = from_f<T>(s / (float)(n - f));
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// host-side launch (called from bind.cpp)
+// ---------------------------------------------------------------------------
+
+template <typename T>
+static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
+                                int f, hipStream_t stream) {
+  if (n <= 64) {
+    const int block = 256;
+    const long want = (d + block - 1) / block;
+    const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
+#define DISPATCH_REG(P)                                                        \
+  do {                                                                         \
+    if (mode == MEDIAN)                                                        \
+      hipLaunchKernelGGL((colsel_reg_kernel<P, MEDIAN, T>), dim3(grid),        \
+                         dim3(block), 0, stream, X, out, n, d, f);             \
+    else if (mode == TRIMMED)                                                  \
+      hipLaunchKernelGGL((colsel_reg_kernel<P, TRIMMED, T>), dim3(grid),       \
+                         dim3(block), 0, stream, X, out, n, d, f);             \
+    else                                                                       \
+      hipLaunchKernelGGL((colsel_reg_kernel<P, MEAMED, T>), dim3(grid),        \
+                         dim3(block), 0, stream, X, out, n, d, f);             \
+  } while (0)
+    if (n <= 8) DISPATCH_REG(8);
+    else if (n <= 16) DISPATCH_REG(16);
+    else if (n <= 32) DISPATCH_REG(32);
+    else DISPATCH_REG(64);
+#undef DISPATCH_REG
+  } else {
+    int P = 128;
+    while (P < n) P <<= 1;  // 128/256/512
+    const long grid = (d + WAVE - 1) / WAVE;
+    const size_t lds = (size_t)P * WAVE * sizeof(float);
+    if (mode == MEDIAN)
+      hipLaunchKernelGGL((colsel_lds_kernel<MEDIAN, T>), dim3(grid), dim3(WAVE),
+                         lds, stream, X, out, n, d, f, P);
+    else if (mode == TRIMMED)
+      hipLaunchKernelGGL((colsel_lds_kernel<TRIMMED, T>), dim3(grid), dim3(WAVE),
+                         lds, stream, X, out, n, d, f, P);
+    else
+      hipLaunchKernelGGL((colsel_lds_kernel<MEAMED, T>), dim3(grid), dim3(WAVE),
+                         lds, stream, X, out, n, d, f, P);
+  }
+}
+
+void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,
+                       int f, hipStream_t stream) {
+  launch_colsel_typed<float>(X, out, n, d, mode, f, stream);
+}
+
+void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
+                        long d, int mode, int f, hipStream_t stream) {
+  launch_colsel_typed<__hip_bfloat16>(X, out, n, d, mode, f, stream);
+}

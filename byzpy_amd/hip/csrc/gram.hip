@@ -1,0 +1,182 @@
+// MFMA split-K Gram kernel G = X @ X^T for gfx950.
+//
+// Shape regime: n small (8..512), d huge (up to 8e9) — the GEMM is skinny,
+// so the kernel is HBM-bound (read X once ~= n*d*2B at ~6.3 TB/s) and the
+// design goal is exactly-once HBM traffic + enough blocks to fill 256 CUs:
+//   - one workgroup = one 64x64 output tile x one K-slab (split-K across
+//     the grid; f32 atomicAdd combine into the tiny n x n output),
+//   - 16 waves per block (4x4 of 16x16 MFMA tiles); A-rows and B-rows of
+//     the same tile are shared through L2; a __syncthreads() every few
+//     K-steps keeps the waves' streams inside one L2 window so HBM traffic
+//     stays ~1x X,
+//   - bf16 path: v_mfma_f32_16x16x32_bf16 (both fragments are 8 contiguous
+//     k-elements of a row of X = one 16 B load);
+//     f32 path: v_mfma_f32_16x16x4_f32 (exact f32 at the vector rate,
+//     guide §3 — no xf32 on gfx950).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+namespace {
+
+constexpr int TILE = 64;    // output tile side
+constexpr int WAVES = 16;   // 4x4 waves of 16x16
+constexpr int SYNC_EVERY = 8;
+
+// Load an 8-element bf16 fragment from row `row` at column k0 (guarded).
+DEV bf16x8 load_frag_bf16(const __hip_bfloat16* __restrict__ X, int row, int n,
+                          long d, long k0, bool vec_ok) {
+  bf16x8 out;
+  if (row >= n) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = (__bf16)0.0f;
+    return out;
+  }
+  const __hip_bfloat16* p = X + (long)row * d + k0;
+  if (vec_ok) {
+    out = *reinterpret_cast<const bf16x8*>(p);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out[j] = *reinterpret_cast<const __bf16*>(p + j);
+  }
+  return out;
+}
+
+__global__ void gram_bf16_kernel(const __hip_bfloat16* __restrict__ X,
+                                 float* __restrict__ G, int n, long d,
+                                 long k_per_block, int vec_ok) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int row_base = blockIdx.y * TILE + wr * 16;
+  const int col_base = blockIdx.z * TILE + wc * 16;
+
+  const long k_lo = (long)blockIdx.x * k_per_block;
+  const long k_hi = min(d, k_lo + k_per_block);
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  const int frag_row = lane & 15;
+  const long frag_k = (long)(lane >> 4) * 8;
+
+  int step = 0;
+  for (long k0 = k_lo; k0 + 32 <= k_hi; k0 += 32) {
+    const bf16x8 a =
+        load_frag_bf16(X, row_base + frag_row, n, d, k0 + frag_k, vec_ok);
+    const bf16x8 b =
+        load_frag_bf16(X, col_base + frag_row, n, d, k0 + frag_k, vec_ok);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
+  }
+  // K tail (< 32): scalar FMA into the same accumulator positions would
+  // need the C layout; do it as a 1-wide MFMA with zero-padded fragments.
+  const long k_rem = k_hi - ((k_hi - k_lo) / 32) * 32 - k_lo;
+  if (k_rem > 0) {
+    const long k0 = k_hi - k_rem;
+    bf16x8 a, b;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const long k = k0 + frag_k + j;
+      const bool in = k < k_hi;
+      a[j] = (in && row_base + frag_row < n)
+                 ? *reinterpret_cast<const __bf16*>(
+                       X + (long)(row_base + frag_row) * d + k)
+                 : (__bf16)0.0f;
+      b[j] = (in && col_base + frag_row < n)
+                 ? *reinterpret_cast<const __bf16*>(
+                       X + (long)(col_base + frag_row) * d + k)
+                 : (__bf16)0.0f;
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+
+  // C/D map for 16x16: col = lane&15, row = (lane>>4)*4 + reg
+  const int out_col = col_base + (lane & 15);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int out_row = row_base + (lane >> 4) * 4 + r;
+    if (out_row < n && out_col < n)
+      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+  }
+}
+
+__global__ void gram_f32_kernel(const float* __restrict__ X,
+                                float* __restrict__ G, int n, long d,
+                                long k_per_block) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int row_base = blockIdx.y * TILE + wr * 16;
+  const int col_base = blockIdx.z * TILE + wc * 16;
+
+  const long k_lo = (long)blockIdx.x * k_per_block;
+  const long k_hi = min(d, k_lo + k_per_block);
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  const int a_row = row_base + (lane & 15);
+  const int b_row = col_base + (lane & 15);
+  const long lane_k = lane >> 4;  // k = k0 + lane_k, K-step 4
+
+  int step = 0;
+  long k0 = k_lo;
+  for (; k0 + 32 <= k_hi; k0 += 32) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const long k = k0 + u * 4 + lane_k;
+      const float a = (a_row < n) ? X[(long)a_row * d + k] : 0.0f;
+      const float b = (b_row < n) ? X[(long)b_row * d + k] : 0.0f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+    if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
+  }
+  for (; k0 < k_hi; k0 += 4) {
+    const long k = k0 + lane_k;
+    const bool in = k < k_hi;
+    const float a = (in && a_row < n) ? X[(long)a_row * d + k] : 0.0f;
+    const float b = (in && b_row < n) ? X[(long)b_row * d + k] : 0.0f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+
+  const int out_col = col_base + (lane & 15);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int out_row = row_base + (lane >> 4) * 4 + r;
+    if (out_row < n && out_col < n)
+      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+  }
+}
+
+inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
+  const int tiles = (n + TILE - 1) / TILE;
+  const long tile_blocks = (long)tiles * tiles;
+  long want = 1024 / tile_blocks;
+  if (want < 1) want = 1;
+  const long max_by_d = (d + 4095) / 4096;  // keep slabs >= 4096 elements
+  if (want > max_by_d) want = max_by_d;
+  splitk = (int)want;
+  k_per_block = (d + splitk - 1) / splitk;
+}
+
+}  // namespace
+
+void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
+                      hipStream_t stream) {
+  int splitk; long kpb;
+  split_geometry(n, d, splitk, kpb);
+  const int tiles = (n + TILE - 1) / TILE;
+  dim3 grid(splitk, tiles, tiles);
+  const int vec_ok = ((d % 8) == 0) ? 1 : 0;
+  hipLaunchKernelGGL(gram_bf16_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
+                     n, d, kpb, vec_ok);
+}
+
+void launch_gram_f32(const float* X, float* G, int n, long d,
+                     hipStream_t stream) {
+  int splitk; long kpb;
+  split_geometry(n, d, splitk, kpb);
+  const int tiles = (n + TILE - 1) / TILE;
+  dim3 grid(splitk, tiles, tiles);
+  hipLaunchKernelGGL(gram_f32_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
+                     n, d, kpb);
+}

@@ -21,9 +21,9 @@ _COLSEL_MEDIAN = 0
 _COLSEL_TRIMMED = 1
 _COLSEL_MEAMED = 2
 
-# Largest n the register-resident column-sort kernels support (padded to a
-# power of two; VGPR-bound). Larger n falls back to a documented torch path.
-COLSEL_MAX_N = 128
+# Largest n the column-sort kernels support: registers to n=64, LDS-staged
+# to n=512. Larger n falls back to a documented torch path.
+COLSEL_MAX_N = 512
 
 
 def _gpu(X: torch.Tensor) -> bool:

@@ -1,0 +1,100 @@
+"""RCCL-over-xGMI distributed fabric (torch.distributed).
+
+MI355X design: one process per GPU; backend "nccl" IS
+RCCL on ROCm; gloo on CPU boxes so the same code paths run in CI. All
+gradient movement is collectives — broadcast / all-reduce / all-gather /
+reduce-scatter / all-to-all — sized for 7 p2p xGMI links per GPU.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def is_initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if is_initialized() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if is_initialized() else 1
+
+
+def init_from_env(timeout_s: float = 600.0) -> int:
+    """Initialize from torchrun-style env vars; returns local rank. Uses
+    RCCL when a GPU is visible, gloo otherwise. Safe to call twice."""
+    if is_initialized():
+        return int(os.environ.get("LOCAL_RANK", 0))
+    if "RANK" not in os.environ:
+        return 0  # single-process mode: no process group
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", 0)))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+    dist.init_process_group(
+        backend=backend, timeout=datetime.timedelta(seconds=timeout_s)
+    )
+    return local_rank
+
+
+def barrier() -> None:
+    if is_initialized():
+        dist.barrier()
+
+
+def all_reduce_(t: torch.Tensor) -> torch.Tensor:
+    if is_initialized():
+        dist.all_reduce(t)
+    return t
+
+
+def broadcast_(t: torch.Tensor, src: int = 0) -> torch.Tensor:
+    if is_initialized():
+        dist.broadcast(t, src)
+    return t
+
+
+def all_gather_rows(local: torch.Tensor) -> torch.Tensor:
+    """All-gather equal-sized row blocks into one stacked matrix (PS gather
+    C1: workers -> all)."""
+    if not is_initialized():
+        return local
+    world = get_world_size()
+    out = torch.empty(
+        (local.shape[0] * world,) + tuple(local.shape[1:]),
+        dtype=local.dtype,
+        device=local.device,
+    )
+    dist.all_gather_into_tensor(out, local.contiguous())
+    return out
+
+
+def all_to_all_rows(local: torch.Tensor) -> torch.Tensor:
+    """Row-blocked all-to-all: rank r sends row-block j to rank j and
+    receives block r from everyone (n-sharded -> d-sharded transposition,
+    SURVEY.md C3/C4)."""
+    if not is_initialized():
+        return local
+    world = get_world_size()
+    assert local.shape[0] % world == 0
+    out = torch.empty_like(local)
+    dist.all_to_all_single(out, local.contiguous())
+    return out
+
+
+def column_shard(X: torch.Tensor, rank: Optional[int] = None) -> torch.Tensor:
+    """This rank's contiguous d-shard of an (n, d) matrix."""
+    world = get_world_size()
+    if world == 1:
+        return X
+    r = get_rank() if rank is None else rank
+    d = X.shape[1]
+    per = (d + world - 1) // world
+    return X[:, r * per : min(d, (r + 1) * per)].contiguous()
