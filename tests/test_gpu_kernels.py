@@ -54,7 +54,22 @@ def test_meamed(n, d, dtype):
     X = _rand(n, d, dtype)
     out = D.mean_of_medians(X, 2)
     ref = F.mean_of_medians(X.float().cpu(), 2)
-    assert torch.allclose(out.float().cpu(), ref, **_tol(dtype))
+    if dtype == torch.float32:
+        assert torch.allclose(out.float().cpu(), ref, **_tol(dtype))
+    else:
+        # bf16 quantization produces exact deviation TIES at the window
+        # boundary; GPU (sorted two-pointer) and CPU (topk index order) may
+        # keep different-but-equally-close elements. Valid results differ by
+        # at most 2*tau/(n-f) per boundary tie, tau = the largest kept
+        # deviation — check against that bound instead of allclose.
+        Xf = X.float().cpu()
+        f = 2
+        med = F.median(Xf)
+        dev = (Xf - med[None, :]).abs()
+        tau = torch.topk(dev, k=n - f, dim=0, largest=False).values[-1]
+        bound = 2.0 * tau / (n - f) + 0.02
+        diff = (out.float().cpu() - ref).abs()
+        assert (diff <= bound).all(), f"max excess {(diff - bound).max()}"
 
 
 @pytest.mark.parametrize("n", [72, 100, 128, 200, 256, 512])
