@@ -37,6 +37,8 @@ namespace {
 
 hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
 
+constexpr int kMaxRowsLds = 1024;  // per-row weight staging in rowops LDS
+
 void check_matrix(const torch::Tensor& X) {
   TORCH_CHECK(X.is_cuda(), "expected a device tensor");
   TORCH_CHECK(X.dim() == 2, "expected (n, d)");
@@ -193,6 +195,7 @@ torch::Tensor weiszfeld_iter(torch::Tensor X, torch::Tensor z, double eps,
               z.is_contiguous() && z.numel() == X.size(1));
   TORCH_CHECK(shift.is_cuda() && shift.scalar_type() == torch::kFloat32 &&
               shift.numel() == 1);
+  TORCH_CHECK(X.size(0) <= kMaxRowsLds, "weiszfeld supports n <= 1024");
   const int n = (int)X.size(0);
   const long d = (long)X.size(1);
   auto dist2 = torch::zeros({n}, X.options().dtype(torch::kFloat32));
@@ -268,6 +271,7 @@ torch::Tensor cc_apply(torch::Tensor X, torch::Tensor v, torch::Tensor dist2,
 torch::Tensor cc_iter(torch::Tensor X, torch::Tensor v, double c_tau,
                       double eps) {
   check_matrix(X);
+  TORCH_CHECK(X.size(0) <= kMaxRowsLds, "cc supports n <= 1024");
   TORCH_CHECK(v.is_cuda() && v.scalar_type() == torch::kFloat32 &&
               v.is_contiguous() && v.numel() == X.size(1));
   const int n = (int)X.size(0);

@@ -38,7 +38,9 @@ DEV void bitonic_sort_reg(float (&v)[P]) {
         if (l > i) {
           const bool asc = (i & k) == 0;
           const float a = v[i], b = v[l];
-          if (asc ? (a > b) : (a < b)) { v[i] = b; v[l] = a; }
+          const float lo = fminf(a, b), hi = fmaxf(a, b);
+          v[i] = asc ? lo : hi;
+          v[l] = asc ? hi : lo;
         }
       }
     }
@@ -77,12 +79,14 @@ DEV float extract_at(const float (&v)[P], int pos) {
   return r;
 }
 
+// MEAMED keeps two live P-float arrays (values + deviation keys); at P=64
+// that needs ~140 VGPRs, above the occupancy heuristic's 128 cap — the
+// MEAMED instantiations get __launch_bounds__(256, 2) via this trait so the
+// allocator may use 256 VGPRs instead of spilling, while MEDIAN/TRIMMED
+// keep the default 4-waves/SIMD occupancy.
 template <int P, int MODE, typename T>
 __global__ void
-// MEAMED keeps two live P-float arrays (values + deviation keys); at P=64
-// that needs ~140 VGPRs, above the occupancy heuristic's 128 cap — ask for
-// 2 waves/SIMD so the allocator may use up to 256 VGPRs instead of spilling.
-__launch_bounds__(256, 2)
+__launch_bounds__(256, (MODE == 2 && P >= 64) ? 2 : 4)
 colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
                                   int n, long d, int f) {
   const long col0 = (long)blockIdx.x * blockDim.x + threadIdx.x;

@@ -60,14 +60,30 @@ __global__ void gram_bf16_kernel(const __hip_bfloat16* __restrict__ X,
   const int frag_row = lane & 15;
   const long frag_k = (long)(lane >> 4) * 8;
 
+  // K-unroll x4: issue all 8 fragment loads before the MFMA cluster so the
+  // HBM latency of one iteration hides under the previous one's MFMAs.
   int step = 0;
-  for (long k0 = k_lo; k0 + 32 <= k_hi; k0 += 32) {
+  long k0 = k_lo;
+  for (; k0 + 128 <= k_hi; k0 += 128) {
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      a[u] = load_frag_bf16(X, row_base + frag_row, n, d, k0 + u * 32 + frag_k,
+                            vec_ok);
+      b[u] = load_frag_bf16(X, col_base + frag_row, n, d, k0 + u * 32 + frag_k,
+                            vec_ok);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc, 0, 0, 0);
+    if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
+  }
+  for (; k0 + 32 <= k_hi; k0 += 32) {
     const bf16x8 a =
         load_frag_bf16(X, row_base + frag_row, n, d, k0 + frag_k, vec_ok);
     const bf16x8 b =
         load_frag_bf16(X, col_base + frag_row, n, d, k0 + frag_k, vec_ok);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-    if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
   }
   // K tail (< 32): scalar FMA into the same accumulator positions would
   // need the C layout; do it as a 1-wide MFMA with zero-padded fragments.

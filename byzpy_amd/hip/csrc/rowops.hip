@@ -8,31 +8,98 @@
 //   weiszfeld_update   fused w=1/max(dist,eps); z' = sum(w x)/sum(w); ||dz||^2
 //   cc_update          fused alpha=min(1,c/dist); v' = v + sum(alpha (x-v))/n
 //
-// Row reductions: 2D grid (k-slab, row), block-reduce + one f32 atomic per
-// block — the (n,) outputs are tiny, contention is nil (guide G12).
-// Column kernels: one thread per coordinate, row loop inside; adjacent
-// lanes read adjacent coordinates so every row iteration is one coalesced
-// wave transaction.
+// All kernels are HBM-bound; loads are vectorized to 16 B/lane (f32x4 /
+// bf16x8 — guide G13: hipcc does not auto-vectorize bf16) with a scalar
+// fallback when d is not a multiple of the vector width. Row reductions:
+// 2D grid (k-slab, row) + block reduce + one f32 atomic per block (G12).
+// Column kernels: V consecutive coordinates per thread, row loop inside;
+// per-row weights (1/dist etc.) are staged once in LDS per block.
 #include "common.h"
 
 namespace {
 
+constexpr int MAX_N_LDS = 1024;  // per-row weight stage (4 KB)
+
+// -- vector load/store ------------------------------------------------------
+
+template <typename T>
+struct VecTraits;
+
+template <>
+struct VecTraits<float> {
+  static constexpr int V = 4;
+  static DEV void load(const float* p, float (&o)[4]) {
+    const float4 v = *reinterpret_cast<const float4*>(p);
+    o[0] = v.x; o[1] = v.y; o[2] = v.z; o[3] = v.w;
+  }
+  static DEV void store(float* p, const float (&o)[4]) {
+    *reinterpret_cast<float4*>(p) = make_float4(o[0], o[1], o[2], o[3]);
+  }
+};
+
+template <>
+struct VecTraits<__hip_bfloat16> {
+  static constexpr int V = 8;
+  static DEV float unpack(unsigned u) {
+    union { unsigned short s; __hip_bfloat16 h; } c;
+    c.s = (unsigned short)u;
+    return __bfloat162float(c.h);
+  }
+  static DEV unsigned short pack(float x) {
+    union { unsigned short s; __hip_bfloat16 h; } c;
+    c.h = __float2bfloat16(x);
+    return c.s;
+  }
+  static DEV void load(const __hip_bfloat16* p, float (&o)[8]) {
+    const uint4 v = *reinterpret_cast<const uint4*>(p);
+    o[0] = unpack(v.x); o[1] = unpack(v.x >> 16);
+    o[2] = unpack(v.y); o[3] = unpack(v.y >> 16);
+    o[4] = unpack(v.z); o[5] = unpack(v.z >> 16);
+    o[6] = unpack(v.w); o[7] = unpack(v.w >> 16);
+  }
+  static DEV void store(__hip_bfloat16* p, const float (&o)[8]) {
+    uint4 v;
+    v.x = pack(o[0]) | ((unsigned)pack(o[1]) << 16);
+    v.y = pack(o[2]) | ((unsigned)pack(o[3]) << 16);
+    v.z = pack(o[4]) | ((unsigned)pack(o[5]) << 16);
+    v.w = pack(o[6]) | ((unsigned)pack(o[7]) << 16);
+    *reinterpret_cast<uint4*>(p) = v;
+  }
+};
+
 // -- row reductions ---------------------------------------------------------
 
-template <typename T, bool CENTER>
+template <typename T, bool CENTER, bool VEC>
 __global__ void row_red_kernel(const T* __restrict__ X,
                                const float* __restrict__ z,
                                float* __restrict__ out, long d) {
   __shared__ float lds[16];
+  constexpr int V = VecTraits<T>::V;
   const int row = blockIdx.y;
   const T* xr = X + (long)row * d;
   float acc = 0.0f;
-  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long j = start; j < d; j += stride) {
-    float v = to_f<T>(xr[j]);
-    if (CENTER) v -= z[j];
-    acc += v * v;
+  if (VEC) {
+    const long dv = d / V;
+    const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long jv = start; jv < dv; jv += stride) {
+      float x[V];
+      VecTraits<T>::load(xr + jv * V, x);
+#pragma unroll
+      for (int c = 0; c < V; ++c) {
+        float v = x[c];
+        if (CENTER) v -= z[jv * V + c];
+        acc += v * v;
+      }
+    }
+  } else {
+    const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long j = start; j < d; j += stride) {
+      float v = to_f<T>(xr[j]);
+      if (CENTER) v -= z[j];
+      acc += v * v;
+    }
   }
   acc = block_reduce_sum(acc, lds);
   if (threadIdx.x == 0) atomicAdd(&out[row], acc);
@@ -40,124 +107,210 @@ __global__ void row_red_kernel(const T* __restrict__ X,
 
 // -- row scaling ------------------------------------------------------------
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void row_scale_kernel(const T* __restrict__ X,
                                  const float* __restrict__ s,
                                  T* __restrict__ out, long d) {
+  constexpr int V = VecTraits<T>::V;
   const int row = blockIdx.y;
   const float sc = s[row];
   const T* xr = X + (long)row * d;
   T* yr = out + (long)row * d;
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long j = start; j < d; j += stride)
-    yr[j] = from_f<T>(to_f<T>(xr[j]) * sc);
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float x[V];
+      VecTraits<T>::load(xr + jv * V, x);
+#pragma unroll
+      for (int c = 0; c < V; ++c) x[c] *= sc;
+      VecTraits<T>::store(yr + jv * V, x);
+    }
+  } else {
+    for (long j = start; j < d; j += stride)
+      yr[j] = from_f<T>(to_f<T>(xr[j]) * sc);
+  }
 }
 
 // -- gather means -----------------------------------------------------------
+// GROUPED=false: idx is (k,) and blockIdx.y==0 -> out (d,).
+// GROUPED=true:  idx is (g,k) rows per group g=blockIdx.y -> out (g,d).
 
-template <typename T>
-__global__ void mean_rows_kernel(const T* __restrict__ X,
-                                 const int* __restrict__ idx, int k,
-                                 T* __restrict__ out, long d) {
-  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  const float inv = 1.0f / (float)k;
-  for (long j = start; j < d; j += stride) {
-    float acc = 0.0f;
-    for (int i = 0; i < k; ++i) acc += to_f<T>(X[(long)idx[i] * d + j]);
-    out[j] = from_f<T>(acc * inv);
-  }
-}
-
-template <typename T>
-__global__ void group_mean_rows_kernel(const T* __restrict__ X,
-                                       const int* __restrict__ idx, int k,
-                                       T* __restrict__ out, long d) {
-  const int g = blockIdx.y;
+template <typename T, bool VEC, bool GROUPED>
+__global__ void gather_mean_kernel(const T* __restrict__ X,
+                                   const int* __restrict__ idx, int k,
+                                   T* __restrict__ out, long d) {
+  constexpr int V = VecTraits<T>::V;
+  const int g = GROUPED ? blockIdx.y : 0;
   const int* gi = idx + (long)g * k;
   T* og = out + (long)g * d;
+  const float inv = 1.0f / (float)k;
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  const float inv = 1.0f / (float)k;
-  for (long j = start; j < d; j += stride) {
-    float acc = 0.0f;
-    for (int i = 0; i < k; ++i) acc += to_f<T>(X[(long)gi[i] * d + j]);
-    og[j] = from_f<T>(acc * inv);
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float acc[V] = {0};
+      for (int i = 0; i < k; ++i) {
+        float x[V];
+        VecTraits<T>::load(X + (long)gi[i] * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) acc[c] *= inv;
+      VecTraits<T>::store(og + jv * V, acc);
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      float acc = 0.0f;
+      for (int i = 0; i < k; ++i) acc += to_f<T>(X[(long)gi[i] * d + j]);
+      og[j] = from_f<T>(acc * inv);
+    }
   }
 }
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void bucket_mean_kernel(const T* __restrict__ X,
                                    const int* __restrict__ perm, int n,
                                    int bucket, T* __restrict__ out, long d) {
+  constexpr int V = VecTraits<T>::V;
   const int b = blockIdx.y;
   const int lo = b * bucket;
   const int hi = min(n, lo + bucket);
   T* ob = out + (long)b * d;
+  const float inv = 1.0f / (float)(hi - lo);
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  const float inv = 1.0f / (float)(hi - lo);
-  for (long j = start; j < d; j += stride) {
-    float acc = 0.0f;
-    for (int i = lo; i < hi; ++i) acc += to_f<T>(X[(long)perm[i] * d + j]);
-    ob[j] = from_f<T>(acc * inv);
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float acc[V] = {0};
+      for (int i = lo; i < hi; ++i) {
+        float x[V];
+        VecTraits<T>::load(X + (long)perm[i] * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) acc[c] *= inv;
+      VecTraits<T>::store(ob + jv * V, acc);
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      float acc = 0.0f;
+      for (int i = lo; i < hi; ++i) acc += to_f<T>(X[(long)perm[i] * d + j]);
+      ob[j] = from_f<T>(acc * inv);
+    }
   }
 }
 
 // -- fused fixed-point iterations ------------------------------------------
+// Per-row weights are a function of the global distances only; stage them
+// in LDS once per block instead of recomputing per column.
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void weiszfeld_update_kernel(const T* __restrict__ X,
                                         const float* __restrict__ z,
                                         const float* __restrict__ dist2,
                                         float* __restrict__ z_new,
                                         float* __restrict__ shift2, int n,
                                         long d, float eps) {
-  __shared__ float lds[16];
+  __shared__ float w_lds[MAX_N_LDS];
+  __shared__ float red[16];
+  constexpr int V = VecTraits<T>::V;
+  for (int i = threadIdx.x; i < n; i += blockDim.x)
+    w_lds[i] = 1.0f / fmaxf(sqrtf(dist2[i]), eps);
+  __syncthreads();
+  float den = 0.0f;
+  for (int i = 0; i < n; ++i) den += w_lds[i];
+  const float inv_den = 1.0f / den;
+
+  float local_shift = 0.0f;
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  float local_shift = 0.0f;
-  for (long j = start; j < d; j += stride) {
-    float num = 0.0f, den = 0.0f;
-    for (int i = 0; i < n; ++i) {
-      const float dist = fmaxf(sqrtf(dist2[i]), eps);
-      const float w = 1.0f / dist;
-      num += w * to_f<T>(X[(long)i * d + j]);
-      den += w;
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float num[V] = {0};
+      for (int i = 0; i < n; ++i) {
+        const float w = w_lds[i];
+        float x[V];
+        VecTraits<T>::load(X + (long)i * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) num[c] += w * x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) {
+        const float zi = num[c] * inv_den;
+        const float dz = zi - z[jv * V + c];
+        local_shift += dz * dz;
+        z_new[jv * V + c] = zi;
+      }
     }
-    const float zi = num / den;
-    const float dz = zi - z[j];
-    local_shift += dz * dz;
-    z_new[j] = zi;
+  } else {
+    for (long j = start; j < d; j += stride) {
+      float num = 0.0f;
+      for (int i = 0; i < n; ++i)
+        num += w_lds[i] * to_f<T>(X[(long)i * d + j]);
+      const float zi = num * inv_den;
+      const float dz = zi - z[j];
+      local_shift += dz * dz;
+      z_new[j] = zi;
+    }
   }
-  const float s = block_reduce_sum(local_shift, lds);
+  const float s = block_reduce_sum(local_shift, red);
   if (threadIdx.x == 0) atomicAdd(shift2, s);
 }
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void cc_update_kernel(const T* __restrict__ X,
                                  const float* __restrict__ v,
                                  const float* __restrict__ dist2,
                                  float* __restrict__ v_new, int n, long d,
                                  float c_tau, float eps) {
+  __shared__ float a_lds[MAX_N_LDS];
+  constexpr int V = VecTraits<T>::V;
+  for (int i = threadIdx.x; i < n; i += blockDim.x)
+    a_lds[i] = fminf(1.0f, c_tau / fmaxf(sqrtf(dist2[i]), eps));
+  __syncthreads();
+  float alpha_sum = 0.0f;
+  for (int i = 0; i < n; ++i) alpha_sum += a_lds[i];
+  const float inv_n = 1.0f / (float)n;
+
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  const float inv_n = 1.0f / (float)n;
-  for (long j = start; j < d; j += stride) {
-    const float vj = v[j];
-    float acc = 0.0f;
-    for (int i = 0; i < n; ++i) {
-      const float dist = fmaxf(sqrtf(dist2[i]), eps);
-      const float alpha = fminf(1.0f, c_tau / dist);
-      acc += alpha * (to_f<T>(X[(long)i * d + j]) - vj);
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float acc[V] = {0};
+      for (int i = 0; i < n; ++i) {
+        const float a = a_lds[i];
+        float x[V];
+        VecTraits<T>::load(X + (long)i * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += a * x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) {
+        const float vj = v[jv * V + c];
+        v_new[jv * V + c] = vj + (acc[c] - alpha_sum * vj) * inv_n;
+      }
     }
-    v_new[j] = vj + acc * inv_n;
+  } else {
+    for (long j = start; j < d; j += stride) {
+      const float vj = v[j];
+      float acc = 0.0f;
+      for (int i = 0; i < n; ++i)
+        acc += a_lds[i] * (to_f<T>(X[(long)i * d + j]) - vj);
+      v_new[j] = vj + acc * inv_n;
+    }
   }
 }
 
-inline int col_grid(long d, int block) {
-  const long want = (d + block - 1) / block;
+inline int col_grid(long work, int block) {
+  const long want = (work + block - 1) / block;
   return (int)(want < 4096 ? (want > 0 ? want : 1) : 4096);
 }
 
@@ -167,6 +320,11 @@ inline int slab_grid(long d, int n, int block) {
   long cap = 2048 / (n > 0 ? n : 1);
   if (cap < 1) cap = 1;
   return (int)(per_row < cap ? (per_row > 0 ? per_row : 1) : cap);
+}
+
+template <typename T>
+inline bool vec_ok(long d) {
+  return (d % VecTraits<T>::V) == 0;
 }
 
 }  // namespace
@@ -182,8 +340,12 @@ void launch_row_sqnorms(const T* X, float* out, int n, long d,
                         hipStream_t stream) {
   const int block = 256;
   dim3 grid(slab_grid(d, n, block), n);
-  hipLaunchKernelGGL((row_red_kernel<T, false>), grid, dim3(block), 0, stream,
-                     X, nullptr, out, d);
+  if (vec_ok<T>(d))
+    hipLaunchKernelGGL((row_red_kernel<T, false, true>), grid, dim3(block), 0,
+                       stream, X, nullptr, out, d);
+  else
+    hipLaunchKernelGGL((row_red_kernel<T, false, false>), grid, dim3(block), 0,
+                       stream, X, nullptr, out, d);
 }
 
 template <typename T>
@@ -191,8 +353,12 @@ void launch_row_center_sqdists(const T* X, const float* z, float* out, int n,
                                long d, hipStream_t stream) {
   const int block = 256;
   dim3 grid(slab_grid(d, n, block), n);
-  hipLaunchKernelGGL((row_red_kernel<T, true>), grid, dim3(block), 0, stream,
-                     X, z, out, d);
+  if (vec_ok<T>(d))
+    hipLaunchKernelGGL((row_red_kernel<T, true, true>), grid, dim3(block), 0,
+                       stream, X, z, out, d);
+  else
+    hipLaunchKernelGGL((row_red_kernel<T, true, false>), grid, dim3(block), 0,
+                       stream, X, z, out, d);
 }
 
 template <typename T>
@@ -200,34 +366,58 @@ void launch_row_scale(const T* X, const float* s, T* out, int n, long d,
                       hipStream_t stream) {
   const int block = 256;
   dim3 grid(slab_grid(d, n, block), n);
-  hipLaunchKernelGGL((row_scale_kernel<T>), grid, dim3(block), 0, stream, X, s,
-                     out, d);
+  if (vec_ok<T>(d))
+    hipLaunchKernelGGL((row_scale_kernel<T, true>), grid, dim3(block), 0,
+                       stream, X, s, out, d);
+  else
+    hipLaunchKernelGGL((row_scale_kernel<T, false>), grid, dim3(block), 0,
+                       stream, X, s, out, d);
 }
 
 template <typename T>
 void launch_mean_rows(const T* X, const int* idx, int k, T* out, long d,
                       hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL((mean_rows_kernel<T>), dim3(col_grid(d, block)),
-                     dim3(block), 0, stream, X, idx, k, out, d);
+  const bool v = vec_ok<T>(d);
+  const long work = v ? d / VecTraits<T>::V : d;
+  if (v)
+    hipLaunchKernelGGL((gather_mean_kernel<T, true, false>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       idx, k, out, d);
+  else
+    hipLaunchKernelGGL((gather_mean_kernel<T, false, false>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       idx, k, out, d);
 }
 
 template <typename T>
 void launch_group_mean_rows(const T* X, const int* idx, int g, int k, T* out,
                             long d, hipStream_t stream) {
   const int block = 256;
-  dim3 grid(col_grid(d, block) / (g > 4 ? 4 : 1) + 1, g);
-  hipLaunchKernelGGL((group_mean_rows_kernel<T>), grid, dim3(block), 0, stream,
-                     X, idx, k, out, d);
+  const bool v = vec_ok<T>(d);
+  const long work = (v ? d / VecTraits<T>::V : d + 0) / (g > 4 ? 4 : 1) + 1;
+  dim3 grid(col_grid(work, block), g);
+  if (v)
+    hipLaunchKernelGGL((gather_mean_kernel<T, true, true>), grid, dim3(block),
+                       0, stream, X, idx, k, out, d);
+  else
+    hipLaunchKernelGGL((gather_mean_kernel<T, false, true>), grid, dim3(block),
+                       0, stream, X, idx, k, out, d);
 }
 
 template <typename T>
 void launch_bucket_mean(const T* X, const int* perm, int n, int bucket, int nb,
                         T* out, long d, hipStream_t stream) {
   const int block = 256;
-  dim3 grid(col_grid(d, block) / (nb > 4 ? 4 : 1) + 1, nb);
-  hipLaunchKernelGGL((bucket_mean_kernel<T>), grid, dim3(block), 0, stream, X,
-                     perm, n, bucket, out, d);
+  const bool v = vec_ok<T>(d);
+  const long work = (v ? d / VecTraits<T>::V : d + 0) / (nb > 4 ? 4 : 1) + 1;
+  dim3 grid(col_grid(work, block), nb);
+  if (v)
+    hipLaunchKernelGGL((bucket_mean_kernel<T, true>), grid, dim3(block), 0,
+                       stream, X, perm, n, bucket, out, d);
+  else
+    hipLaunchKernelGGL((bucket_mean_kernel<T, false>), grid, dim3(block), 0,
+                       stream, X, perm, n, bucket, out, d);
 }
 
 template <typename T>
@@ -235,9 +425,16 @@ void launch_weiszfeld_update(const T* X, const float* z, const float* dist2,
                              float* z_new, float* shift2, int n, long d,
                              float eps, hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL((weiszfeld_update_kernel<T>), dim3(col_grid(d, block)),
-                     dim3(block), 0, stream, X, z, dist2, z_new, shift2, n, d,
-                     eps);
+  const bool v = vec_ok<T>(d);
+  const long work = v ? d / VecTraits<T>::V : d;
+  if (v)
+    hipLaunchKernelGGL((weiszfeld_update_kernel<T, true>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       z, dist2, z_new, shift2, n, d, eps);
+  else
+    hipLaunchKernelGGL((weiszfeld_update_kernel<T, false>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       z, dist2, z_new, shift2, n, d, eps);
 }
 
 template <typename T>
@@ -245,9 +442,16 @@ void launch_cc_update(const T* X, const float* v, const float* dist2,
                       float* v_new, int n, long d, float c_tau, float eps,
                       hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL((cc_update_kernel<T>), dim3(col_grid(d, block)),
-                     dim3(block), 0, stream, X, v, dist2, v_new, n, d, c_tau,
-                     eps);
+  const bool vo = vec_ok<T>(d);
+  const long work = vo ? d / VecTraits<T>::V : d;
+  if (vo)
+    hipLaunchKernelGGL((cc_update_kernel<T, true>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       v, dist2, v_new, n, d, c_tau, eps);
+  else
+    hipLaunchKernelGGL((cc_update_kernel<T, false>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       v, dist2, v_new, n, d, c_tau, eps);
 }
 
 // explicit instantiations for bind.cpp
