@@ -93,9 +93,15 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long col = col0; col < d; col += stride) {
     float v[P];
+    // load from a CLAMPED row unconditionally and pad with a select after:
+    // a branch around each load would serialize the 64-load burst behind
+    // per-load waits (guide §5 ".s-level traps" (c)).
 #pragma unroll
     for (int i = 0; i < P; ++i)
-      v[i] = (i < n) ? to_f<T>(X[(long)i * d + col]) : PAD;
+      v[i] = to_f<T>(X[(long)(i < n ? i : n - 1) * d + col]);
+#pragma unroll
+    for (int i = 0; i < P; ++i)
+      if (i >= n) v[i] = PAD;
     bitonic_sort_reg<P>(v);  // in place: meamed only needs sorted order
 
     const float med_lo = extract_at<P>(v, (n - 1) >> 1);

@@ -24,29 +24,31 @@ constexpr int TILE = 64;    // output tile side
 constexpr int WAVES = 16;   // 4x4 waves of 16x16
 constexpr int SYNC_EVERY = 8;
 
-// Load an 8-element bf16 fragment from row `row` at column k0 (guarded).
-DEV bf16x8 load_frag_bf16(const __hip_bfloat16* __restrict__ X, int row, int n,
-                          long d, long k0, bool vec_ok) {
+// Unconditional fragment load from a pre-clamped row pointer. Out-of-range
+// rows are CLAMPED to a valid row by the caller and zeroed with a select
+// AFTER the load — a branch around the load would serialize the whole
+// K-stream behind per-load vmcnt(0) waits (guide §5 ".s-level traps" (c)).
+template <bool VEC>
+DEV bf16x8 load_frag8(const __hip_bfloat16* p, bool ok) {
   bf16x8 out;
-  if (row >= n) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) out[j] = (__bf16)0.0f;
-    return out;
-  }
-  const __hip_bfloat16* p = X + (long)row * d + k0;
-  if (vec_ok) {
+  if (VEC) {
     out = *reinterpret_cast<const bf16x8*>(p);
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
       out[j] = *reinterpret_cast<const __bf16*>(p + j);
   }
+  if (!ok) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = (__bf16)0.0f;
+  }
   return out;
 }
 
+template <bool VEC>
 __global__ void gram_bf16_kernel(const __hip_bfloat16* __restrict__ X,
                                  float* __restrict__ G, int n, long d,
-                                 long k_per_block, int vec_ok) {
+                                 long k_per_block) {
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wr = wave >> 2, wc = wave & 3;
@@ -59,19 +61,41 @@ __global__ void gram_bf16_kernel(const __hip_bfloat16* __restrict__ X,
   f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
   const int frag_row = lane & 15;
   const long frag_k = (long)(lane >> 4) * 8;
+  const int a_row = row_base + frag_row;
+  const int b_row = col_base + frag_row;
+  const bool a_ok = a_row < n, b_ok = b_row < n;
+  // loop-invariant row pointers (clamped): no 64-bit mul in the K-loop
+  const __hip_bfloat16* a_base = X + (long)min(a_row, n - 1) * d + frag_k;
+  const __hip_bfloat16* b_base = X + (long)min(b_row, n - 1) * d + frag_k;
 
   // K-unroll x4: issue all 8 fragment loads before the MFMA cluster so the
   // HBM latency of one iteration hides under the previous one's MFMAs.
+  // Fully in-range tiles (always, when n % 16 == 0) take the select-free
+  // loop — the zero-select costs 32 VALU/iteration otherwise.
+  const bool tile_ok = ((int)(blockIdx.y + 1) * TILE <= n) &&
+                       ((int)(blockIdx.z + 1) * TILE <= n);
   int step = 0;
   long k0 = k_lo;
+  if (tile_ok) {
+    for (; k0 + 128 <= k_hi; k0 += 128) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        a[u] = load_frag8<VEC>(a_base + k0 + u * 32, true);
+        b[u] = load_frag8<VEC>(b_base + k0 + u * 32, true);
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc, 0, 0, 0);
+      if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
+    }
+  }
   for (; k0 + 128 <= k_hi; k0 += 128) {
     bf16x8 a[4], b[4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      a[u] = load_frag_bf16(X, row_base + frag_row, n, d, k0 + u * 32 + frag_k,
-                            vec_ok);
-      b[u] = load_frag_bf16(X, col_base + frag_row, n, d, k0 + u * 32 + frag_k,
-                            vec_ok);
+      a[u] = load_frag8<VEC>(a_base + k0 + u * 32, a_ok);
+      b[u] = load_frag8<VEC>(b_base + k0 + u * 32, b_ok);
     }
 #pragma unroll
     for (int u = 0; u < 4; ++u)
@@ -79,30 +103,21 @@ __global__ void gram_bf16_kernel(const __hip_bfloat16* __restrict__ X,
     if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
   }
   for (; k0 + 32 <= k_hi; k0 += 32) {
-    const bf16x8 a =
-        load_frag_bf16(X, row_base + frag_row, n, d, k0 + frag_k, vec_ok);
-    const bf16x8 b =
-        load_frag_bf16(X, col_base + frag_row, n, d, k0 + frag_k, vec_ok);
+    const bf16x8 a = load_frag8<VEC>(a_base + k0, a_ok);
+    const bf16x8 b = load_frag8<VEC>(b_base + k0, b_ok);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
   }
-  // K tail (< 32): scalar FMA into the same accumulator positions would
-  // need the C layout; do it as a 1-wide MFMA with zero-padded fragments.
-  const long k_rem = k_hi - ((k_hi - k_lo) / 32) * 32 - k_lo;
-  if (k_rem > 0) {
-    const long k0 = k_hi - k_rem;
+  // K tail (< 32): one MFMA on zero-padded fragments.
+  if (k0 < k_hi) {
     bf16x8 a, b;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const long k = k0 + frag_k + j;
       const bool in = k < k_hi;
-      a[j] = (in && row_base + frag_row < n)
-                 ? *reinterpret_cast<const __bf16*>(
-                       X + (long)(row_base + frag_row) * d + k)
-                 : (__bf16)0.0f;
-      b[j] = (in && col_base + frag_row < n)
-                 ? *reinterpret_cast<const __bf16*>(
-                       X + (long)(col_base + frag_row) * d + k)
-                 : (__bf16)0.0f;
+      a[j] = (in && a_ok) ? *reinterpret_cast<const __bf16*>(a_base + k0 + j)
+                          : (__bf16)0.0f;
+      b[j] = (in && b_ok) ? *reinterpret_cast<const __bf16*>(b_base + k0 + j)
+                          : (__bf16)0.0f;
     }
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
   }
@@ -132,16 +147,24 @@ __global__ void gram_f32_kernel(const float* __restrict__ X,
   f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
   const int a_row = row_base + (lane & 15);
   const int b_row = col_base + (lane & 15);
+  const bool a_ok = a_row < n, b_ok = b_row < n;
   const long lane_k = lane >> 4;  // k = k0 + lane_k, K-step 4
+  const float* a_base = X + (long)min(a_row, n - 1) * d + lane_k;
+  const float* b_base = X + (long)min(b_row, n - 1) * d + lane_k;
 
   int step = 0;
   long k0 = k_lo;
   for (; k0 + 32 <= k_hi; k0 += 32) {
+    float av[8], bv[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const long k = k0 + u * 4 + lane_k;
-      const float a = (a_row < n) ? X[(long)a_row * d + k] : 0.0f;
-      const float b = (b_row < n) ? X[(long)b_row * d + k] : 0.0f;
+      av[u] = a_base[k0 + u * 4];
+      bv[u] = b_base[k0 + u * 4];
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const float a = a_ok ? av[u] : 0.0f;
+      const float b = b_ok ? bv[u] : 0.0f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
     if (++step == SYNC_EVERY) { step = 0; __syncthreads(); }
@@ -149,8 +172,8 @@ __global__ void gram_f32_kernel(const float* __restrict__ X,
   for (; k0 < k_hi; k0 += 4) {
     const long k = k0 + lane_k;
     const bool in = k < k_hi;
-    const float a = (in && a_row < n) ? X[(long)a_row * d + k] : 0.0f;
-    const float b = (in && b_row < n) ? X[(long)b_row * d + k] : 0.0f;
+    const float a = (in && a_ok) ? a_base[k0] : 0.0f;
+    const float b = (in && b_ok) ? b_base[k0] : 0.0f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
 
@@ -172,6 +195,7 @@ inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
   if (want > max_by_d) want = max_by_d;
   splitk = (int)want;
   k_per_block = (d + splitk - 1) / splitk;
+  k_per_block = ((k_per_block + 127) / 128) * 128;  // keep the vector path
 }
 
 }  // namespace
@@ -182,9 +206,12 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
   dim3 grid(splitk, tiles, tiles);
-  const int vec_ok = ((d % 8) == 0) ? 1 : 0;
-  hipLaunchKernelGGL(gram_bf16_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
-                     n, d, kpb, vec_ok);
+  if ((d % 8) == 0 && (kpb % 8) == 0)
+    hipLaunchKernelGGL((gram_bf16_kernel<true>), grid, dim3(WAVES * 64), 0,
+                       stream, X, G, n, d, kpb);
+  else
+    hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
+                       stream, X, G, n, d, kpb);
 }
 
 void launch_gram_f32(const float* X, float* G, int n, long d,
