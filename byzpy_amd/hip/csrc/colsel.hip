@@ -84,7 +84,7 @@ DEV float extract_at(const float (&v)[P], int pos) {
 // MEAMED instantiations get __launch_bounds__(256, 2) via this trait so the
 // allocator may use 256 VGPRs instead of spilling, while MEDIAN/TRIMMED
 // keep the default 4-waves/SIMD occupancy.
-template <int P, int MODE, typename T>
+template <int P, int MODE, bool EXACT, typename T>
 __global__ void
 __launch_bounds__(256, (MODE == 2 && P >= 64) ? 2 : 4)
 colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
@@ -93,15 +93,26 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long col = col0; col < d; col += stride) {
     float v[P];
-    // load from a CLAMPED row unconditionally and pad with a select after:
-    // a branch around each load would serialize the 64-load burst behind
-    // per-load waits (guide §5 ".s-level traps" (c)).
+    if (EXACT) {
+      // n == P: plain strided loads; the compiler strength-reduces the
+      // address chain and batches the load burst.
 #pragma unroll
-    for (int i = 0; i < P; ++i)
-      v[i] = to_f<T>(X[(long)(i < n ? i : n - 1) * d + col]);
+      for (int i = 0; i < P; ++i) v[i] = to_f<T>(X[(long)i * d + col]);
+    } else {
+      // n < P: walk a row pointer that stops advancing at the last valid
+      // row (one live address pair — batching P clamped addresses spills),
+      // then pad with selects. No branch around any load (guide §5
+      // ".s-level traps" (c)).
+      const T* p = X + col;
 #pragma unroll
-    for (int i = 0; i < P; ++i)
-      if (i >= n) v[i] = PAD;
+      for (int i = 0; i < P; ++i) {
+        v[i] = to_f<T>(*p);
+        if (i + 1 < n) p += d;
+      }
+#pragma unroll
+      for (int i = 0; i < P; ++i)
+        if (i >= n) v[i] = PAD;
+    }
     bitonic_sort_reg<P>(v);  // in place: meamed only needs sorted order
 
     const float med_lo = extract_at<P>(v, (n - 1) >> 1);
@@ -203,23 +214,25 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
     const int block = 256;
     const long want = (d + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
+#define DISPATCH_ONE(P, MODE, EXACT)                                           \
+  hipLaunchKernelGGL((colsel_reg_kernel<P, MODE, EXACT, T>), dim3(grid),       \
+                     dim3(block), 0, stream, X, out, n, d, f)
 #define DISPATCH_REG(P)                                                        \
   do {                                                                         \
+    const bool exact = (n == P);                                               \
     if (mode == MEDIAN)                                                        \
-      hipLaunchKernelGGL((colsel_reg_kernel<P, MEDIAN, T>), dim3(grid),        \
-                         dim3(block), 0, stream, X, out, n, d, f);             \
+      exact ? DISPATCH_ONE(P, MEDIAN, true) : DISPATCH_ONE(P, MEDIAN, false);  \
     else if (mode == TRIMMED)                                                  \
-      hipLaunchKernelGGL((colsel_reg_kernel<P, TRIMMED, T>), dim3(grid),       \
-                         dim3(block), 0, stream, X, out, n, d, f);             \
+      exact ? DISPATCH_ONE(P, TRIMMED, true) : DISPATCH_ONE(P, TRIMMED, false);\
     else                                                                       \
-      hipLaunchKernelGGL((colsel_reg_kernel<P, MEAMED, T>), dim3(grid),        \
-                         dim3(block), 0, stream, X, out, n, d, f);             \
+      exact ? DISPATCH_ONE(P, MEAMED, true) : DISPATCH_ONE(P, MEAMED, false);  \
   } while (0)
     if (n <= 8) DISPATCH_REG(8);
     else if (n <= 16) DISPATCH_REG(16);
     else if (n <= 32) DISPATCH_REG(32);
     else DISPATCH_REG(64);
 #undef DISPATCH_REG
+#undef DISPATCH_ONE
   } else {
     int P = 128;
     while (P < n) P <<= 1;  // 128/256/512

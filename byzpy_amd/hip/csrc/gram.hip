@@ -186,6 +186,126 @@ __global__ void gram_f32_kernel(const float* __restrict__ X,
   }
 }
 
+// ---------------------------------------------------------------------------
+// LDS-staged bf16 Gram (v3, the fast path for d % 8 == 0).
+//
+// Per chunk (64 rows x 128 k = 16 KB): 1024 threads register-load one 16 B
+// piece each (T14 split: loads issue BEFORE the compute phase, the ds_write
+// lands after the barrier), waves then read MFMA fragments from LDS —
+// HBM/L2 sees each line ONCE per block instead of 8x. LDS slots are
+// XOR-swizzled (slot ^= row & 15) so a 16-lane ds_read_b128 group touches
+// 16 distinct 16 B slots (guide §5.5 T2; linear layout would be 16-way).
+// DIAG=true (blockIdx.y == blockIdx.z, always at n <= 64): the A and B
+// images coincide — staged once, halving staging traffic.
+// ---------------------------------------------------------------------------
+
+template <bool DIAG>
+__global__ void __launch_bounds__(1024, 2)
+gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
+                     float* __restrict__ G, int n, long d, long k_per_block) {
+  constexpr int BK = 128;                       // k elements per chunk
+  constexpr int CHUNK_BYTES = TILE * BK * 2;    // 16 KB
+  __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
+  char* bufA[2] = {smem, smem + CHUNK_BYTES};
+  char* bufB[2] = {DIAG ? smem : smem + 2 * CHUNK_BYTES,
+                   DIAG ? smem + CHUNK_BYTES : smem + 3 * CHUNK_BYTES};
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int row_base = blockIdx.y * TILE;
+  const int col_base = blockIdx.z * TILE;
+
+  const long k_lo = (long)blockIdx.x * k_per_block;
+  const long k_hi = min(d, k_lo + k_per_block);
+  const long nchunks = (k_hi - k_lo + BK - 1) / BK;
+
+  // stage geometry: thread t handles LDS row st_row, memory slot st_slot;
+  // the CONTENT of that slot is global slot (st_slot ^ (st_row & 15)).
+  const int st_row = t >> 4;
+  const int st_slot = t & 15;
+  const int src_slot = st_slot ^ (st_row & 15);
+  const int a_rows = min(TILE, n - row_base);
+  const int b_rows = min(TILE, n - col_base);
+  const __hip_bfloat16* a_src =
+      X + (long)(row_base + min(st_row, a_rows - 1)) * d + (long)src_slot * 8;
+  const __hip_bfloat16* b_src =
+      X + (long)(col_base + min(st_row, b_rows - 1)) * d + (long)src_slot * 8;
+
+  bf16x8 ra, rb;
+  auto stage_load = [&](long c) {
+    const long k0 = k_lo + c * BK;
+    const bool full = (k0 + BK <= k_hi);
+    if (full && a_rows == TILE) {
+      ra = *reinterpret_cast<const bf16x8*>(a_src + k0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const long k = k0 + (long)src_slot * 8 + j;
+        ra[j] = (st_row < a_rows && k < k_hi)
+                    ? *reinterpret_cast<const __bf16*>(a_src + k0 + j)
+                    : (__bf16)0.0f;
+      }
+    }
+    if (!DIAG) {
+      if (full && b_rows == TILE) {
+        rb = *reinterpret_cast<const bf16x8*>(b_src + k0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const long k = k0 + (long)src_slot * 8 + j;
+          rb[j] = (st_row < b_rows && k < k_hi)
+                      ? *reinterpret_cast<const __bf16*>(b_src + k0 + j)
+                      : (__bf16)0.0f;
+        }
+      }
+    }
+  };
+  auto stage_write = [&](int which) {
+    *reinterpret_cast<bf16x8*>(bufA[which] + st_row * 256 + st_slot * 16) = ra;
+    if (!DIAG)
+      *reinterpret_cast<bf16x8*>(bufB[which] + st_row * 256 + st_slot * 16) =
+          rb;
+  };
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  const int rowA = wr * 16 + (lane & 15);
+  const int rowB = wc * 16 + (lane & 15);
+  const int grp = lane >> 4;  // 16 B slot group within the K step
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  for (long c = 0; c < nchunks; ++c) {
+    if (c + 1 < nchunks) stage_load(c + 1);
+    const char* A = bufA[c & 1];
+    const char* B = bufB[c & 1];
+#pragma unroll
+    for (int step = 0; step < 4; ++step) {
+      const int q = step * 4 + grp;
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          A + rowA * 256 + ((q ^ (rowA & 15)) * 16));
+      const bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          B + rowB * 256 + ((q ^ (rowB & 15)) * 16));
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+    if (c + 1 < nchunks) {
+      stage_write((c + 1) & 1);
+      __syncthreads();
+    }
+  }
+
+  const int out_col = col_base + wc * 16 + (lane & 15);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
+    if (out_row < n && out_col < n)
+      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+  }
+}
+
 inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
   const int tiles = (n + TILE - 1) / TILE;
   const long tile_blocks = (long)tiles * tiles;
@@ -206,12 +326,19 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
   dim3 grid(splitk, tiles, tiles);
-  if ((d % 8) == 0 && (kpb % 8) == 0)
-    hipLaunchKernelGGL((gram_bf16_kernel<true>), grid, dim3(WAVES * 64), 0,
-                       stream, X, G, n, d, kpb);
-  else
+  if ((d % 8) == 0 && (kpb % 128) == 0) {
+    if (tiles == 1)
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<true>), grid, dim3(WAVES * 64),
+                         0, stream, X, G, n, d, kpb);
+    else
+      // off-diagonal tiles need separate A/B images; diagonal blocks of a
+      // multi-tile grid still produce correct results with DIAG=false.
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<false>), grid, dim3(WAVES * 64),
+                         0, stream, X, G, n, d, kpb);
+  } else {
     hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
                        stream, X, G, n, d, kpb);
+  }
 }
 
 void launch_gram_f32(const float* X, float* G, int n, long d,
