@@ -217,21 +217,23 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
 #define DISPATCH_ONE(P, MODE, EXACT)                                           \
   hipLaunchKernelGGL((colsel_reg_kernel<P, MODE, EXACT, T>), dim3(grid),       \
                      dim3(block), 0, stream, X, out, n, d, f)
+#define DISPATCH_MODE(P, MODE)                                                 \
+  do {                                                                         \
+    if (n == P) DISPATCH_ONE(P, MODE, true);                                   \
+    else        DISPATCH_ONE(P, MODE, false);                                  \
+  } while (0)
 #define DISPATCH_REG(P)                                                        \
   do {                                                                         \
-    const bool exact = (n == P);                                               \
-    if (mode == MEDIAN)                                                        \
-      exact ? DISPATCH_ONE(P, MEDIAN, true) : DISPATCH_ONE(P, MEDIAN, false);  \
-    else if (mode == TRIMMED)                                                  \
-      exact ? DISPATCH_ONE(P, TRIMMED, true) : DISPATCH_ONE(P, TRIMMED, false);\
-    else                                                                       \
-      exact ? DISPATCH_ONE(P, MEAMED, true) : DISPATCH_ONE(P, MEAMED, false);  \
+    if (mode == MEDIAN) DISPATCH_MODE(P, MEDIAN);                              \
+    else if (mode == TRIMMED) DISPATCH_MODE(P, TRIMMED);                       \
+    else DISPATCH_MODE(P, MEAMED);                                             \
   } while (0)
     if (n <= 8) DISPATCH_REG(8);
     else if (n <= 16) DISPATCH_REG(16);
     else if (n <= 32) DISPATCH_REG(32);
     else DISPATCH_REG(64);
 #undef DISPATCH_REG
+#undef DISPATCH_MODE
 #undef DISPATCH_ONE
   } else {
     int P = 128;

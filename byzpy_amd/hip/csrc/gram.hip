@@ -206,9 +206,12 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   constexpr int BK = 128;                       // k elements per chunk
   constexpr int CHUNK_BYTES = TILE * BK * 2;    // 16 KB
   __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
-  char* bufA[2] = {smem, smem + CHUNK_BYTES};
-  char* bufB[2] = {DIAG ? smem : smem + 2 * CHUNK_BYTES,
-                   DIAG ? smem + CHUNK_BYTES : smem + 3 * CHUNK_BYTES};
+  // no pointer ARRAYS into LDS (clang rejects the addrspace-cast array
+  // initializer); select buffers with ternaries instead
+  char* const bufA0 = smem;
+  char* const bufA1 = smem + CHUNK_BYTES;
+  char* const bufB0 = DIAG ? smem : smem + 2 * CHUNK_BYTES;
+  char* const bufB1 = DIAG ? smem + CHUNK_BYTES : smem + 3 * CHUNK_BYTES;
 
   const int t = threadIdx.x;
   const int wave = t >> 6;
@@ -263,10 +266,12 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
     }
   };
   auto stage_write = [&](int which) {
-    *reinterpret_cast<bf16x8*>(bufA[which] + st_row * 256 + st_slot * 16) = ra;
-    if (!DIAG)
-      *reinterpret_cast<bf16x8*>(bufB[which] + st_row * 256 + st_slot * 16) =
-          rb;
+    char* a = which ? bufA1 : bufA0;
+    *reinterpret_cast<bf16x8*>(a + st_row * 256 + st_slot * 16) = ra;
+    if (!DIAG) {
+      char* b = which ? bufB1 : bufB0;
+      *reinterpret_cast<bf16x8*>(b + st_row * 256 + st_slot * 16) = rb;
+    }
   };
 
   f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
@@ -279,8 +284,8 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   __syncthreads();
   for (long c = 0; c < nchunks; ++c) {
     if (c + 1 < nchunks) stage_load(c + 1);
-    const char* A = bufA[c & 1];
-    const char* B = bufB[c & 1];
+    const char* A = (c & 1) ? bufA1 : bufA0;
+    const char* B = (c & 1) ? bufB1 : bufB0;
 #pragma unroll
     for (int step = 0; step < 4; ++step) {
       const int q = step * 4 + grp;
