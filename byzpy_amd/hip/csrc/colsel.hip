@@ -84,7 +84,7 @@ DEV float extract_at(const float (&v)[P], int pos) {
 // MEAMED instantiations get __launch_bounds__(256, 2) via this trait so the
 // allocator may use 256 VGPRs instead of spilling, while MEDIAN/TRIMMED
 // keep the default 4-waves/SIMD occupancy.
-template <int P, int MODE, bool EXACT, typename T>
+template <int P, int MODE, typename T>
 __global__ void
 __launch_bounds__(256, (MODE == 2 && P >= 64) ? 2 : 4)
 colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
@@ -93,16 +93,11 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long col = col0; col < d; col += stride) {
     float v[P];
-    if (EXACT) {
-      // n == P: plain strided loads; the compiler strength-reduces the
-      // address chain and batches the load burst.
-#pragma unroll
-      for (int i = 0; i < P; ++i) v[i] = to_f<T>(X[(long)i * d + col]);
-    } else {
-      // n < P: walk a row pointer that stops advancing at the last valid
-      // row (one live address pair — batching P clamped addresses spills),
-      // then pad with selects. No branch around any load (guide §5
-      // ".s-level traps" (c)).
+    // Walk a row pointer that stops advancing at the last valid row (ONE
+    // live address pair — batching P strided addresses spills ~P VGPR
+    // pairs), then pad with selects. No branch around any load (guide §5
+    // ".s-level traps" (c)).
+    {
       const T* p = X + col;
 #pragma unroll
       for (int i = 0; i < P; ++i) {
@@ -214,26 +209,20 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
     const int block = 256;
     const long want = (d + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
-#define DISPATCH_ONE(P, MODE, EXACT)                                           \
-  hipLaunchKernelGGL((colsel_reg_kernel<P, MODE, EXACT, T>), dim3(grid),       \
+#define DISPATCH_ONE(P, MODE)                                                  \
+  hipLaunchKernelGGL((colsel_reg_kernel<P, MODE, T>), dim3(grid),              \
                      dim3(block), 0, stream, X, out, n, d, f)
-#define DISPATCH_MODE(P, MODE)                                                 \
-  do {                                                                         \
-    if (n == P) DISPATCH_ONE(P, MODE, true);                                   \
-    else        DISPATCH_ONE(P, MODE, false);                                  \
-  } while (0)
 #define DISPATCH_REG(P)                                                        \
   do {                                                                         \
-    if (mode == MEDIAN) DISPATCH_MODE(P, MEDIAN);                              \
-    else if (mode == TRIMMED) DISPATCH_MODE(P, TRIMMED);                       \
-    else DISPATCH_MODE(P, MEAMED);                                             \
+    if (mode == MEDIAN) DISPATCH_ONE(P, MEDIAN);                               \
+    else if (mode == TRIMMED) DISPATCH_ONE(P, TRIMMED);                        \
+    else DISPATCH_ONE(P, MEAMED);                                              \
   } while (0)
     if (n <= 8) DISPATCH_REG(8);
     else if (n <= 16) DISPATCH_REG(16);
     else if (n <= 32) DISPATCH_REG(32);
     else DISPATCH_REG(64);
 #undef DISPATCH_REG
-#undef DISPATCH_MODE
 #undef DISPATCH_ONE
   } else {
     int P = 128;
