@@ -1,0 +1,92 @@
+"""RCCL parameter-server engine — one process per GPU over xGMI.
+
+The MI355X-native replacement for the reference's actor-RPC gather
+(SURVEY.md C1/C2): every rank computes its worker gradients locally, the
+n-sharded gradient matrix is transposed to d-sharded with ONE
+all-to-all (each GPU then owns a contiguous d/world slice of every worker
+gradient), robust aggregation runs shard-local with only tiny (n,)/(n,n)
+all-reduces (byzpy_amd/parallel/sharded.py), and the d-sharded update is
+either consumed in place or assembled with an all-gather. xGMI note: the
+all-to-all form spreads traffic over all 7 p2p links instead of
+serializing a ring.
+
+BASELINE config 3 shape: ResNet-50 grads, 8 honest + 3 SignFlip workers,
+TrimmedMean, 8 GPUs.
+"""
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+
+from byzpy_amd.parallel import dist as pdist
+from byzpy_amd.parallel import sharded
+
+
+class RcclParameterServer:
+    """Symmetric (serverless) PS: every rank ends each round with the same
+    d-shard of the robust aggregate (or the full vector with gather=True).
+
+    ``local_gradient_fns``: this rank's worker-gradient callables (honest
+    or byzantine), each returning a flat (d,) tensor on this rank's GPU.
+    The global worker count n = sum over ranks of len(local_gradient_fns)
+    and must be equal per rank (pad with byzantine/duplicate workers if
+    needed).
+    """
+
+    def __init__(
+        self,
+        local_gradient_fns: List[Callable[[], torch.Tensor]],
+        aggregate_fn: Callable[[torch.Tensor], torch.Tensor],
+        *,
+        gather_result: bool = False,
+    ) -> None:
+        self.local_gradient_fns = list(local_gradient_fns)
+        self.aggregate_fn = aggregate_fn
+        self.gather_result = gather_result
+
+    def round(self) -> torch.Tensor:
+        world = pdist.get_world_size()
+        grads = [fn().reshape(-1) for fn in self.local_gradient_fns]
+        local = torch.stack(grads)  # (n_local, d)
+        n_local, d = local.shape
+        if world == 1:
+            out = self.aggregate_fn(local)
+            return out
+        # pad d to a multiple of world so shards are equal-sized
+        pad = (-d) % world
+        if pad:
+            local = torch.nn.functional.pad(local, (0, pad))
+        d_pad = d + pad
+        shard = d_pad // world
+        # (n_local, d) -> world blocks of (n_local, shard) -> all-to-all ->
+        # this rank holds every worker's shard: (n_global, shard)
+        blocks = local.reshape(n_local, world, shard).transpose(0, 1).contiguous()
+        recv = pdist.all_to_all_rows(blocks.reshape(world * n_local, shard))
+        X_shard = recv.reshape(world, n_local, shard).reshape(world * n_local, shard)
+        out_shard = self.aggregate_fn(X_shard)  # (shard,)
+        if not self.gather_result:
+            return out_shard
+        full = pdist.all_gather_rows(out_shard.reshape(1, -1)).reshape(-1)
+        return full[:d]
+
+
+def trimmed_mean_aggregate(f: int):
+    def _agg(X_shard: torch.Tensor) -> torch.Tensor:
+        return sharded.trimmed_mean(X_shard, f)
+
+    return _agg
+
+
+def median_aggregate():
+    def _agg(X_shard: torch.Tensor) -> torch.Tensor:
+        return sharded.median(X_shard)
+
+    return _agg
+
+
+def multi_krum_aggregate(f: int, q: int):
+    def _agg(X_shard: torch.Tensor) -> torch.Tensor:
+        return sharded.multi_krum(X_shard, f, q)
+
+    return _agg

@@ -1,0 +1,67 @@
+"""CLI, checkpoint layout and tracing tests."""
+import json
+
+import torch
+
+from byzpy_amd.cli import main as cli_main
+from byzpy_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+from byzpy_amd.utils.tracing import mark, trace_range
+
+
+def test_cli_version(capsys):
+    assert cli_main(["version"]) == 0
+    out = capsys.readouterr().out.strip()
+    assert out.count(".") >= 1
+
+
+def test_cli_doctor_json(capsys):
+    assert cli_main(["doctor", "--format", "json"]) == 0
+    info = json.loads(capsys.readouterr().out)
+    assert "torch" in info
+    assert "hip_extension" in info
+
+
+def test_cli_list(capsys):
+    assert cli_main(["list", "aggregators"]) == 0
+    out = capsys.readouterr().out
+    for name in ("CoordinateWiseMedian", "MultiKrum", "GeometricMedian", "CAF"):
+        assert name in out
+    assert cli_main(["list", "attacks"]) == 0
+    assert "EmpireAttack" in capsys.readouterr().out
+    assert cli_main(["list", "pre-aggregators"]) == 0
+    assert "Bucketing" in capsys.readouterr().out
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    model_state = {"w": torch.randn(4, 3), "b": torch.randn(3)}
+    agg_state = {"center": torch.randn(12)}
+    save_checkpoint(
+        str(tmp_path),
+        round_idx=17,
+        model_state=model_state,
+        aggregator_state=agg_state,
+    )
+    out = load_checkpoint(str(tmp_path))
+    assert out["meta"]["round"] == 17
+    assert torch.allclose(out["model_state"]["w"], model_state["w"])
+    assert torch.allclose(out["aggregator_state"]["center"], agg_state["center"])
+
+
+def test_checkpoint_sharded(tmp_path):
+    for rank in range(2):
+        save_checkpoint(
+            str(tmp_path),
+            round_idx=3,
+            model_state={"shard": torch.full((4,), float(rank))},
+            rank=rank,
+            world_size=2,
+        )
+    out0 = load_checkpoint(str(tmp_path), rank=0)
+    out1 = load_checkpoint(str(tmp_path), rank=1)
+    assert float(out0["model_state"]["shard"][0]) == 0.0
+    assert float(out1["model_state"]["shard"][0]) == 1.0
+
+
+def test_tracing_noop_on_cpu():
+    with trace_range("test"):
+        mark("point")  # must not raise regardless of roctx availability

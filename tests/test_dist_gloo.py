@@ -1,0 +1,148 @@
+"""Multi-process distributed tests over gloo, world_size=2 (SURVEY.md §4
+pattern 6b + the MI355X sharded-aggregation engine). These exercise the
+exact collective call patterns the RCCL path uses on GPUs."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as tmp
+
+import byzpy_amd.ops.functional as F
+
+WORLD = 2
+
+
+def _run_workers(target, *args):
+    ctx = tmp.get_context("spawn")
+    port = 29531 + abs(hash(target.__name__)) % 2000
+    procs = []
+    for rank in range(WORLD):
+        p = ctx.Process(target=_entry, args=(target, rank, port) + args)
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(120)
+    for p in procs:
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+def _entry(target, rank, port, *args):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    os.environ["LOCAL_RANK"] = str(rank)
+    import torch.distributed as dist
+
+    from byzpy_amd.parallel import dist as pdist
+
+    pdist.init_from_env()
+    try:
+        target(rank, *args)
+    finally:
+        dist.destroy_process_group()
+
+
+# -- worker bodies (module-level: spawn-picklable) --------------------------
+
+
+def _body_sharded_ops(rank):
+    from byzpy_amd.parallel import sharded
+    from byzpy_amd.parallel.dist import column_shard
+
+    g = torch.Generator().manual_seed(7)
+    X = torch.randn(10, 64, generator=g)  # identical on both ranks
+    Xs = column_shard(X, rank)
+    lo, hi = rank * 32, rank * 32 + 32
+
+    med = sharded.median(Xs)
+    assert torch.allclose(med, F.median(X)[lo:hi], atol=1e-5)
+
+    mk = sharded.multi_krum(Xs, 2, 3)
+    assert torch.allclose(mk, F.multi_krum(X, 2, 3)[lo:hi], atol=1e-4)
+
+    cge = sharded.cge(Xs, 2)
+    assert torch.allclose(cge, F.cge(X, 2)[lo:hi], atol=1e-4)
+
+    gm = sharded.geometric_median(Xs, tol=1e-8, max_iter=100)
+    ref = F.geometric_median(X, tol=1e-8, max_iter=100)
+    assert (gm - ref[lo:hi]).norm() < 1e-3
+
+    cc = sharded.centered_clipping(Xs, c_tau=0.5, M=5)
+    ref_cc = F.centered_clipping(X, c_tau=0.5, M=5)
+    assert torch.allclose(cc, ref_cc[lo:hi], atol=1e-4)
+
+    cl = sharded.clip_rows(Xs, 1.0)
+    assert torch.allclose(cl, F.clip_rows(X, 1.0)[:, lo:hi], atol=1e-5)
+
+    nn_ = sharded.nnm(Xs, 2)
+    assert torch.allclose(nn_, F.nnm(X, 2)[:, lo:hi], atol=1e-4)
+
+
+def _body_rccl_ps(rank):
+    from byzpy_amd.engine.parameter_server.rccl import (
+        RcclParameterServer,
+        trimmed_mean_aggregate,
+    )
+
+    d = 37  # odd on purpose: exercises the pad path
+
+    def honest(v):
+        return lambda: torch.full((d,), float(v))
+
+    def byz():
+        return lambda: torch.full((d,), 1.0e6)
+
+    fns = [honest(1 + rank), byz()] if rank == 1 else [honest(1 + rank), honest(5)]
+    ps = RcclParameterServer(fns, trimmed_mean_aggregate(1), gather_result=True)
+    out = ps.round()
+    assert out.shape == (d,)
+    # n=4 grads: {1, 5, 2, 1e6}; trimmed f=1 -> mean(2, 5) = 3.5
+    assert torch.allclose(out, torch.full((d,), 3.5), atol=1e-4), out[:3]
+
+
+def _body_rccl_p2p(rank):
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+    from byzpy_amd.engine.peer_to_peer.rccl import RcclPeerToPeer
+
+    state = {"params": torch.full((5,), float(rank + 1))}
+
+    def half_step():
+        return state["params"] * 2
+
+    def write(params):
+        state["params"] = params
+
+    p2p = RcclPeerToPeer(half_step, write, CoordinateWiseMedian())
+    out = p2p.round()
+    # vectors: rank0 -> 2, rank1 -> 4; median = 3
+    assert torch.allclose(out, torch.full((5,), 3.0), atol=1e-5)
+    assert torch.allclose(state["params"], torch.full((5,), 3.0), atol=1e-5)
+
+
+def _body_collectives(rank):
+    from byzpy_amd.parallel import dist as pdist
+
+    t = torch.full((4,), float(rank + 1))
+    s = pdist.all_reduce_(t.clone())
+    assert torch.allclose(s, torch.full((4,), 3.0))
+    g = pdist.all_gather_rows(t.reshape(1, -1))
+    assert g.shape == (2, 4)
+    a2a = pdist.all_to_all_rows(torch.arange(4.0).reshape(2, 2) + 10 * rank)
+    assert a2a.shape == (2, 2)
+
+
+def test_collectives():
+    _run_workers(_body_collectives)
+
+
+def test_sharded_aggregation_parity():
+    _run_workers(_body_sharded_ops)
+
+
+def test_rccl_parameter_server():
+    _run_workers(_body_rccl_ps)
+
+
+def test_rccl_peer_to_peer():
+    _run_workers(_body_rccl_p2p)
