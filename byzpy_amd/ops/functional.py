@@ -386,7 +386,11 @@ def nnm(X: torch.Tensor, f: int) -> torch.Tensor:
     D2 = pairwise_sq_dists(X)
     idx = torch.topk(D2, k=k, dim=1, largest=False).indices  # (n, k)
     Xf = X.float()
-    out = Xf[idx].mean(dim=1)
+    # mixing as a 0/1-mask matmul (one GEMM) rather than a gather that
+    # materializes an (n, k, d) tensor
+    mask = torch.zeros((n, n), device=X.device, dtype=torch.float32)
+    mask.scatter_(1, idx, 1.0)
+    out = (mask @ Xf) / k
     return out.to(X.dtype)
 
 

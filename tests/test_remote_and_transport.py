@@ -1,0 +1,136 @@
+"""Remote node server/client, mesh contexts, and prototype transports
+(SURVEY.md §4 patterns 6c/7) — all over loopback TCP."""
+import asyncio
+
+import pytest
+
+from byzpy_amd.engine.node.decentralized import DecentralizedNode
+from byzpy_amd.engine.node.remote import (
+    MeshRemoteContext,
+    RemoteContext,
+    RemoteNodeServer,
+)
+from byzpy_amd.engine.node_runner import NodeCluster, NodeRunner
+from byzpy_amd.engine.transport import LocalTransport, TcpTransport
+
+
+def test_remote_server_hosted_nodes():
+    async def main():
+        server = RemoteNodeServer()
+        a = DecentralizedNode("a")
+        b = DecentralizedNode("b")
+        got = []
+        b.register_handler("ping", lambda m: got.append(m["v"]))
+        server.host_node(a)
+        server.host_node(b)
+        await server.start()
+        try:
+            await a.router.route_direct("b", {"type": "ping", "sender": "a", "v": 1})
+            await asyncio.sleep(0.05)
+            assert got == [1]
+        finally:
+            await server.stop()
+
+    asyncio.run(main())
+
+
+def test_remote_client_to_hosted_node():
+    async def main():
+        server = RemoteNodeServer()
+        hosted = DecentralizedNode("hosted")
+        got = []
+        hosted.register_handler("hello", lambda m: got.append(m["v"]))
+        server.host_node(hosted)
+        await server.start()
+        try:
+            remote = DecentralizedNode(
+                "client-node",
+                context=RemoteContext("client-node", "127.0.0.1", server.port),
+            )
+            await remote.start()
+            await remote.context.send_message(
+                "hosted", {"type": "hello", "sender": "client-node", "v": 42}
+            )
+            await asyncio.sleep(0.1)
+            assert got == [42]
+            # reply path: hosted -> client
+            got2 = []
+            remote.register_handler("reply", lambda m: got2.append(m["v"]))
+            await hosted.context.send_message(
+                "client-node", {"type": "reply", "sender": "hosted", "v": 7}
+            )
+            await asyncio.sleep(0.3)
+            assert got2 == [7]
+            await remote.stop()
+        finally:
+            await server.stop()
+
+    asyncio.run(main())
+
+
+def test_mesh_context_two_nodes():
+    async def main():
+        n1 = DecentralizedNode("m1", context=MeshRemoteContext("m1"))
+        n2 = DecentralizedNode("m2", context=MeshRemoteContext("m2"))
+        got = []
+        n2.register_handler("g", lambda m: got.append(m["v"]))
+        await n1.start()
+        await n2.start()
+        # wire peers after ports are known
+        n1.context.add_peer("m2", "127.0.0.1", n2.context.port)
+        n2.context.add_peer("m1", "127.0.0.1", n1.context.port)
+        await n1.context.send_message("m2", {"type": "g", "sender": "m1", "v": 5})
+        await asyncio.sleep(0.1)
+        assert got == [5]
+        await n1.stop()
+        await n2.stop()
+
+    asyncio.run(main())
+
+
+def test_local_transport_runner_cluster():
+    async def main():
+        inbox = []
+        r1 = NodeRunner("r1", step_fn=lambda i: i * 2)
+        r2 = NodeRunner("r2", step_fn=lambda i: i, on_msg=inbox.append)
+        cluster = NodeCluster()
+        cluster.add(r1)
+        cluster.add(r2)
+        await cluster.start_all()
+        out = cluster.step_all()
+        assert out == {"r1": 0, "r2": 0}
+        assert r1.step() == 2
+        await r1.send("r2", {"x": 1})
+        assert inbox == [{"x": 1}]
+        await cluster.stop_all()
+
+    asyncio.run(main())
+
+
+def test_tcp_transport_loopback():
+    async def main():
+        got = []
+        t1 = TcpTransport("t1")
+        t2 = TcpTransport("t2")
+        t2.on_message(got.append)
+        await t1.start()
+        await t2.start()
+        t1.add_peer("t2", "127.0.0.1", t2.port)
+        await t1.send("t2", {"payload": [1, 2, 3]})
+        await asyncio.sleep(0.1)
+        assert got == [{"payload": [1, 2, 3]}]
+        await t1.stop()
+        await t2.stop()
+
+    asyncio.run(main())
+
+
+def test_auto_step_runner():
+    async def main():
+        r = NodeRunner("auto", step_fn=lambda i: i, auto_step_interval=0.02)
+        await r.start()
+        await asyncio.sleep(0.1)
+        await r.stop()
+        assert r.steps_done >= 2
+
+    asyncio.run(main())
