@@ -5,6 +5,11 @@
 #include <hip/hip_bf16.h>
 #include <c10/hip/HIPStream.h>
 
+#include <algorithm>
+#include <functional>
+#include <limits>
+#include <vector>
+
 // launchers (colsel.hip, rowops.hip, gram.hip)
 void launch_colsel_f32(const float*, float*, int, long, int, int, hipStream_t);
 void launch_colsel_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, long, int,
@@ -297,6 +302,86 @@ torch::Tensor cc_iter(torch::Tensor X, torch::Tensor v, double c_tau,
   return v_new;
 }
 
+// Exact minimum-diameter (n-f)-subset search (MDA) — HOST code: the n x n
+// squared-distance matrix is tiny; the combinatorial DFS belongs in native
+// C++, not Python (reference ran a seeded-DFS process pool for this,
+// minimum_diameter_average.py:267-386). Branch-and-bound over candidates
+// ordered by greedy insertion cost, seeded with a greedy upper bound.
+torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
+  TORCH_CHECK(!D2in.is_cuda(), "mda_search runs on the host copy of D2");
+  auto D2 = D2in.to(torch::kFloat64).contiguous();
+  const int n = (int)D2.size(0);
+  const int m = n - (int)f;
+  TORCH_CHECK(m >= 1 && D2.size(1) == n);
+  auto D = D2.accessor<double, 2>();
+
+  // greedy seed: for each anchor row, grow the subset by min-added-diameter
+  double best_diam = std::numeric_limits<double>::infinity();
+  std::vector<int> best;
+  std::vector<int> cur;
+  std::vector<double> maxd(n);
+  for (int a = 0; a < n; ++a) {
+    cur.assign(1, a);
+    std::fill(maxd.begin(), maxd.end(), 0.0);
+    std::vector<char> used(n, 0);
+    used[a] = 1;
+    for (int j = 0; j < n; ++j) maxd[j] = D[a][j];
+    double diam = 0.0;
+    for (int step = 1; step < m; ++step) {
+      int pick = -1;
+      double pick_cost = std::numeric_limits<double>::infinity();
+      for (int j = 0; j < n; ++j)
+        if (!used[j] && maxd[j] < pick_cost) { pick_cost = maxd[j]; pick = j; }
+      used[pick] = 1;
+      cur.push_back(pick);
+      diam = std::max(diam, pick_cost);
+      for (int j = 0; j < n; ++j) maxd[j] = std::max(maxd[j], D[pick][j]);
+    }
+    if (diam < best_diam) { best_diam = diam; best = cur; }
+  }
+
+  // exact DFS with prefix-max pruning (candidates in greedy-friendly order)
+  std::vector<int> order(n);
+  for (int i = 0; i < n; ++i) order[i] = i;
+  // order by row-sum of distances: central rows first
+  std::vector<double> rowsum(n, 0.0);
+  for (int i = 0; i < n; ++i)
+    for (int j = 0; j < n; ++j) rowsum[i] += D[i][j];
+  std::sort(order.begin(), order.end(),
+            [&](int a, int b) { return rowsum[a] < rowsum[b]; });
+
+  std::vector<int> chosen;
+  chosen.reserve(m);
+  std::function<void(int, double)> dfs = [&](int start, double diam) {
+    if ((int)chosen.size() == m) {
+      if (diam < best_diam) { best_diam = diam; best = chosen; }
+      return;
+    }
+    const int need = m - (int)chosen.size();
+    for (int oi = start; oi <= n - need; ++oi) {
+      const int j = order[oi];
+      double dj = diam;
+      bool ok = true;
+      for (int c : chosen) {
+        const double dc = D[c][j];
+        if (dc >= best_diam) { ok = false; break; }
+        if (dc > dj) dj = dc;
+      }
+      if (!ok || dj >= best_diam) continue;
+      chosen.push_back(j);
+      dfs(oi + 1, dj);
+      chosen.pop_back();
+    }
+  };
+  dfs(0, 0.0);
+
+  std::sort(best.begin(), best.end());
+  auto out = torch::empty({(long)best.size()}, torch::kInt64);
+  auto acc = out.accessor<int64_t, 1>();
+  for (size_t i = 0; i < best.size(); ++i) acc[i] = best[i];
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -312,4 +397,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("weiszfeld_apply", &weiszfeld_apply);
   m.def("cc_iter", &cc_iter);
   m.def("cc_apply", &cc_apply);
+  m.def("mda_search", &mda_search, "exact min-diameter subset (host DFS)");
 }

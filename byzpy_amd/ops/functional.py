@@ -154,6 +154,12 @@ def mda_subset(D2: torch.Tensor, f: int) -> Tuple[int, ...]:
     m = n - f
     if m < 1:
         raise ValueError(f"need n - f >= 1, got n={n}, f={f}")
+    from byzpy_amd import hip as _hip
+
+    ext = _hip.extension()
+    if ext is not None:  # native C++ branch-and-bound (host)
+        idx = ext.mda_search(D2.detach().float().cpu(), int(f))
+        return tuple(int(i) for i in idx)
     D = D2.detach().cpu().double().numpy()
     best_diam = math.inf
     best: Tuple[int, ...] = tuple(range(m))
@@ -298,15 +304,18 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     w = torch.ones(n, device=Xf.device)
     gen = torch.Generator(device="cpu")
     gen.manual_seed(0)
-    best_lambda = math.inf
+    # pregenerate every round's power-iteration init: ONE host->device copy
+    # instead of one per round (the seeded CPU rng is part of the contract)
+    V0 = torch.randn(n, Xf.shape[1], generator=gen).to(Xf.device)
+    best_lambda = torch.full((), math.inf, device=Xf.device)
     best_mu = Xf.mean(dim=0)
     target = float(n - 2 * f)
-    for _ in range(n):  # at most n rounds of downweighting
+    for r in range(n):  # at most n rounds of downweighting
         wsum = w.sum()
         mu = (w[:, None] * Xf).sum(dim=0) / wsum
         diffs = Xf - mu[None, :]
         # dominant eigenpair of (1/wsum) * diffs^T W diffs via power iteration
-        v = torch.randn(Xf.shape[1], generator=gen).to(Xf.device)
+        v = V0[r]
         v = v / v.norm().clamp_min(1e-20)
         lam = torch.zeros((), device=Xf.device)
         for _ in range(max(1, power_iters)):
@@ -315,18 +324,19 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
             t = t[0] / wsum
             lam = t.norm()
             v = t / lam.clamp_min(1e-20)
-        if float(lam) < best_lambda:
-            best_lambda = float(lam)
-            best_mu = mu
-        if float(wsum) <= target:
-            break
+        # device-side best tracking (no per-round host sync)
+        better = lam < best_lambda
+        best_lambda = torch.where(better, lam, best_lambda)
+        best_mu = torch.where(better, mu, best_mu)
         # downweight along v proportionally to projection^2
         proj = (diffs @ v) ** 2
         pmax = proj.max().clamp_min(1e-20)
-        w = w * (1.0 - proj / pmax)
-        w = w.clamp_min(0.0)
-        if float(w.sum()) <= 0:
+        w_next = (w * (1.0 - proj / pmax)).clamp_min(0.0)
+        # one host sync per round for the two break conditions
+        wsum_f, wnext_f = torch.stack([wsum, w_next.sum()]).tolist()
+        if wsum_f <= target or wnext_f <= 0:
             break
+        w = w_next
     return best_mu.to(X.dtype)
 
 
