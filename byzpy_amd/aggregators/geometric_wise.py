@@ -214,7 +214,8 @@ class MinimumDiameterAveraging(Aggregator):
         X, like, handles = self._pending
         self._pending = None
         try:
-            best_diam, best = min(results, key=lambda r: r[0])
+            # min diameter, then lexicographic subset (canonical tie-break)
+            best_diam, best = min(results, key=lambda r: (r[0], r[1]))
             idx = torch.tensor(best, dtype=torch.long)
             out = X.float()[idx].mean(dim=0).to(X.dtype)
             return to_like(out, like)

@@ -375,6 +375,31 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
   };
   dfs(0, 0.0);
 
+  // Canonicalize: among all subsets achieving best_diam, return the
+  // lexicographically smallest (optimal subsets are generically non-unique
+  // — every subset containing the binding edge ties). Pure index-order DFS
+  // with the now-tight bound; the first complete subset found is the
+  // lexicographic minimum.
+  const double bound = best_diam;
+  std::vector<int> lex;
+  lex.reserve(m);
+  bool found = false;
+  std::function<void(int)> dfs2 = [&](int start) {
+    if (found) return;
+    if ((int)lex.size() == m) { best = lex; found = true; return; }
+    const int need = m - (int)lex.size();
+    for (int j = start; j <= n - need && !found; ++j) {
+      bool ok = true;
+      for (int c : lex)
+        if (D[c][j] > bound) { ok = false; break; }
+      if (!ok) continue;
+      lex.push_back(j);
+      dfs2(j + 1);
+      lex.pop_back();
+    }
+  };
+  dfs2(0);
+
   std::sort(best.begin(), best.end());
   auto out = torch::empty({(long)best.size()}, torch::kInt64);
   auto acc = out.accessor<int64_t, 1>();

@@ -195,6 +195,26 @@ def mda_subset(D2: torch.Tensor, f: int) -> Tuple[int, ...]:
                 chosen.pop()
 
     dfs(0, [], 0.0)
+
+    # canonicalize to the lexicographically smallest optimal subset (ties
+    # are generic: every subset containing the binding edge can tie)
+    bound = best_diam
+    lex: List[int] = []
+
+    def dfs2(start: int) -> bool:
+        if len(lex) == m:
+            return True
+        need = m - len(lex)
+        for j in range(start, n - need + 1):
+            if all(D[c][j] <= bound for c in lex):
+                lex.append(j)
+                if dfs2(j + 1):
+                    return True
+                lex.pop()
+        return False
+
+    if dfs2(0):
+        return tuple(lex)
     return best
 
 
