@@ -324,9 +324,6 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     w = torch.ones(n, device=Xf.device)
     gen = torch.Generator(device="cpu")
     gen.manual_seed(0)
-    # pregenerate every round's power-iteration init: ONE host->device copy
-    # instead of one per round (the seeded CPU rng is part of the contract)
-    V0 = torch.randn(n, Xf.shape[1], generator=gen).to(Xf.device)
     best_lambda = torch.full((), math.inf, device=Xf.device)
     best_mu = Xf.mean(dim=0)
     target = float(n - 2 * f)
@@ -335,7 +332,8 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
         mu = (w[:, None] * Xf).sum(dim=0) / wsum
         diffs = Xf - mu[None, :]
         # dominant eigenpair of (1/wsum) * diffs^T W diffs via power iteration
-        v = V0[r]
+        # (seeded CPU rng per round is part of the determinism contract)
+        v = torch.randn(Xf.shape[1], generator=gen).to(Xf.device)
         v = v / v.norm().clamp_min(1e-20)
         lam = torch.zeros((), device=Xf.device)
         for _ in range(max(1, power_iters)):

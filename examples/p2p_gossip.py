@@ -5,6 +5,10 @@
 from __future__ import annotations
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import asyncio
 
 import torch
