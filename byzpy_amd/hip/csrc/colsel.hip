@@ -140,6 +140,111 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
+// Packed bf16 median (n <= 64, even d): TWO adjacent columns per thread.
+// Order statistics only need a monotone key, so bf16 values become
+// sortable u16 keys (sign-flip transform) and the bitonic network runs on
+// v_pk_min_u16 / v_pk_max_u16 — one instruction per compare-exchange for
+// BOTH columns (no packed f32 min/max exists on gfx950), with half the
+// register footprint (P u32 regs for 2 columns) and 4 B/lane loads.
+// ---------------------------------------------------------------------------
+
+typedef unsigned int u32;
+
+DEV u32 pk_min_u16(u32 a, u32 b) {
+  u32 r;
+  asm("v_pk_min_u16 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+  return r;
+}
+DEV u32 pk_max_u16(u32 a, u32 b) {
+  u32 r;
+  asm("v_pk_max_u16 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+  return r;
+}
+
+// bf16 bits -> sortable u16 key, two lanes at once (32-bit arithmetic:
+// per half key = bits ^ (sign ? 0xFFFF : 0x8000); the multiply cannot
+// carry across halves)
+DEV u32 pk_key_from_bf16(u32 bits) {
+  const u32 s = (bits >> 15) & 0x00010001u;
+  return bits ^ (0x80008000u + s * 0x7FFFu);
+}
+
+DEV float key_to_float(u32 key16) {
+  unsigned short bits =
+      (key16 & 0x8000u) ? (unsigned short)(key16 ^ 0x8000u)
+                        : (unsigned short)(~key16 & 0xFFFFu);
+  union { unsigned short s; __hip_bfloat16 h; } c;
+  c.s = bits;
+  return __bfloat162float(c.h);
+}
+
+template <int P>
+DEV void bitonic_sort_pk(u32 (&v)[P]) {
+#pragma unroll
+  for (int k = 2; k <= P; k <<= 1) {
+#pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = (i & k) == 0;
+          const u32 a = v[i], b = v[l];
+          const u32 lo = pk_min_u16(a, b), hi = pk_max_u16(a, b);
+          v[i] = asc ? lo : hi;
+          v[l] = asc ? hi : lo;
+        }
+      }
+    }
+  }
+}
+
+template <int P>
+DEV u32 extract_at_pk(const u32 (&v)[P], int pos) {
+  u32 r = 0;
+#pragma unroll
+  for (int i = 0; i < P; ++i)
+    if (i == pos) r = v[i];
+  return r;
+}
+
+template <int P>
+__global__ void colsel_pk_median_bf16(const unsigned short* __restrict__ X,
+                                      unsigned short* __restrict__ out, int n,
+                                      long d, int f) {
+  const long npairs = d >> 1;
+  const long pair0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long rowstride = d >> 1;  // in u32 units
+  for (long pair = pair0; pair < npairs; pair += stride) {
+    u32 v[P];
+    const u32* p = reinterpret_cast<const u32*>(X) + pair;
+#pragma unroll
+    for (int i = 0; i < P; ++i) {
+      v[i] = pk_key_from_bf16(*p);
+      if (i + 1 < n) p += rowstride;
+    }
+#pragma unroll
+    for (int i = 0; i < P; ++i)
+      if (i >= n) v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
+    bitonic_sort_pk<P>(v);
+    const u32 lo = extract_at_pk<P>(v, (n - 1) >> 1);
+    const u32 hi = extract_at_pk<P>(v, n >> 1);
+    const float m0 =
+        0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
+    const float m1 =
+        0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
+    union { unsigned short s[2]; u32 w; } o;
+    union { unsigned short s; __hip_bfloat16 h; } c0, c1;
+    c0.h = __float2bfloat16(m0);
+    c1.h = __float2bfloat16(m1);
+    o.s[0] = c0.s;
+    o.s[1] = c1.s;
+    reinterpret_cast<u32*>(out)[pair] = o.w;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // LDS variant, 64 < n <= 512; one wave per block, lane-owned columns
 // ---------------------------------------------------------------------------
 
@@ -248,5 +353,26 @@ void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,
 
 void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
+  if (mode == MEDIAN && n <= 64 && (d % 2) == 0) {
+    const int block = 256;
+    const long pairs = d >> 1;
+    const long want = (pairs + block - 1) / block;
+    const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
+    const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
+    unsigned short* Ou = reinterpret_cast<unsigned short*>(out);
+    if (n <= 8)
+      hipLaunchKernelGGL((colsel_pk_median_bf16<8>), dim3(grid), dim3(block),
+                         0, stream, Xu, Ou, n, d, f);
+    else if (n <= 16)
+      hipLaunchKernelGGL((colsel_pk_median_bf16<16>), dim3(grid), dim3(block),
+                         0, stream, Xu, Ou, n, d, f);
+    else if (n <= 32)
+      hipLaunchKernelGGL((colsel_pk_median_bf16<32>), dim3(grid), dim3(block),
+                         0, stream, Xu, Ou, n, d, f);
+    else
+      hipLaunchKernelGGL((colsel_pk_median_bf16<64>), dim3(grid), dim3(block),
+                         0, stream, Xu, Ou, n, d, f);
+    return;
+  }
   launch_colsel_typed<__hip_bfloat16>(X, out, n, d, mode, f, stream);
 }
