@@ -20,6 +20,17 @@ namespace {
 
 constexpr float PAD = 3.0e38f;  // sorts after every real value
 
+// Copy a wave-uniform int into a VGPR. Unrolled predicates like
+// `i == pos` / `i >= n` (i compile-time, pos/n uniform) otherwise become
+// batched s_cmp/s_cselect_b64 SGPR mask pairs — ~64 live masks spill
+// through v_writelane/readlane and dominate the kernel. Against a VGPR
+// operand they lower to v_cmp/v_cndmask instead (2 VALU each, no spills).
+DEV int vecify(int x) {
+  int r;
+  asm("v_mov_b32_e32 %0, %1" : "=v"(r) : "s"(x));
+  return r;
+}
+
 enum Mode { MEDIAN = 0, TRIMMED = 1, MEAMED = 2 };
 
 // ---------------------------------------------------------------------------
@@ -93,46 +104,58 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long col = col0; col < d; col += stride) {
     float v[P];
-    // Walk a row pointer that stops advancing at the last valid row (ONE
-    // live address pair — batching P strided addresses spills ~P VGPR
-    // pairs), then pad with selects. No branch around any load (guide §5
-    // ".s-level traps" (c)).
+    // Load phase. Two rules keep the address walk at ONE live base:
+    // (1) raw loads only — any data-dependent convert inside the walk
+    //     would force a wait per element; conversion happens after;
+    // (2) sched_barrier(0) pins the load/advance alternation — without it
+    //     the compiler batches all P addresses into SGPR pairs and spills
+    //     them through v_writelane/readlane (~1100 VALU per column).
+    // Once a load ISSUES its address registers are free; the loads stay
+    // in flight and the converts below wait on counted vmcnt.
     {
+      T raw[P];
       const T* p = X + col;
 #pragma unroll
       for (int i = 0; i < P; ++i) {
-        v[i] = to_f<T>(*p);
+        raw[i] = *p;
         if (i + 1 < n) p += d;
+        __builtin_amdgcn_sched_barrier(0);
       }
 #pragma unroll
+      for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
+      const int nv = vecify(n);
+#pragma unroll
       for (int i = 0; i < P; ++i)
-        if (i >= n) v[i] = PAD;
+        if (i >= nv) v[i] = PAD;
     }
     bitonic_sort_reg<P>(v);  // in place: meamed only needs sorted order
 
-    const float med_lo = extract_at<P>(v, (n - 1) >> 1);
-    const float med_hi = extract_at<P>(v, n >> 1);
+    const int pos_lo = vecify((n - 1) >> 1), pos_hi = vecify(n >> 1);
+    const float med_lo = extract_at<P>(v, pos_lo);
+    const float med_hi = extract_at<P>(v, pos_hi);
     const float med = 0.5f * (med_lo + med_hi);
 
     float result;
     if (MODE == MEDIAN) {
       result = med;
     } else if (MODE == TRIMMED) {
+      const int fv = vecify(f), nfv = vecify(n - f);
       float s = 0.0f;
 #pragma unroll
       for (int i = 0; i < P; ++i)
-        if (i >= f && i < n - f) s += v[i];
+        if (i >= fv && i < nfv) s += v[i];
       result = s / (float)(n - 2 * f);
     } else {  // MEAMED: mean of the n-f values closest to the median
+      const int nv2 = vecify(n), nfv = vecify(n - f);
       float dev[P];
 #pragma unroll
       for (int i = 0; i < P; ++i)
-        dev[i] = (i < n) ? fabsf(v[i] - med) : PAD;
+        dev[i] = (i < nv2) ? fabsf(v[i] - med) : PAD;
       bitonic_sort_kv_reg<P>(dev, v);
       float s = 0.0f;
 #pragma unroll
       for (int i = 0; i < P; ++i)
-        if (i < n - f) s += v[i];
+        if (i < nfv) s += v[i];
       result = s / (float)(n - f);
     }
     out[col] = from_f<T>(result);
@@ -218,18 +241,24 @@ __global__ void colsel_pk_median_bf16(const unsigned short* __restrict__ X,
   const long rowstride = d >> 1;  // in u32 units
   for (long pair = pair0; pair < npairs; pair += stride) {
     u32 v[P];
+    // raw loads first + pinned alternation: see colsel_reg_kernel's load
+    // phase for why (one live base, no SGPR spill storm)
     const u32* p = reinterpret_cast<const u32*>(X) + pair;
 #pragma unroll
     for (int i = 0; i < P; ++i) {
-      v[i] = pk_key_from_bf16(*p);
+      v[i] = *p;
       if (i + 1 < n) p += rowstride;
+      __builtin_amdgcn_sched_barrier(0);
     }
 #pragma unroll
+    for (int i = 0; i < P; ++i) v[i] = pk_key_from_bf16(v[i]);
+    const int nv = vecify(n);
+#pragma unroll
     for (int i = 0; i < P; ++i)
-      if (i >= n) v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
+      if (i >= nv) v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
     bitonic_sort_pk<P>(v);
-    const u32 lo = extract_at_pk<P>(v, (n - 1) >> 1);
-    const u32 hi = extract_at_pk<P>(v, n >> 1);
+    const u32 lo = extract_at_pk<P>(v, vecify((n - 1) >> 1));
+    const u32 hi = extract_at_pk<P>(v, vecify(n >> 1));
     const float m0 =
         0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
     const float m1 =
