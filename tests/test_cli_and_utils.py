@@ -65,3 +65,42 @@ def test_checkpoint_sharded(tmp_path):
 def test_tracing_noop_on_cpu():
     with trace_range("test"):
         mark("point")  # must not raise regardless of roctx availability
+
+
+def test_backend_config_surface():
+    import pytest as _pytest
+
+    from byzpy_amd.configs.backend import get_backend, set_backend, use_backend
+
+    set_backend("torch")
+    with use_backend("torch"):
+        pass
+    import torch as _torch
+
+    assert get_backend() is _torch
+    with _pytest.raises(ValueError):
+        set_backend("numpy")
+
+
+def test_parameter_server_runner():
+    import asyncio
+
+    import torch
+
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+    from byzpy_amd.engine.parameter_server.runner import ParameterServerRunner
+
+    async def main():
+        applied = []
+        runner = ParameterServerRunner(
+            [lambda i, v=v: torch.full((4,), float(v)) for v in (1, 2, 3)],
+            CoordinateWiseMedian(),
+            apply_fns=[applied.append],
+        )
+        await runner.start()
+        out = await runner.round()
+        assert torch.allclose(out, torch.full((4,), 2.0))
+        assert len(applied) == 1
+        await runner.stop()
+
+    asyncio.run(main())
