@@ -37,6 +37,7 @@ void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
+void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
 
 namespace {
 
@@ -225,6 +226,21 @@ torch::Tensor weiszfeld_iter(torch::Tensor X, torch::Tensor z, double eps,
         cur_stream());
   }
   return z_new;
+}
+
+// Fused Krum selection: Gram -> q winner indices (one kernel, replaces
+// the D2-build + two topk torch chains whose launch overhead caps the
+// d-sharded multi-GPU scaling).
+torch::Tensor krum_select(torch::Tensor G, int64_t f, int64_t q) {
+  TORCH_CHECK(G.is_cuda() && G.dim() == 2 && G.size(0) == G.size(1) &&
+              G.scalar_type() == torch::kFloat32 && G.is_contiguous());
+  const int n = (int)G.size(0);
+  TORCH_CHECK(n <= 512, "krum_select supports n <= 512");
+  TORCH_CHECK(q >= 1 && q <= n && f >= 0 && n - f - 1 >= 1);
+  auto idx = torch::empty({q}, G.options().dtype(torch::kInt32));
+  launch_krum_select(G.data_ptr<float>(), n, (int)f, (int)q,
+                     idx.data_ptr<int>(), nullptr, cur_stream());
+  return idx;
 }
 
 // Sharded form: apply the Weiszfeld update with externally-reduced global
@@ -418,6 +434,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_mean_rows", &group_mean_rows);
   m.def("bucket_mean", &bucket_mean);
   m.def("gram", &gram);
+  m.def("krum_select", &krum_select);
   m.def("weiszfeld_iter", &weiszfeld_iter);
   m.def("weiszfeld_apply", &weiszfeld_apply);
   m.def("cc_iter", &cc_iter);

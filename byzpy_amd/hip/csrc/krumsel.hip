@@ -1,0 +1,105 @@
+// Fused Krum selection: Gram (n,n) -> q winner indices.
+//
+// One workgroup (16 waves). Phase 1: wave w handles rows w, w+16, ...;
+// for each row it materializes the squared-distance row
+// D2[i][j] = G[i][i] + G[j][j] - 2 G[i][j]  (diag -> +inf), sorts it in
+// its private LDS slab with a wave-synchronous bitonic (lanes of one wave
+// are lockstep; no block barrier needed inside a row), and sums the
+// n-f-1 smallest into scores[i]. Phase 2: one wave selects the q smallest
+// scores by repeated masked min. Replaces ~8 small torch launches of
+// <=0.1 ms each — fixed overhead that caps multi-GPU scaling of the
+// d-sharded Krum (the (n,n) work is rank-replicated).
+#include "common.h"
+
+namespace {
+
+constexpr int WAVES_K = 16;
+constexpr int MAX_N = 512;
+constexpr float INF = 3.0e38f;
+
+__global__ void __launch_bounds__(WAVES_K * 64)
+krum_select_kernel(const float* __restrict__ G, int n, int f, int q,
+                   int* __restrict__ out_idx, float* __restrict__ out_scores) {
+  __shared__ float rowbuf[WAVES_K][MAX_N];
+  __shared__ float scores[MAX_N];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  int P2 = 1;
+  while (P2 < n) P2 <<= 1;
+
+  const int k = n - f - 1;  // distances summed per row
+  for (int row = wave; row < n; row += WAVES_K) {
+    float* buf = rowbuf[wave];
+    const float gii = G[(long)row * n + row];
+    for (int j = lane; j < P2; j += 64) {
+      float v;
+      if (j >= n || j == row) {
+        v = INF;
+      } else {
+        const float d2 =
+            gii + G[(long)j * n + j] - 2.0f * G[(long)row * n + j];
+        v = fmaxf(d2, 0.0f);
+      }
+      buf[j] = v;
+    }
+    // wave-synchronous bitonic sort of buf[0..P2)
+    for (int kk = 2; kk <= P2; kk <<= 1) {
+      for (int jj = kk >> 1; jj > 0; jj >>= 1) {
+        for (int i = lane; i < P2; i += 64) {
+          const int l = i ^ jj;
+          if (l > i) {
+            const bool asc = (i & kk) == 0;
+            const float a = buf[i], b = buf[l];
+            const float lo = fminf(a, b), hi = fmaxf(a, b);
+            buf[i] = asc ? lo : hi;
+            buf[l] = asc ? hi : lo;
+          }
+        }
+        // lanes of one wave run in lockstep; LDS within the wave is
+        // ordered by program order — no barrier needed
+      }
+    }
+    float acc = 0.0f;
+    for (int i = lane; i < k; i += 64) acc += buf[i];
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) scores[row] = acc;
+  }
+  __syncthreads();
+
+  // phase 2: q smallest scores (first wave only), ties -> smallest index
+  if (wave == 0) {
+    for (int pick = 0; pick < q; ++pick) {
+      float best = INF;
+      int best_i = -1;
+      for (int i = lane; i < n; i += 64) {
+        const float s = scores[i];
+        if (s < best || (s == best && i < best_i)) { best = s; best_i = i; }
+      }
+      // wave reduce (value, index) preferring smaller value then index
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        const float ov = __shfl_down(best, off, 64);
+        const int oi = __shfl_down(best_i, off, 64);
+        if (ov < best || (ov == best && oi != -1 && (best_i == -1 || oi < best_i))) {
+          best = ov;
+          best_i = oi;
+        }
+      }
+      best_i = __shfl(best_i, 0, 64);
+      if (lane == 0) {
+        out_idx[pick] = best_i;
+        if (out_scores != nullptr) out_scores[pick] = best;
+        scores[best_i] = INF;  // mask out
+      }
+      // lane 0's LDS write is visible to the wave (lockstep)
+    }
+  }
+}
+
+}  // namespace
+
+void launch_krum_select(const float* G, int n, int f, int q, int* out_idx,
+                        float* out_scores, hipStream_t stream) {
+  hipLaunchKernelGGL(krum_select_kernel, dim3(1), dim3(WAVES_K * 64), 0,
+                     stream, G, n, f, q, out_idx, out_scores);
+}

@@ -88,13 +88,29 @@ def multi_krum_scores(X: torch.Tensor, f: int) -> torch.Tensor:
 
 def multi_krum(X: torch.Tensor, f: int, q: int) -> torch.Tensor:
     if _gpu(X):
-        scores = multi_krum_scores(X, f)
-        winners = torch.topk(scores, k=q, largest=False).indices
-        return mean_rows(X, winners)
+        n = X.shape[0]
+        G = gram(X)
+        if n <= 512:
+            idx = _hip.require().krum_select(G, int(f), int(q))
+        else:
+            scores = _scores_from_gram(G, n, f)
+            idx = torch.topk(scores, k=q, largest=False).indices.to(torch.int32)
+        return mean_rows(X, idx)
     return F.multi_krum(X, f, q)
 
 
+def _scores_from_gram(G: torch.Tensor, n: int, f: int) -> torch.Tensor:
+    norms = torch.diagonal(G)
+    D2 = (norms[:, None] + norms[None, :] - 2.0 * G).clamp_(min=0.0)
+    D2 = D2 + torch.diag(torch.full((n,), float("inf"), device=G.device, dtype=D2.dtype))
+    return torch.topk(D2, k=n - f - 1, dim=1, largest=False).values.sum(dim=1)
+
+
 def krum(X: torch.Tensor, f: int) -> torch.Tensor:
+    if _gpu(X) and X.shape[0] <= 512:
+        G = gram(X)
+        idx = _hip.require().krum_select(G, int(f), 1)
+        return X[idx.long()[0]].clone()
     scores = multi_krum_scores(X, f)
     return X[int(torch.argmin(scores))].clone()
 

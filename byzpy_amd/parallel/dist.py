@@ -76,6 +76,23 @@ def all_gather_rows(local: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def reduce_scatter_cat(full: torch.Tensor) -> torch.Tensor:
+    """Reduce-scatter a replicated-layout (world*m, ...) tensor: rank r gets
+    the SUM over ranks of row-block r (SURVEY.md C4 — coordinate-chunked
+    partial reduction; the xGMI-friendly half of an all-reduce)."""
+    if not is_initialized():
+        return full
+    world = get_world_size()
+    assert full.shape[0] % world == 0
+    out = torch.empty(
+        (full.shape[0] // world,) + tuple(full.shape[1:]),
+        dtype=full.dtype,
+        device=full.device,
+    )
+    dist.reduce_scatter_tensor(out, full.contiguous())
+    return out
+
+
 def all_to_all_rows(local: torch.Tensor) -> torch.Tensor:
     """Row-blocked all-to-all: rank r sends row-block j to rank j and
     receives block r from everyone (n-sharded -> d-sharded transposition,

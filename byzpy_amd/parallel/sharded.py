@@ -45,6 +45,11 @@ def _global_gram(X_local: torch.Tensor) -> torch.Tensor:
 def multi_krum(X_local: torch.Tensor, f: int, q: int) -> torch.Tensor:
     n = X_local.shape[0]
     G = _global_gram(X_local)
+    if X_local.is_cuda and n <= 512:
+        from byzpy_amd.hip import require
+
+        winners = require().krum_select(G, int(f), int(q))
+        return D.mean_rows(X_local, winners)
     norms = torch.diagonal(G)
     D2 = (norms[:, None] + norms[None, :] - 2.0 * G).clamp_(min=0.0)
     D2 = D2 + torch.diag(torch.full((n,), float("inf"), device=X_local.device))

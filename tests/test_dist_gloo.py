@@ -130,6 +130,12 @@ def _body_collectives(rank):
     assert g.shape == (2, 4)
     a2a = pdist.all_to_all_rows(torch.arange(4.0).reshape(2, 2) + 10 * rank)
     assert a2a.shape == (2, 2)
+    # reduce-scatter: rank r gets sum over ranks of row-block r
+    full = torch.arange(8.0).reshape(4, 2) * (rank + 1)
+    rs = pdist.reduce_scatter_cat(full)
+    lo = rank * 2
+    expect = (torch.arange(8.0).reshape(4, 2) * 3)[lo : lo + 2]
+    assert torch.allclose(rs, expect), rs
 
 
 def test_collectives():

@@ -190,3 +190,19 @@ def test_aggregator_classes_on_gpu():
     assert out.is_cuda and out.dtype == torch.bfloat16
     out2 = MultiKrum(2, 3).aggregate(list(X))
     assert out2.is_cuda and out2.shape == (513,)
+
+
+@pytest.mark.parametrize("n,f,q", [(16, 3, 4), (64, 16, 12), (100, 20, 8)])
+def test_krum_select_kernel(n, f, q):
+    from byzpy_amd.hip import require
+
+    X = _rand(n, 2048, seed=21)
+    G = D.gram(X)
+    idx = require().krum_select(G, f, q)
+    # reference: torch scores path on the same Gram
+    norms = torch.diagonal(G)
+    D2 = (norms[:, None] + norms[None, :] - 2.0 * G).clamp_(min=0.0)
+    D2 = D2 + torch.diag(torch.full((n,), float("inf"), device=G.device))
+    scores = torch.topk(D2, k=n - f - 1, dim=1, largest=False).values.sum(dim=1)
+    ref = torch.topk(scores, k=q, largest=False).indices
+    assert set(idx.cpu().tolist()) == set(ref.cpu().tolist())
