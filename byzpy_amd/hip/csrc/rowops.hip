@@ -82,7 +82,27 @@ __global__ void row_red_kernel(const T* __restrict__ X,
     const long dv = d / V;
     const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
     const long stride = (long)gridDim.x * blockDim.x;
-    for (long jv = start; jv < dv; jv += stride) {
+    long jv = start;
+    // 4 grid-stride positions per step: 4 loads in flight
+    for (; jv + 3 * stride < dv; jv += 4 * stride) {
+      float x0[V], x1[V], x2[V], x3[V];
+      VecTraits<T>::load(xr + jv * V, x0);
+      VecTraits<T>::load(xr + (jv + stride) * V, x1);
+      VecTraits<T>::load(xr + (jv + 2 * stride) * V, x2);
+      VecTraits<T>::load(xr + (jv + 3 * stride) * V, x3);
+#pragma unroll
+      for (int c = 0; c < V; ++c) {
+        float v0 = x0[c], v1 = x1[c], v2 = x2[c], v3 = x3[c];
+        if (CENTER) {
+          v0 -= z[jv * V + c];
+          v1 -= z[(jv + stride) * V + c];
+          v2 -= z[(jv + 2 * stride) * V + c];
+          v3 -= z[(jv + 3 * stride) * V + c];
+        }
+        acc += (v0 * v0 + v1 * v1) + (v2 * v2 + v3 * v3);
+      }
+    }
+    for (; jv < dv; jv += stride) {
       float x[V];
       VecTraits<T>::load(xr + jv * V, x);
 #pragma unroll
@@ -152,7 +172,19 @@ __global__ void gather_mean_kernel(const T* __restrict__ X,
     const long dv = d / V;
     for (long jv = start; jv < dv; jv += stride) {
       float acc[V] = {0};
-      for (int i = 0; i < k; ++i) {
+      // 4 independent row loads in flight per step (a single runtime-n
+      // loop leaves one load outstanding and goes latency-bound)
+      int i = 0;
+      for (; i + 4 <= k; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)gi[i + 0] * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)gi[i + 1] * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)gi[i + 2] * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)gi[i + 3] * d + jv * V, x3);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += (x0[c] + x1[c]) + (x2[c] + x3[c]);
+      }
+      for (; i < k; ++i) {
         float x[V];
         VecTraits<T>::load(X + (long)gi[i] * d + jv * V, x);
 #pragma unroll
@@ -187,7 +219,17 @@ __global__ void bucket_mean_kernel(const T* __restrict__ X,
     const long dv = d / V;
     for (long jv = start; jv < dv; jv += stride) {
       float acc[V] = {0};
-      for (int i = lo; i < hi; ++i) {
+      int i = lo;
+      for (; i + 4 <= hi; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)perm[i + 0] * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)perm[i + 1] * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)perm[i + 2] * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)perm[i + 3] * d + jv * V, x3);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += (x0[c] + x1[c]) + (x2[c] + x3[c]);
+      }
+      for (; i < hi; ++i) {
         float x[V];
         VecTraits<T>::load(X + (long)perm[i] * d + jv * V, x);
 #pragma unroll
@@ -234,7 +276,20 @@ __global__ void weiszfeld_update_kernel(const T* __restrict__ X,
     const long dv = d / V;
     for (long jv = start; jv < dv; jv += stride) {
       float num[V] = {0};
-      for (int i = 0; i < n; ++i) {
+      int i = 0;
+      for (; i + 4 <= n; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)(i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)(i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)(i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)(i + 3) * d + jv * V, x3);
+        const float w0 = w_lds[i], w1 = w_lds[i + 1], w2 = w_lds[i + 2],
+                    w3 = w_lds[i + 3];
+#pragma unroll
+        for (int c = 0; c < V; ++c)
+          num[c] += (w0 * x0[c] + w1 * x1[c]) + (w2 * x2[c] + w3 * x3[c]);
+      }
+      for (; i < n; ++i) {
         const float w = w_lds[i];
         float x[V];
         VecTraits<T>::load(X + (long)i * d + jv * V, x);
@@ -285,7 +340,20 @@ __global__ void cc_update_kernel(const T* __restrict__ X,
     const long dv = d / V;
     for (long jv = start; jv < dv; jv += stride) {
       float acc[V] = {0};
-      for (int i = 0; i < n; ++i) {
+      int i = 0;
+      for (; i + 4 <= n; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)(i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)(i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)(i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)(i + 3) * d + jv * V, x3);
+        const float a0 = a_lds[i], a1 = a_lds[i + 1], a2 = a_lds[i + 2],
+                    a3 = a_lds[i + 3];
+#pragma unroll
+        for (int c = 0; c < V; ++c)
+          acc[c] += (a0 * x0[c] + a1 * x1[c]) + (a2 * x2[c] + a3 * x3[c]);
+      }
+      for (; i < n; ++i) {
         const float a = a_lds[i];
         float x[V];
         VecTraits<T>::load(X + (long)i * d + jv * V, x);
