@@ -42,15 +42,15 @@ Workload = Tuple[str, int, int, Callable[[torch.Tensor], Any], Optional[float]]
 
 
 def _agg(op):
-    return lambda X: op.aggregate(X)
+    return ("agg", op, lambda X: op.aggregate(X))
 
 
 def _pre(op):
-    return lambda X: op.pre_aggregate(X)
+    return ("pre", op, lambda X: op.pre_aggregate(X))
 
 
 def _atk(op):
-    return lambda X: op.apply(honest_grads=X)
+    return ("atk", op, lambda X: op.apply(honest_grads=X))
 
 
 def workloads() -> List[Workload]:
@@ -76,6 +76,33 @@ def workloads() -> List[Workload]:
     ]
 
 
+def bench_one_pooled(op, kind, X, warmup, repeat, pool_backend, pool_n):
+    """Reference-style ActorPool xN timing (CPU plumbing path)."""
+    import asyncio
+
+    from byzpy_amd.graph.executor import OperatorExecutor
+    from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
+
+    key = {"agg": "gradients", "pre": "vectors", "atk": "honest_grads"}[kind]
+
+    async def go():
+        pool = ActorPool(ActorPoolConfig(backend=pool_backend, count=pool_n))
+        await pool.start()
+        ex = OperatorExecutor(op, pool=pool)
+        inputs = {key: list(X)}
+        for _ in range(warmup):
+            await ex.run(inputs)
+        best = float("inf")
+        for _ in range(repeat):
+            t0 = time.perf_counter()
+            await ex.run(inputs)
+            best = min(best, time.perf_counter() - t0)
+        await pool.close()
+        return best * 1000.0
+
+    return asyncio.run(go())
+
+
 def bench_one(run, X, warmup: int, repeat: int, device) -> float:
     for _ in range(warmup):
         run(X)
@@ -99,18 +126,24 @@ def main() -> None:
     p.add_argument("--dtype", default="f32", choices=["f32", "bf16"])
     p.add_argument("--out", default=None)
     p.add_argument("--only", default=None, help="substring filter")
+    p.add_argument("--pool", type=int, default=0, help="ActorPool worker count (0 = direct)")
+    p.add_argument("--pool-backend", default="process")
     args = p.parse_args()
     device = torch.device(args.device)
     dtype = torch.float32 if args.dtype == "f32" else torch.bfloat16
 
     rows = []
-    for name, n, d, run, ref_ms in workloads():
+    for name, n, d, (kind, op, run), ref_ms in workloads():
         if args.only and args.only.lower() not in name.lower():
             continue
         g = torch.Generator().manual_seed(0)
         X = torch.randn(n, d, generator=g).to(device, dtype)
         try:
-            ms = bench_one(run, X, args.warmup, args.repeat, device)
+            if args.pool > 0:
+                ms = bench_one_pooled(op, kind, X, args.warmup, args.repeat,
+                                      args.pool_backend, args.pool)
+            else:
+                ms = bench_one(run, X, args.warmup, args.repeat, device)
             speed = f"{ref_ms / ms:.1f}x" if ref_ms else "-"
             rows.append((name, ms, ref_ms, speed))
             print(f"{name:42s} {ms:9.3f} ms   ref {ref_ms or '-':>8} ms   {speed}")

@@ -31,10 +31,18 @@ def _pickle_fn(fn: Any) -> bytes:
 
 
 class _SubTaskWorker:
-    """Constructed ON the actor backend; executes shipped subtask fns."""
+    """Constructed ON the actor backend; executes shipped subtask fns.
 
-    def __init__(self) -> None:
+    ``torch_threads``: out-of-process workers cap their intra-op thread
+    count so k worker processes don't oversubscribe the host (reference
+    krum.py:425-475 did the same inside its chunk fns)."""
+
+    def __init__(self, torch_threads: int = 0) -> None:
         self._fn_cache: Dict[bytes, Any] = {}
+        if torch_threads > 0:
+            import torch
+
+            torch.set_num_threads(torch_threads)
 
     def execute(self, fn_blob: bytes, args: tuple, kwargs: dict) -> Any:
         fn = self._fn_cache.get(fn_blob)
@@ -78,10 +86,11 @@ class _PoolWorker:
         self.label = label
         self.ref = ActorRef(backend)
         self._in_process = getattr(backend, "scheme", "") in ("thread", "stream")
+        self.torch_threads = 0  # set by the pool for out-of-process workers
 
     async def start(self) -> None:
         await self.backend.start()
-        await self.backend.construct(_SubTaskWorker)
+        await self.backend.construct(_SubTaskWorker, self.torch_threads)
 
     async def run(self, subtask: SubTask) -> Any:
         if self._in_process:
@@ -136,6 +145,14 @@ class ActorPool:
                     cfg.backend if isinstance(cfg.backend, str) else cfg.backend
                 )
                 self.workers.append(_PoolWorker(backend, caps, label))
+        import os
+
+        n_out = sum(1 for w in self.workers if not w._in_process)
+        if n_out:
+            per = max(1, (os.cpu_count() or 4) // n_out)
+            for w in self.workers:
+                if not w._in_process:
+                    w.torch_threads = per
         await asyncio.gather(*(w.start() for w in self.workers))
         self._idle.extend(range(len(self.workers)))
         self._started = True
