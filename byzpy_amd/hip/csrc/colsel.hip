@@ -18,7 +18,10 @@
 
 namespace {
 
-constexpr float PAD = 3.0e38f;  // sorts after every real value
+// +inf (not a big-finite sentinel): adversarial rows can BE +inf
+// (InfAttack) and pads must never sort below a real value — with inf pads
+// a pad/value tie still yields the mathematically-correct inf statistics.
+#define PAD __builtin_huge_valf()
 
 // Copy a wave-uniform int into a VGPR. Unrolled predicates like
 // `i == pos` / `i >= n` (i compile-time, pos/n uniform) otherwise become
