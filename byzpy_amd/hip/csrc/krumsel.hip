@@ -38,7 +38,11 @@ krum_select_kernel(const float* __restrict__ G, int n, int f, int q,
       } else {
         const float d2 =
             gii + G[(long)j * n + j] - 2.0f * G[(long)row * n + j];
-        v = fmaxf(d2, 0.0f);
+        // adversarial rows produce inf/NaN distances (inf - inf in the
+        // Gram identity); NaN breaks the min/max sort's total order, so
+        // clamp every non-finite distance to +inf — it then sorts last
+        // and never enters a k-smallest sum that has finite candidates
+        v = isfinite(d2) ? fmaxf(d2, 0.0f) : INF;
       }
       buf[j] = v;
     }
