@@ -104,3 +104,18 @@ def test_parameter_server_runner():
         await runner.stop()
 
     asyncio.run(main())
+
+
+def test_package_import_smoke():
+    """Wheel-style import smoke (reference CI tests.yml:59-110)."""
+    import byzpy_amd
+    import byzpy_amd.aggregators
+    import byzpy_amd.attacks
+    import byzpy_amd.engine.parameter_server
+    import byzpy_amd.engine.peer_to_peer
+    import byzpy_amd.graph
+    import byzpy_amd.parallel
+    import byzpy_amd.pre_aggregators
+
+    assert callable(byzpy_amd.run_operator)
+    assert byzpy_amd.__version__
