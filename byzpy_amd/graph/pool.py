@@ -59,6 +59,12 @@ class _SubTaskWorker:
         return "pong"
 
 
+def _looks_dead(e: BaseException) -> bool:
+    if isinstance(e, (ConnectionError, EOFError, BrokenPipeError)):
+        return True
+    return "closed" in str(e).lower() or "pipe" in str(e).lower()
+
+
 def _infer_capabilities(backend_spec: Any) -> Tuple[str, ...]:
     spec = backend_spec if isinstance(backend_spec, str) else getattr(backend_spec, "scheme", "")
     if isinstance(spec, str) and (spec.startswith(("stream", "gpu"))):
@@ -80,10 +86,17 @@ class ActorPoolConfig:
 
 
 class _PoolWorker:
-    def __init__(self, backend: Any, capabilities: Tuple[str, ...], label: str) -> None:
+    def __init__(
+        self,
+        backend: Any,
+        capabilities: Tuple[str, ...],
+        label: str,
+        spec: Any = None,
+    ) -> None:
         self.backend = backend
         self.capabilities = set(capabilities) | {label}
         self.label = label
+        self.spec = spec  # original string spec: enables respawn after a crash
         self.ref = ActorRef(backend)
         self._in_process = getattr(backend, "scheme", "") in ("thread", "stream")
         self.torch_threads = 0  # set by the pool for out-of-process workers
@@ -101,6 +114,21 @@ class _PoolWorker:
 
     async def close(self) -> None:
         await self.backend.close()
+
+    async def respawn(self) -> bool:
+        """Replace a crashed worker with a fresh backend (string specs
+        only — instance backends cannot be re-created)."""
+        if not isinstance(self.spec, str):
+            return False
+        try:
+            await self.backend.close()
+        except Exception:
+            pass
+        self.backend = resolve_backend(self.spec)
+        self.ref = ActorRef(self.backend)
+        await self.backend.start()
+        await self.backend.construct(_SubTaskWorker, self.torch_threads)
+        return True
 
 
 class ActorPoolChannel:
@@ -144,7 +172,9 @@ class ActorPool:
                 backend = resolve_backend(
                     cfg.backend if isinstance(cfg.backend, str) else cfg.backend
                 )
-                self.workers.append(_PoolWorker(backend, caps, label))
+                self.workers.append(
+                    _PoolWorker(backend, caps, label, spec=cfg.backend)
+                )
         import os
 
         n_out = sum(1 for w in self.workers if not w._in_process)
@@ -227,6 +257,15 @@ class ActorPool:
             idx = await self._acquire(subtask.affinity)
             try:
                 return await self.workers[idx].run(subtask)
+            except (ConnectionError, EOFError, OSError, RuntimeError) as e:
+                # worker may have died (killed process, broken pipe):
+                # respawn it so the pool heals before the next attempt
+                last_err = e
+                if _looks_dead(e):
+                    try:
+                        await self.workers[idx].respawn()
+                    except Exception:  # noqa: BLE001 — stays broken, others serve
+                        pass
             except BaseException as e:  # noqa: BLE001
                 last_err = e
             finally:

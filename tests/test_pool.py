@@ -130,3 +130,29 @@ def test_many_concurrent_subtasks():
         await pool.close()
 
     asyncio.run(main())
+
+
+def _pid():
+    import os
+
+    return os.getpid()
+
+
+def test_process_worker_respawn_after_crash():
+    """A killed worker process heals on retry (aux §5.3 fault recovery)."""
+
+    async def main():
+        pool = ActorPool(ActorPoolConfig(backend="process", count=1))
+        await pool.start()
+        pid1 = await pool.run_subtask(SubTask(fn=_pid))
+        # kill the worker out from under the pool
+        import os
+        import signal
+
+        os.kill(pid1, signal.SIGKILL)
+        await asyncio.sleep(0.2)
+        pid2 = await pool.run_subtask(SubTask(fn=_pid, max_retries=2))
+        assert pid2 != pid1
+        await pool.close()
+
+    asyncio.run(main())

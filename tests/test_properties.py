@@ -1,0 +1,107 @@
+"""Property-based invariants (hypothesis) for the robust aggregators."""
+import numpy as np
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from byzpy_amd.ops import functional as F
+
+
+def _matrix(n, d, seed):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(n, d, generator=g)
+
+
+matrix_params = st.tuples(
+    st.integers(min_value=3, max_value=24),
+    st.integers(min_value=1, max_value=65),
+    st.integers(min_value=0, max_value=10_000),
+)
+
+
+@settings(max_examples=25, deadline=None)
+@given(matrix_params)
+def test_median_permutation_invariant(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    perm = torch.randperm(n, generator=torch.Generator().manual_seed(seed + 1))
+    assert torch.allclose(F.median(X), F.median(X[perm]), atol=1e-5)
+
+
+@settings(max_examples=25, deadline=None)
+@given(matrix_params)
+def test_trimmed_mean_permutation_invariant(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = (n - 1) // 2
+    perm = torch.randperm(n, generator=torch.Generator().manual_seed(seed + 1))
+    assert torch.allclose(F.trimmed_mean(X, f), F.trimmed_mean(X[perm], f), atol=1e-5)
+
+
+@settings(max_examples=25, deadline=None)
+@given(matrix_params)
+def test_median_between_min_max(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    med = F.median(X)
+    assert (med >= X.min(dim=0).values - 1e-6).all()
+    assert (med <= X.max(dim=0).values + 1e-6).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_clip_rows_norm_bound_and_idempotent(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    thr = 1.0
+    Y = F.clip_rows(X, thr)
+    assert (Y.norm(dim=1) <= thr + 1e-4).all()
+    Z = F.clip_rows(Y, thr)
+    assert torch.allclose(Y, Z, atol=1e-5)
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_krum_winner_is_input_row(p):
+    n, d, seed = p
+    if n < 4:
+        return
+    X = _matrix(n, d, seed)
+    f = (n - 2) // 2
+    if n - f - 1 < 1:
+        return
+    out = F.krum(X, f)
+    assert any(torch.allclose(out, X[i]) for i in range(n))
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_bucketing_mean_preserving(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    b = max(1, n // 3)
+    if n % b != 0:
+        return  # equal buckets only: mean of bucket-means == global mean
+    out = F.bucketing(X, b, perm=list(range(n)))
+    assert torch.allclose(out.mean(dim=0), X.mean(dim=0), atol=1e-4)
+
+
+@settings(max_examples=15, deadline=None)
+@given(matrix_params)
+def test_geometric_median_translation_equivariant(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    shift = torch.randn(d, generator=torch.Generator().manual_seed(seed + 2))
+    a = F.geometric_median(X, tol=1e-8, max_iter=300)
+    b = F.geometric_median(X + shift, tol=1e-8, max_iter=300)
+    assert torch.allclose(a + shift, b, atol=1e-2)
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_nnm_rows_in_convex_hull_bounds(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = max(0, n // 4)
+    out = F.nnm(X, f)
+    assert (out.max() <= X.max() + 1e-5) and (out.min() >= X.min() - 1e-5)
