@@ -311,6 +311,126 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   }
 }
 
+// ---------------------------------------------------------------------------
+// LDS-staged f32 Gram (fast path for d % 4 == 0): same structure as the
+// bf16 v3 kernel with v_mfma_f32_16x16x4_f32 fragments (exact f32 at the
+// vector rate, guide §3). Chunk = 64 rows x 64 k f32 = 16 KB. The MFMA
+// A/B operand is ONE float per lane (A[l&15][k0 + (l>>4)]), read from LDS
+// with ds_read_b32 — a 16-lane group shares k and spans 16 rows, so a
+// linear [row][k] image is 16-way bank-conflicted; a slot-granular XOR
+// (16 B slot s -> s ^ (row & 7)) spreads the group over 8 slot positions
+// (<= 2-way).
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
+template <bool DIAG>
+__global__ void __launch_bounds__(1024, 2)
+gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
+                    int n, long d, long k_per_block) {
+  constexpr int BKF = 64;                        // k elements per chunk
+  constexpr int CHUNK_BYTES = TILE * BKF * 4;    // 16 KB
+  __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
+  char* const bufA0 = smem;
+  char* const bufA1 = smem + CHUNK_BYTES;
+  char* const bufB0 = DIAG ? smem : smem + 2 * CHUNK_BYTES;
+  char* const bufB1 = DIAG ? smem + CHUNK_BYTES : smem + 3 * CHUNK_BYTES;
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int row_base = blockIdx.y * TILE;
+  const int col_base = blockIdx.z * TILE;
+
+  const long k_lo = (long)blockIdx.x * k_per_block;
+  const long k_hi = min(d, k_lo + k_per_block);
+  const long nchunks = (k_hi - k_lo + BKF - 1) / BKF;
+
+  // stage: thread t owns LDS (row = t>>4, slot = t&15); content of that
+  // slot is global slot (slot ^ (row & 7)) — both-sides swizzle
+  const int st_row = t >> 4;
+  const int st_slot = t & 15;
+  const int src_slot = st_slot ^ (st_row & 7);
+  const int a_rows = min(TILE, n - row_base);
+  const int b_rows = min(TILE, n - col_base);
+  const float* a_src =
+      X + (long)(row_base + min(st_row, a_rows - 1)) * d + (long)src_slot * 4;
+  const float* b_src =
+      X + (long)(col_base + min(st_row, b_rows - 1)) * d + (long)src_slot * 4;
+
+  f32x4v ra, rb;
+  auto stage_load = [&](long c) {
+    const long k0 = k_lo + c * BKF;
+    const bool full = (k0 + BKF <= k_hi);
+    if (full && a_rows == TILE) {
+      ra = *reinterpret_cast<const f32x4v*>(a_src + k0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long k = k0 + (long)src_slot * 4 + j;
+        ra[j] = (st_row < a_rows && k < k_hi) ? a_src[k0 + j] : 0.0f;
+      }
+    }
+    if (!DIAG) {
+      if (full && b_rows == TILE) {
+        rb = *reinterpret_cast<const f32x4v*>(b_src + k0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const long k = k0 + (long)src_slot * 4 + j;
+          rb[j] = (st_row < b_rows && k < k_hi) ? b_src[k0 + j] : 0.0f;
+        }
+      }
+    }
+  };
+  auto stage_write = [&](int which) {
+    char* a = which ? bufA1 : bufA0;
+    *reinterpret_cast<f32x4v*>(a + st_row * 256 + st_slot * 16) = ra;
+    if (!DIAG) {
+      char* b = which ? bufB1 : bufB0;
+      *reinterpret_cast<f32x4v*>(b + st_row * 256 + st_slot * 16) = rb;
+    }
+  };
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  const int rowA = wr * 16 + (lane & 15);
+  const int rowB = wc * 16 + (lane & 15);
+  const int kgrp = lane >> 4;  // k offset within each 4-k step
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  for (long c = 0; c < nchunks; ++c) {
+    if (c + 1 < nchunks) stage_load(c + 1);
+    const char* A = (c & 1) ? bufA1 : bufA0;
+    const char* B = (c & 1) ? bufB1 : bufB0;
+#pragma unroll
+    for (int step = 0; step < 16; ++step) {
+      const int k = step * 4 + kgrp;  // element column 0..63
+      // element k lives in slot (k>>2), swizzled by row
+      const int sa = ((k >> 2) ^ (rowA & 7)) * 16 + (k & 3) * 4;
+      const int sb = ((k >> 2) ^ (rowB & 7)) * 16 + (k & 3) * 4;
+      const float a = *reinterpret_cast<const float*>(A + rowA * 256 + sa);
+      const float b = *reinterpret_cast<const float*>(B + rowB * 256 + sb);
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+    if (c + 1 < nchunks) {
+      stage_write((c + 1) & 1);
+      __syncthreads();
+    }
+  }
+
+  const int out_col = col_base + wc * 16 + (lane & 15);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
+    if (out_row < n && out_col < n)
+      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+  }
+}
+
 inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
   const int tiles = (n + TILE - 1) / TILE;
   const long tile_blocks = (long)tiles * tiles;
@@ -352,6 +472,15 @@ void launch_gram_f32(const float* X, float* G, int n, long d,
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
   dim3 grid(splitk, tiles, tiles);
+  if ((d % 4) == 0 && (kpb % 64) == 0) {
+    if (tiles == 1)
+      hipLaunchKernelGGL((gram_f32_lds_kernel<true>), grid, dim3(WAVES * 64),
+                         0, stream, X, G, n, d, kpb);
+    else
+      hipLaunchKernelGGL((gram_f32_lds_kernel<false>), grid, dim3(WAVES * 64),
+                         0, stream, X, G, n, d, kpb);
+    return;
+  }
   hipLaunchKernelGGL(gram_f32_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
                      n, d, kpb);
 }
