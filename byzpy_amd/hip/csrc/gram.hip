@@ -393,7 +393,10 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
     }
   };
 
-  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  // two accumulators break the dependent-MFMA chain (16x16x4 f32 has a
+  // 40-cycle dependent-accumulator latency, guide §3) — summed at the end
+  f32x4 acc0 = {0.0f, 0.0f, 0.0f, 0.0f};
+  f32x4 acc1 = {0.0f, 0.0f, 0.0f, 0.0f};
   const int rowA = wr * 16 + (lane & 15);
   const int rowB = wc * 16 + (lane & 15);
   const int kgrp = lane >> 4;  // k offset within each 4-k step
@@ -413,7 +416,10 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
       const int sb = ((k >> 2) ^ (rowB & 7)) * 16 + (k & 3) * 4;
       const float a = *reinterpret_cast<const float*>(A + rowA * 256 + sa);
       const float b = *reinterpret_cast<const float*>(B + rowB * 256 + sb);
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+      if (step & 1)
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc1, 0, 0, 0);
+      else
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc0, 0, 0, 0);
     }
     __syncthreads();
     if (c + 1 < nchunks) {
@@ -427,7 +433,7 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
   for (int r = 0; r < 4; ++r) {
     const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
     if (out_row < n && out_col < n)
-      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+      atomicAdd(&G[(long)out_row * n + out_col], acc0[r] + acc1[r]);
   }
 }
 
