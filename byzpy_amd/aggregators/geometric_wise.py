@@ -19,7 +19,7 @@ from byzpy_amd.graph.subtask import SubTask
 from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops import functional as F
 from byzpy_amd.ops.base import OpContext
-from byzpy_amd.storage.shared_store import register_tensor
+from byzpy_amd.storage.shared_store import register_tensor, write_handle
 from byzpy_amd.utils.flatten import stack_gradients, to_like
 
 
@@ -136,27 +136,29 @@ class GeometricMedian(Aggregator):
             self._cleanup(handles)
             return to_like(self._aggregate(X), like)
         use_shm = handles != []
+        center_handle = None
         try:
             n = X.shape[0]
             chunk = max(1, min(self.chunk_size, n))
             z = (F.median(X) if self.init == "median" else X.float().mean(dim=0)).float()
+            if use_shm:
+                # ONE segment, rewritten per iteration (reference
+                # geometric_median.py:126 _write_handle semantics)
+                center_handle = register_tensor(z)
             for _ in range(self.max_iter):
                 if use_shm:
-                    center_ref = register_tensor(z)
+                    write_handle(center_handle, z)
+                    center_ref = center_handle
                 else:
                     center_ref = z
-                try:
-                    tasks = [
-                        SubTask(
-                            fn=SF.weiszfeld_chunk,
-                            args=(ref, lo, hi, center_ref, self.eps),
-                        )
-                        for lo, hi in chunk_ranges(n, chunk)
-                    ]
-                    partials = await self._run_subtasks(ctx, tasks)
-                finally:
-                    if use_shm:
-                        self._cleanup([center_ref])
+                tasks = [
+                    SubTask(
+                        fn=SF.weiszfeld_chunk,
+                        args=(ref, lo, hi, center_ref, self.eps),
+                    )
+                    for lo, hi in chunk_ranges(n, chunk)
+                ]
+                partials = await self._run_subtasks(ctx, tasks)
                 num = sum(p[0] for p in partials)
                 den = sum(p[1] for p in partials)
                 z_new = num / den
@@ -166,6 +168,8 @@ class GeometricMedian(Aggregator):
                     break
             return to_like(z.to(X.dtype), like)
         finally:
+            if center_handle is not None:
+                self._cleanup([center_handle])
             self._cleanup(handles)
 
 

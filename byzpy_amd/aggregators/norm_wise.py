@@ -17,7 +17,7 @@ from byzpy_amd.graph.subtask import SubTask
 from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops import functional as F
 from byzpy_amd.ops.base import OpContext
-from byzpy_amd.storage.shared_store import register_tensor
+from byzpy_amd.storage.shared_store import register_tensor, write_handle
 from byzpy_amd.utils.flatten import to_like
 
 
@@ -60,6 +60,7 @@ class CenteredClipping(Aggregator):
             self._cleanup(handles)
             return to_like(self._aggregate(X), like)
         use_shm = handles != []
+        center_handle = None
         try:
             n = X.shape[0]
             chunk = max(1, min(self.chunk_size, n))
@@ -69,23 +70,27 @@ class CenteredClipping(Aggregator):
                 v = F.median(X).float()
             else:
                 v = torch.zeros(X.shape[1], dtype=torch.float32)
+            if use_shm:
+                center_handle = register_tensor(v)
             for _ in range(self.M):
-                center_ref = register_tensor(v) if use_shm else v
-                try:
-                    tasks = [
-                        SubTask(
-                            fn=SF.cc_chunk,
-                            args=(ref, lo, hi, center_ref, self.c_tau, self.eps),
-                        )
-                        for lo, hi in chunk_ranges(n, chunk)
-                    ]
-                    partials = await self._run_subtasks(ctx, tasks)
-                finally:
-                    if use_shm:
-                        self._cleanup([center_ref])
+                if use_shm:
+                    write_handle(center_handle, v)
+                    center_ref = center_handle
+                else:
+                    center_ref = v
+                tasks = [
+                    SubTask(
+                        fn=SF.cc_chunk,
+                        args=(ref, lo, hi, center_ref, self.c_tau, self.eps),
+                    )
+                    for lo, hi in chunk_ranges(n, chunk)
+                ]
+                partials = await self._run_subtasks(ctx, tasks)
                 v = v + sum(partials) / n
             return to_like(v.to(X.dtype), like)
         finally:
+            if center_handle is not None:
+                self._cleanup([center_handle])
             self._cleanup(handles)
 
 

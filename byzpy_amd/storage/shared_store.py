@@ -80,6 +80,23 @@ def open_tensor_copy(handle: SharedTensorHandle) -> torch.Tensor:
         return view.clone()
 
 
+def write_handle(handle: SharedTensorHandle, t: Any) -> None:
+    """Overwrite an existing segment's contents in place (iterative
+    aggregators re-broadcast their center each iteration — reference
+    geometric_median.py:204-206)."""
+    if isinstance(t, torch.Tensor):
+        t = t.detach().cpu()
+        arr = t.view(torch.uint16).numpy() if t.dtype == torch.bfloat16 else t.numpy()
+    else:
+        arr = np.asarray(t)
+    seg = shared_memory.SharedMemory(name=handle.name)
+    try:
+        dst = np.ndarray(arr.shape, dtype=arr.dtype, buffer=seg.buf)
+        dst[...] = arr
+    finally:
+        seg.close()
+
+
 def cleanup_tensor(handle: SharedTensorHandle) -> None:
     with contextlib.suppress(FileNotFoundError):
         seg = shared_memory.SharedMemory(name=handle.name)
