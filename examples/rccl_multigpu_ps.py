@@ -9,7 +9,11 @@ one all-to-all and TrimmedMean runs shard-local.
 """
 from __future__ import annotations
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
