@@ -234,45 +234,79 @@ DEV u32 extract_at_pk(const u32 (&v)[P], int pos) {
   return r;
 }
 
-template <int P>
-__global__ void colsel_pk_median_bf16(const unsigned short* __restrict__ X,
+// QUADS=true: each thread owns TWO adjacent packed pairs (4 columns,
+// 8 B/lane row reads — wider DRAM bursts; ~150 VGPRs at P=64) with two
+// independent sorting networks; QUADS=false: one pair (4 B/lane).
+template <int P, bool QUADS>
+__global__ void
+// QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
+// so the allocator doesn't cap at 128 and spill
+__launch_bounds__(256, (QUADS && P >= 64) ? 3 : 4)
+colsel_pk_median_bf16(const unsigned short* __restrict__ X,
                                       unsigned short* __restrict__ out, int n,
                                       long d, int f) {
   const long npairs = d >> 1;
-  const long pair0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long nunits = QUADS ? (npairs >> 1) : npairs;
+  const long unit0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const long rowstride = d >> 1;  // in u32 units
-  for (long pair = pair0; pair < npairs; pair += stride) {
-    u32 v[P];
+  for (long unit = unit0; unit < nunits; unit += stride) {
+    const long pair = QUADS ? unit * 2 : unit;
+    u32 v[P], v2[QUADS ? P : 1];
     // raw loads first + pinned alternation: see colsel_reg_kernel's load
-    // phase for why (one live base, no SGPR spill storm)
+    // phase for why (one live base, no SGPR spill storm). The walk
+    // condition is vecified too — 64 uniform selects otherwise become
+    // batched SGPR mask pairs and spill at QUADS register pressure.
+    const int n_walk = vecify(n);
     const u32* p = reinterpret_cast<const u32*>(X) + pair;
 #pragma unroll
     for (int i = 0; i < P; ++i) {
-      v[i] = *p;
-      if (i + 1 < n) p += rowstride;
+      if (QUADS) {
+        const uint2 w = *reinterpret_cast<const uint2*>(p);
+        v[i] = w.x;
+        v2[i] = w.y;
+      } else {
+        v[i] = *p;
+      }
+      p += (i + 1 < n_walk) ? rowstride : 0;
       __builtin_amdgcn_sched_barrier(0);
     }
-#pragma unroll
-    for (int i = 0; i < P; ++i) v[i] = pk_key_from_bf16(v[i]);
     const int nv = vecify(n);
 #pragma unroll
+    for (int i = 0; i < P; ++i) {
+      v[i] = pk_key_from_bf16(v[i]);
+      if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
+    }
+#pragma unroll
     for (int i = 0; i < P; ++i)
-      if (i >= nv) v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
+      if (i >= nv) {
+        v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
+        if (QUADS) v2[i] = 0xFFFFFFFFu;
+      }
     bitonic_sort_pk<P>(v);
-    const u32 lo = extract_at_pk<P>(v, vecify((n - 1) >> 1));
-    const u32 hi = extract_at_pk<P>(v, vecify(n >> 1));
-    const float m0 =
-        0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
-    const float m1 =
-        0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
-    union { unsigned short s[2]; u32 w; } o;
-    union { unsigned short s; __hip_bfloat16 h; } c0, c1;
-    c0.h = __float2bfloat16(m0);
-    c1.h = __float2bfloat16(m1);
-    o.s[0] = c0.s;
-    o.s[1] = c1.s;
-    reinterpret_cast<u32*>(out)[pair] = o.w;
+    if (QUADS) bitonic_sort_pk<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]));
+    const int plo = vecify((n - 1) >> 1), phi = vecify(n >> 1);
+    u32* outw = reinterpret_cast<u32*>(out);
+#pragma unroll
+    for (int half = 0; half < (QUADS ? 2 : 1); ++half) {
+      const u32* arr = half ? v2 : v;
+      u32 lo = 0, hi = 0;
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        if (i == plo) lo = arr[i];
+        if (i == phi) hi = arr[i];
+      }
+      const float m0 =
+          0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
+      const float m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
+      union { unsigned short s[2]; u32 w; } o;
+      union { unsigned short s; __hip_bfloat16 h; } c0, c1;
+      c0.h = __float2bfloat16(m0);
+      c1.h = __float2bfloat16(m1);
+      o.s[0] = c0.s;
+      o.s[1] = c1.s;
+      outw[pair + half] = o.w;
+    }
   }
 }
 
@@ -387,23 +421,26 @@ void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
   if (mode == MEDIAN && n <= 64 && (d % 2) == 0) {
     const int block = 256;
-    const long pairs = d >> 1;
-    const long want = (pairs + block - 1) / block;
+    const bool quads = (d % 4) == 0;
+    const long units = quads ? (d >> 2) : (d >> 1);
+    const long want = (units + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
     const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
     unsigned short* Ou = reinterpret_cast<unsigned short*>(out);
-    if (n <= 8)
-      hipLaunchKernelGGL((colsel_pk_median_bf16<8>), dim3(grid), dim3(block),
-                         0, stream, Xu, Ou, n, d, f);
-    else if (n <= 16)
-      hipLaunchKernelGGL((colsel_pk_median_bf16<16>), dim3(grid), dim3(block),
-                         0, stream, Xu, Ou, n, d, f);
-    else if (n <= 32)
-      hipLaunchKernelGGL((colsel_pk_median_bf16<32>), dim3(grid), dim3(block),
-                         0, stream, Xu, Ou, n, d, f);
-    else
-      hipLaunchKernelGGL((colsel_pk_median_bf16<64>), dim3(grid), dim3(block),
-                         0, stream, Xu, Ou, n, d, f);
+#define PK_LAUNCH(P)                                                          \
+  do {                                                                        \
+    if (quads)                                                                \
+      hipLaunchKernelGGL((colsel_pk_median_bf16<P, true>), dim3(grid),        \
+                         dim3(block), 0, stream, Xu, Ou, n, d, f);            \
+    else                                                                      \
+      hipLaunchKernelGGL((colsel_pk_median_bf16<P, false>), dim3(grid),       \
+                         dim3(block), 0, stream, Xu, Ou, n, d, f);            \
+  } while (0)
+    if (n <= 8) PK_LAUNCH(8);
+    else if (n <= 16) PK_LAUNCH(16);
+    else if (n <= 32) PK_LAUNCH(32);
+    else PK_LAUNCH(64);
+#undef PK_LAUNCH
     return;
   }
   launch_colsel_typed<__hip_bfloat16>(X, out, n, d, mode, f, stream);
