@@ -421,7 +421,10 @@ void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
   if (mode == MEDIAN && n <= 64 && (d % 2) == 0) {
     const int block = 256;
-    const bool quads = (d % 4) == 0;
+    // A/B'd on MI355X: QUADS (8 B/lane) ties the single-pair variant
+    // (3.9 vs 3.8 ms at 64 x 125M) — the occupancy drop offsets the wider
+    // bursts. Keep single-pair as the shipping path.
+    const bool quads = false;
     const long units = quads ? (d >> 2) : (d >> 1);
     const long want = (units + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
