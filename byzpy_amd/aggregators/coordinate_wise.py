@@ -18,7 +18,7 @@ from byzpy_amd.aggregators.base import Aggregator
 from byzpy_amd.graph.subtask import SubTask
 from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops.base import OpContext
-from byzpy_amd.utils.flatten import stack_gradients, to_like
+from byzpy_amd.utils.flatten import to_like
 
 
 class _FeatureChunkedAggregator(Aggregator):

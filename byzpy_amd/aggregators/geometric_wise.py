@@ -13,7 +13,7 @@ from typing import Any, List, Sequence
 import torch
 
 from byzpy_amd.aggregators import _subtask_fns as SF
-from byzpy_amd.aggregators._chunking import chunk_ranges, select_adaptive_chunk_size
+from byzpy_amd.aggregators._chunking import chunk_ranges
 from byzpy_amd.aggregators.base import Aggregator
 from byzpy_amd.graph.subtask import SubTask
 from byzpy_amd.hip import dispatch as D

@@ -9,8 +9,6 @@ from __future__ import annotations
 
 from typing import Any
 
-import torch
-
 from byzpy_amd.ops.base import Operator, OpContext
 from byzpy_amd.utils.flatten import stack_gradients
 

@@ -14,7 +14,7 @@ from torch import nn
 
 from byzpy_amd.attacks.base import Attack
 from byzpy_amd.ops import functional as F
-from byzpy_amd.utils.flatten import stack_gradients, to_like
+from byzpy_amd.utils.flatten import to_like
 
 
 class EmpireAttack(Attack):

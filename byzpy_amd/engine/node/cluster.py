@@ -5,9 +5,9 @@ map pushed into every router before start_all).
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
-from byzpy_amd.engine.node.context import InProcessContext, ProcessContext
+from byzpy_amd.engine.node.context import ProcessContext
 from byzpy_amd.engine.node.decentralized import DecentralizedNode
 
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import asyncio
 import struct
-from typing import Any, Callable, Dict, Optional
+from typing import Any, Dict, Optional
 
 import cloudpickle
 

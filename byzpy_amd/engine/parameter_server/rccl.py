@@ -15,7 +15,7 @@ TrimmedMean, 8 GPUs.
 """
 from __future__ import annotations
 
-from typing import Callable, List, Optional
+from typing import Callable, List
 
 import torch
 

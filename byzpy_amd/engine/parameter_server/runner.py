@@ -8,7 +8,7 @@ ParameterServer (actors) and RcclParameterServer (one rank per GPU).
 """
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional, Sequence
+from typing import Callable, Optional, Sequence
 
 import torch
 

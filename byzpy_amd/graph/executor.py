@@ -6,14 +6,13 @@ pool creation + graph/scheduler caching 124-258; _InputMappingOperator
 """
 from __future__ import annotations
 
-import asyncio
 from typing import Any, Dict, Optional, Sequence
 
 from byzpy_amd.graph.graph import ComputationGraph
 from byzpy_amd.graph.ops import make_single_operator_graph
 from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
 from byzpy_amd.graph.scheduler import NodeScheduler
-from byzpy_amd.ops.base import Operator, OpContext
+from byzpy_amd.ops.base import Operator
 
 
 def _detect_input_key(op: Operator, inputs: Dict[str, Any]) -> str:

@@ -7,7 +7,7 @@ collection, default output = last topo node).
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Mapping, Optional, Sequence, Union
+from typing import Dict, List, Mapping, Optional, Sequence, Union
 
 from byzpy_amd.ops.base import Operator
 

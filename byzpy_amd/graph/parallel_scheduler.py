@@ -9,7 +9,7 @@ from __future__ import annotations
 import asyncio
 from typing import Any, Dict, List, Optional
 
-from byzpy_amd.graph.graph import ComputationGraph, GraphInput, MessageSource
+from byzpy_amd.graph.graph import ComputationGraph
 from byzpy_amd.graph.scheduler import NodeScheduler
 from byzpy_amd.ops.base import OpContext
 

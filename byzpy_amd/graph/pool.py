@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import asyncio
 import collections
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from functools import lru_cache
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 

@@ -9,7 +9,6 @@ path for plain GEMM shapes.
 """
 from __future__ import annotations
 
-import math
 from typing import Optional, Sequence
 
 import torch

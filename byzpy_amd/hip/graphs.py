@@ -14,7 +14,7 @@ tol=0-style usage or use CenteredClipping-like fixed-M ops).
 """
 from __future__ import annotations
 
-from typing import Any, Callable, Optional
+from typing import Any, Callable
 
 import torch
 

@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -12,13 +12,13 @@ shard of the aggregate.
 """
 from __future__ import annotations
 
-from typing import Optional, Sequence
+from typing import Sequence
 
 import torch
 
 from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops import functional as F
-from byzpy_amd.parallel.dist import all_reduce_, get_world_size, is_initialized
+from byzpy_amd.parallel.dist import all_reduce_
 
 
 # -- coordinate-wise: pure local --------------------------------------------

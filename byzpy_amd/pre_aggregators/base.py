@@ -11,7 +11,7 @@ import torch
 
 from byzpy_amd.ops.base import Operator, OpContext
 from byzpy_amd.aggregators.base import build_matrix_ref, cleanup_handles
-from byzpy_amd.utils.flatten import LikeTemplate, stack_gradients, to_like
+from byzpy_amd.utils.flatten import stack_gradients, to_like
 
 
 class PreAggregator(Operator):

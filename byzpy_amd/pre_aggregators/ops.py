@@ -17,7 +17,7 @@ from byzpy_amd.graph.subtask import SubTask
 from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops.base import OpContext
 from byzpy_amd.pre_aggregators.base import PreAggregator
-from byzpy_amd.utils.flatten import stack_gradients, to_like
+from byzpy_amd.utils.flatten import to_like
 
 
 class Clipping(PreAggregator):

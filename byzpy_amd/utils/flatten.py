@@ -8,7 +8,7 @@ shape (or parameter-list structure) only at the boundary.
 """
 from __future__ import annotations
 
-from typing import Any, List, Sequence, Tuple
+from typing import Any, List, Tuple
 
 import numpy as np
 import torch

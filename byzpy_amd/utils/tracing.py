@@ -8,7 +8,6 @@ from __future__ import annotations
 
 import contextlib
 import ctypes
-import os
 from typing import Iterator, Optional
 
 _lib: Optional[ctypes.CDLL] = None
