@@ -199,12 +199,13 @@ __global__ void gram_f32_kernel(const float* __restrict__ X,
 // images coincide — staged once, halving staging traffic.
 // ---------------------------------------------------------------------------
 
-template <bool DIAG>
+template <bool DIAG, int BK = 128>
 __global__ void __launch_bounds__(1024, 2)
 gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
                      float* __restrict__ G, int n, long d, long k_per_block) {
-  constexpr int BK = 128;                       // k elements per chunk
-  constexpr int CHUNK_BYTES = TILE * BK * 2;    // 16 KB
+  constexpr int SLOTS = BK / 8;                 // 16-B slots per row
+  constexpr int SUB = SLOTS / 16;               // stage passes per thread
+  constexpr int CHUNK_BYTES = TILE * BK * 2;    // 16 KB at BK=128
   __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
   // no pointer ARRAYS into LDS (clang rejects the addrspace-cast array
   // initializer); select buffers with ternaries instead
@@ -227,50 +228,59 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   // stage geometry: thread t handles LDS row st_row, memory slot st_slot;
   // the CONTENT of that slot is global slot (st_slot ^ (st_row & 15)).
   const int st_row = t >> 4;
-  const int st_slot = t & 15;
-  const int src_slot = st_slot ^ (st_row & 15);
   const int a_rows = min(TILE, n - row_base);
   const int b_rows = min(TILE, n - col_base);
-  const __hip_bfloat16* a_src =
-      X + (long)(row_base + min(st_row, a_rows - 1)) * d + (long)src_slot * 8;
-  const __hip_bfloat16* b_src =
-      X + (long)(col_base + min(st_row, b_rows - 1)) * d + (long)src_slot * 8;
+  // SUB slots per thread: pass u stages slot (t&15) + u*16
+  int st_slot[SUB], src_slot[SUB];
+#pragma unroll
+  for (int u = 0; u < SUB; ++u) {
+    st_slot[u] = (t & 15) + u * 16;
+    src_slot[u] = st_slot[u] ^ (st_row & 15);
+  }
+  const __hip_bfloat16* a_row_src =
+      X + (long)(row_base + min(st_row, a_rows - 1)) * d;
+  const __hip_bfloat16* b_row_src =
+      X + (long)(col_base + min(st_row, b_rows - 1)) * d;
 
-  bf16x8 ra, rb;
+  bf16x8 ra[SUB], rb[SUB];
   auto stage_load = [&](long c) {
     const long k0 = k_lo + c * BK;
     const bool full = (k0 + BK <= k_hi);
-    if (full && a_rows == TILE) {
-      ra = *reinterpret_cast<const bf16x8*>(a_src + k0);
-    } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const long k = k0 + (long)src_slot * 8 + j;
-        ra[j] = (st_row < a_rows && k < k_hi)
-                    ? *reinterpret_cast<const __bf16*>(a_src + k0 + j)
-                    : (__bf16)0.0f;
-      }
-    }
-    if (!DIAG) {
-      if (full && b_rows == TILE) {
-        rb = *reinterpret_cast<const bf16x8*>(b_src + k0);
+    for (int u = 0; u < SUB; ++u) {
+      const long off = k0 + (long)src_slot[u] * 8;
+      if (full && a_rows == TILE) {
+        ra[u] = *reinterpret_cast<const bf16x8*>(a_row_src + off);
       } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const long k = k0 + (long)src_slot * 8 + j;
-          rb[j] = (st_row < b_rows && k < k_hi)
-                      ? *reinterpret_cast<const __bf16*>(b_src + k0 + j)
-                      : (__bf16)0.0f;
+        for (int j = 0; j < 8; ++j)
+          ra[u][j] = (st_row < a_rows && off + j < k_hi)
+                         ? *reinterpret_cast<const __bf16*>(a_row_src + off + j)
+                         : (__bf16)0.0f;
+      }
+      if (!DIAG) {
+        if (full && b_rows == TILE) {
+          rb[u] = *reinterpret_cast<const bf16x8*>(b_row_src + off);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            rb[u][j] = (st_row < b_rows && off + j < k_hi)
+                           ? *reinterpret_cast<const __bf16*>(b_row_src + off + j)
+                           : (__bf16)0.0f;
         }
       }
     }
   };
   auto stage_write = [&](int which) {
     char* a = which ? bufA1 : bufA0;
-    *reinterpret_cast<bf16x8*>(a + st_row * 256 + st_slot * 16) = ra;
-    if (!DIAG) {
-      char* b = which ? bufB1 : bufB0;
-      *reinterpret_cast<bf16x8*>(b + st_row * 256 + st_slot * 16) = rb;
+    char* b = which ? bufB1 : bufB0;
+#pragma unroll
+    for (int u = 0; u < SUB; ++u) {
+      *reinterpret_cast<bf16x8*>(a + st_row * (SLOTS * 16) + st_slot[u] * 16) =
+          ra[u];
+      if (!DIAG)
+        *reinterpret_cast<bf16x8*>(b + st_row * (SLOTS * 16) +
+                                   st_slot[u] * 16) = rb[u];
     }
   };
 
@@ -287,12 +297,12 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
     const char* A = (c & 1) ? bufA1 : bufA0;
     const char* B = (c & 1) ? bufB1 : bufB0;
 #pragma unroll
-    for (int step = 0; step < 4; ++step) {
+    for (int step = 0; step < BK / 32; ++step) {
       const int q = step * 4 + grp;
       const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          A + rowA * 256 + ((q ^ (rowA & 15)) * 16));
+          A + rowA * (SLOTS * 16) + ((q ^ (rowA & 15)) * 16));
       const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-          B + rowB * 256 + ((q ^ (rowB & 15)) * 16));
+          B + rowB * (SLOTS * 16) + ((q ^ (rowB & 15)) * 16));
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     __syncthreads();
@@ -446,7 +456,7 @@ inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
   if (want > max_by_d) want = max_by_d;
   splitk = (int)want;
   k_per_block = (d + splitk - 1) / splitk;
-  k_per_block = ((k_per_block + 127) / 128) * 128;  // keep the vector path
+  k_per_block = ((k_per_block + 255) / 256) * 256;  // keep the vector path
 }
 
 }  // namespace
@@ -457,15 +467,24 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
   dim3 grid(splitk, tiles, tiles);
-  if ((d % 8) == 0 && (kpb % 128) == 0) {
+  if ((d % 8) == 0 && (kpb % 256) == 0) {
     if (tiles == 1)
-      hipLaunchKernelGGL((gram_bf16_lds_kernel<true>), grid, dim3(WAVES * 64),
-                         0, stream, X, G, n, d, kpb);
+      // single tile (n <= 64): BK=256 halves the barrier count (LDS
+      // 2 x 32 KB still fits 2 blocks/CU with the single A image)
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 256>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
     else
       // off-diagonal tiles need separate A/B images; diagonal blocks of a
       // multi-tile grid still produce correct results with DIAG=false.
-      hipLaunchKernelGGL((gram_bf16_lds_kernel<false>), grid, dim3(WAVES * 64),
-                         0, stream, X, G, n, d, kpb);
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<false, 128>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+  } else if ((d % 8) == 0 && (kpb % 128) == 0) {
+    if (tiles == 1)
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 128>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+    else
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<false, 128>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
   } else {
     hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
                        stream, X, G, n, d, kpb);
