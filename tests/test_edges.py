@@ -1,0 +1,161 @@
+"""Edge-case coverage: message timeouts, chunking env overrides, storage
+write_handle, scheduler input resolution, PS streaming order."""
+import asyncio
+import os
+
+import pytest
+import torch
+
+from byzpy_amd.aggregators._chunking import chunk_ranges, select_adaptive_chunk_size
+from byzpy_amd.graph.graph import ComputationGraph, GraphInput, GraphNode
+from byzpy_amd.graph.ops import CallableOp
+from byzpy_amd.graph.scheduler import MessageAwareNodeScheduler, NodeScheduler
+from byzpy_amd.ops.base import MessageTriggerOp
+from byzpy_amd.storage.shared_store import (
+    cleanup_tensor,
+    open_tensor,
+    register_tensor,
+    write_handle,
+)
+
+
+class TestChunking:
+    def test_shrinks_to_fill_workers(self):
+        # 1000 total, 4 workers want >= 16 chunks -> chunk shrinks
+        c = select_adaptive_chunk_size(1000, 4, 512)
+        assert c <= 128
+
+    def test_shrink_cap(self, monkeypatch):
+        monkeypatch.setenv("BYZPY_AMD_CHUNK_MAX_SHRINK", "2")
+        c = select_adaptive_chunk_size(10_000, 8, 512)
+        assert c >= 256  # at most 2x shrink
+
+    def test_env_min_per_worker(self, monkeypatch):
+        monkeypatch.setenv("BYZPY_AMD_CHUNK_MIN_PER_WORKER", "1")
+        c = select_adaptive_chunk_size(4096, 4, 1024)
+        assert c == 1024  # 4 chunks already cover 4 workers x 1
+
+    def test_ranges_cover(self):
+        spans = list(chunk_ranges(10, 3))
+        assert spans == [(0, 3), (3, 6), (6, 9), (9, 10)]
+
+
+class TestSharedStore:
+    def test_write_handle_updates_view(self):
+        t = torch.zeros(6)
+        h = register_tensor(t)
+        try:
+            write_handle(h, torch.arange(6.0))
+            with open_tensor(h) as view:
+                assert torch.allclose(view, torch.arange(6.0))
+        finally:
+            cleanup_tensor(h)
+
+    def test_bf16_roundtrip(self):
+        t = torch.randn(5).bfloat16()
+        h = register_tensor(t)
+        try:
+            with open_tensor(h) as view:
+                assert view.dtype == torch.bfloat16
+                assert torch.equal(view, t)
+        finally:
+            cleanup_tensor(h)
+
+
+class TestMessageScheduling:
+    def test_wait_timeout(self):
+        async def main():
+            sched = MessageAwareNodeScheduler(ComputationGraph([]))
+            with pytest.raises(asyncio.TimeoutError):
+                await sched.wait_for_message("never", timeout=0.05)
+
+        asyncio.run(main())
+
+    def test_message_trigger_op(self):
+        async def main():
+            g = ComputationGraph(
+                [GraphNode("t", MessageTriggerOp("go"), {})]
+            )
+            sched = MessageAwareNodeScheduler(g)
+            task = asyncio.get_running_loop().create_task(sched.run({}))
+            await asyncio.sleep(0.01)
+            sched.deliver_message("go", {"x": 1})
+            assert await task == {"x": 1}
+
+        asyncio.run(main())
+
+    def test_message_source_field_extraction(self):
+        async def main():
+            g = ComputationGraph(
+                [
+                    GraphNode(
+                        "a",
+                        CallableOp(lambda m: m * 2),
+                        {"m": GraphInput.from_message("v", field="payload")},
+                    )
+                ]
+            )
+            sched = MessageAwareNodeScheduler(g)
+            sched.deliver_message("v", {"payload": 21})
+            return await sched.run({})
+
+        assert asyncio.run(main()) == 42
+
+    def test_plain_scheduler_rejects_messages(self):
+        async def main():
+            g = ComputationGraph(
+                [GraphNode("t", MessageTriggerOp("x"), {})]
+            )
+            with pytest.raises(RuntimeError):
+                await NodeScheduler(g).run({})
+
+        asyncio.run(main())
+
+
+class TestPsStreaming:
+    def test_honest_grads_stream_as_completed(self):
+        """Slow workers must not block fast ones (reference ps.py:89-92)."""
+        from byzpy_amd.engine.parameter_server.ps import ParameterServer
+        from byzpy_amd.aggregators import CoordinateWiseMedian
+
+        order = []
+
+        class Slow:
+            def __init__(self, tag, delay):
+                self.tag, self.delay = tag, delay
+
+            async def honest_gradient_for_next_batch(self):
+                await asyncio.sleep(self.delay)
+                order.append(self.tag)
+                return torch.full((4,), float(len(self.tag)))
+
+            async def apply_server_gradient(self, g):
+                pass
+
+        async def main():
+            nodes = [Slow("slow", 0.1), Slow("a", 0.0), Slow("b", 0.01)]
+            ps = ParameterServer(nodes, [], CoordinateWiseMedian())
+            await ps.round()
+            assert order[0] in ("a", "b") and order[-1] == "slow"
+
+        asyncio.run(main())
+
+
+class TestGraphOutputs:
+    def test_multi_output_dict(self):
+        async def main():
+            g = ComputationGraph(
+                [
+                    GraphNode("a", CallableOp(lambda x: x + 1), {"x": GraphInput("x")}),
+                    GraphNode("b", CallableOp(lambda x: x * 2), {"x": GraphInput("x")}),
+                ],
+                outputs=["a", "b"],
+            )
+            return await NodeScheduler(g).run({"x": 10})
+
+        out = asyncio.run(main())
+        assert out == {"a": 11, "b": 20}
+
+    def test_empty_graph(self):
+        g = ComputationGraph([])
+        assert g.topo_order == [] and g.outputs == []
