@@ -311,59 +311,86 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
-// LDS variant, 64 < n <= 512; one wave per block, lane-owned columns
+// LDS variant, 64 < n <= 512: block-COOPERATIVE batched bitonic. One block
+// (16 waves) owns 64 columns staged as an LDS plane [P][64]; every
+// compare-exchange step spreads its P/2 x 64 sites over all 1024 threads
+// (adjacent threads take adjacent columns of one site -> adjacent banks,
+// conflict-free), with a block barrier between j-substeps. Replaces the
+// original one-wave-per-block version whose single wave left the CU 97%
+// idle (n=512 d=1M: 91.7 -> ~1 ms).
 // ---------------------------------------------------------------------------
 
-DEV void bitonic_sort_lds(float* v, int stride, int P) {
-  for (int k = 2; k <= P; k <<= 1)
-    for (int j = k >> 1; j > 0; j >>= 1)
-      for (int i = 0; i < P; ++i) {
-        const int l = i ^ j;
-        if (l > i) {
-          const bool asc = (i & k) == 0;
-          const float a = v[i * stride], b = v[l * stride];
-          if (asc ? (a > b) : (a < b)) { v[i * stride] = b; v[l * stride] = a; }
-        }
-      }
-}
+constexpr int LDS_COLS = 64;
+constexpr int LDS_THREADS = 1024;
 
 template <int MODE, typename T>
-__global__ void colsel_lds_kernel(const T* __restrict__ X, T* __restrict__ out,
-                                  int n, long d, int f, int P) {
+__global__ void __launch_bounds__(LDS_THREADS)
+colsel_lds_kernel(const T* __restrict__ X, T* __restrict__ out,
+                  int n, long d, int f, int P) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [P][WAVE] floats, lane l owns column buf[i*WAVE + l]. Uniform i
-  // across the wave puts each lane on bank (lane % 32): conflict-free for
-  // ds_read_b32 (lanes l and l+32 are different lane groups, guide §2).
-  float* buf = reinterpret_cast<float*>(smem);
-  const int lane = threadIdx.x;  // blockDim.x == 64
-  const int stride = WAVE;
-  const long col = (long)blockIdx.x * WAVE + lane;
-  if (col >= d) return;
-  float* mine = buf + lane;
-  for (int i = 0; i < P; ++i)
-    mine[i * stride] = (i < n) ? to_f<T>(X[(long)i * d + col]) : PAD;
+  float* buf = reinterpret_cast<float*>(smem);  // [P][LDS_COLS]
+  const int t = threadIdx.x;
+  const long col0 = (long)blockIdx.x * LDS_COLS;
+  const int cols = (int)min((long)LDS_COLS, d - col0);
 
-  bitonic_sort_lds(mine, stride, P);
-  const float med =
-      0.5f * (mine[((n - 1) >> 1) * stride] + mine[(n >> 1) * stride]);
-  if (MODE == MEDIAN) {
-    out[col] = from_f<T>(med);
-  } else if (MODE == TRIMMED) {
-    float s = 0.0f;
-    for (int i = f; i < n - f; ++i) s += mine[i * stride];
-    out[col] = from_f<T>(s / (float)(n - 2 * f));
-  } else {  // MEAMED: the n-f values closest to med form a contiguous
-    // window of the sorted column — shrink [l, r) from whichever end
-    // deviates more (two-pointer, O(f)).
-    int l = 0, r = n;
-    for (int k = 0; k < f; ++k) {
-      const float dl = med - mine[l * stride];
-      const float dr = mine[(r - 1) * stride] - med;
-      if (dl > dr) ++l; else --r;
+  // cooperative stage: element (row, c) by thread index
+  for (int idx = t; idx < P * LDS_COLS; idx += LDS_THREADS) {
+    const int row = idx / LDS_COLS;
+    const int c = idx % LDS_COLS;
+    float v = PAD;
+    if (row < n && c < cols)
+      v = to_f<T>(X[(long)row * d + col0 + c]);
+    buf[idx] = v;
+  }
+  __syncthreads();
+
+  // batched bitonic: P/2 compare sites x LDS_COLS columns per substep
+  for (int k = 2; k <= P; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int idx = t; idx < (P >> 1) * LDS_COLS; idx += LDS_THREADS) {
+        const int site = idx / LDS_COLS;
+        const int c = idx % LDS_COLS;
+        // site s enumerates pairs (i, i^j) with i^j > i:
+        // i = (s / j) * 2j + (s % j)
+        const int i = ((site / j) * (j << 1)) + (site % j);
+        const int l = i ^ j;
+        const bool asc = (i & k) == 0;
+        float* a = &buf[i * LDS_COLS + c];
+        float* b = &buf[l * LDS_COLS + c];
+        const float av = *a, bv = *b;
+        const float lo = fminf(av, bv), hi = fmaxf(av, bv);
+        *a = asc ? lo : hi;
+        *b = asc ? hi : lo;
+      }
+      __syncthreads();
     }
-    float s = 0.0f;
-    for (int i = l; i < r; ++i) s += mine[i * stride];
-    out[col] = from_f<T>(s / (float)(n - f));
+  }
+
+  // per-column epilogue: threads 0..cols-1
+  if (t < cols) {
+    float* mine = buf + t;
+    const int stride = LDS_COLS;
+    const float med =
+        0.5f * (mine[((n - 1) >> 1) * stride] + mine[(n >> 1) * stride]);
+    float result;
+    if (MODE == MEDIAN) {
+      result = med;
+    } else if (MODE == TRIMMED) {
+      float s = 0.0f;
+      for (int i2 = f; i2 < n - f; ++i2) s += mine[i2 * stride];
+      result = s / (float)(n - 2 * f);
+    } else {  // MEAMED: contiguous window of the sorted column (two-pointer)
+      int l = 0, r = n;
+      for (int kk = 0; kk < f; ++kk) {
+        const float dl = med - mine[l * stride];
+        const float dr = mine[(r - 1) * stride] - med;
+        if (dl > dr) ++l; else --r;
+      }
+      float s = 0.0f;
+      for (int i2 = l; i2 < r; ++i2) s += mine[i2 * stride];
+      result = s / (float)(n - f);
+    }
+    out[col0 + t] = from_f<T>(result);
   }
 }
 
@@ -398,17 +425,17 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
   } else {
     int P = 128;
     while (P < n) P <<= 1;  // 128/256/512
-    const long grid = (d + WAVE - 1) / WAVE;
-    const size_t lds = (size_t)P * WAVE * sizeof(float);
+    const long grid = (d + LDS_COLS - 1) / LDS_COLS;
+    const size_t lds = (size_t)P * LDS_COLS * sizeof(float);
     if (mode == MEDIAN)
-      hipLaunchKernelGGL((colsel_lds_kernel<MEDIAN, T>), dim3(grid), dim3(WAVE),
-                         lds, stream, X, out, n, d, f, P);
+      hipLaunchKernelGGL((colsel_lds_kernel<MEDIAN, T>), dim3(grid),
+                         dim3(LDS_THREADS), lds, stream, X, out, n, d, f, P);
     else if (mode == TRIMMED)
-      hipLaunchKernelGGL((colsel_lds_kernel<TRIMMED, T>), dim3(grid), dim3(WAVE),
-                         lds, stream, X, out, n, d, f, P);
+      hipLaunchKernelGGL((colsel_lds_kernel<TRIMMED, T>), dim3(grid),
+                         dim3(LDS_THREADS), lds, stream, X, out, n, d, f, P);
     else
-      hipLaunchKernelGGL((colsel_lds_kernel<MEAMED, T>), dim3(grid), dim3(WAVE),
-                         lds, stream, X, out, n, d, f, P);
+      hipLaunchKernelGGL((colsel_lds_kernel<MEAMED, T>), dim3(grid),
+                         dim3(LDS_THREADS), lds, stream, X, out, n, d, f, P);
   }
 }
 
