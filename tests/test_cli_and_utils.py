@@ -119,3 +119,12 @@ def test_package_import_smoke():
 
     assert callable(byzpy_amd.run_operator)
     assert byzpy_amd.__version__
+
+
+def test_dependencies_probe():
+    from byzpy_amd._dependencies import CPU_DEPS, probe
+
+    info = probe()
+    assert "torch" in info["cpu_deps"]
+    assert "hip_extension" in info["gpu_stack"]
+    assert "torch" in CPU_DEPS
