@@ -113,13 +113,13 @@ def main():
         step()
     sync()
     elapsed = time.perf_counter() - t0
-    # max over ranks
-    el = torch.tensor([elapsed], dtype=torch.float64)
+    # max over ranks (RCCL needs a DEVICE tensor; gloo takes CPU)
+    el = torch.tensor([elapsed], dtype=torch.float64, device=device if use_cuda else "cpu")
     if pdist.is_initialized():
         import torch.distributed as dist
 
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
-    elapsed = float(el[0])
+    elapsed = float(el.item())
 
     aggs_per_step = n * (2 if args.op == "both" else 1)
     value = aggs_per_step * args.steps / elapsed
