@@ -40,10 +40,16 @@ class RcclParameterServer:
         aggregate_fn: Callable[[torch.Tensor], torch.Tensor],
         *,
         gather_result: bool = False,
+        overlap_chunks: int = 1,
     ) -> None:
         self.local_gradient_fns = list(local_gradient_fns)
         self.aggregate_fn = aggregate_fn
         self.gather_result = gather_result
+        # >1: split the shard into chunks and pipeline — chunk i aggregates
+        # on the compute stream while chunk i+1's all-to-all is in flight on
+        # RCCL's stream (coordinate-chunkable aggregators only: every
+        # coordinate-wise op qualifies; Krum-style global scoring does not).
+        self.overlap_chunks = max(1, int(overlap_chunks))
 
     def round(self) -> torch.Tensor:
         world = pdist.get_world_size()
@@ -62,13 +68,51 @@ class RcclParameterServer:
         # (n_local, d) -> world blocks of (n_local, shard) -> all-to-all ->
         # this rank holds every worker's shard: (n_global, shard)
         blocks = local.reshape(n_local, world, shard).transpose(0, 1).contiguous()
-        recv = pdist.all_to_all_rows(blocks.reshape(world * n_local, shard))
-        X_shard = recv.reshape(world, n_local, shard).reshape(world * n_local, shard)
-        out_shard = self.aggregate_fn(X_shard)  # (shard,)
+        if self.overlap_chunks <= 1:
+            recv = pdist.all_to_all_rows(blocks.reshape(world * n_local, shard))
+            X_shard = recv.reshape(world, n_local, shard).reshape(
+                world * n_local, shard
+            )
+            out_shard = self.aggregate_fn(X_shard)  # (shard,)
+        else:
+            out_shard = self._round_overlapped(blocks, world, n_local, shard)
         if not self.gather_result:
             return out_shard
         full = pdist.all_gather_rows(out_shard.reshape(1, -1)).reshape(-1)
         return full[:d]
+
+    def _round_overlapped(
+        self, blocks: torch.Tensor, world: int, n_local: int, shard: int
+    ) -> torch.Tensor:
+        """Chunked pipeline: issue chunk i+1's all-to-all (async, on RCCL's
+        own stream) while chunk i aggregates on the compute stream
+        (SURVEY.md §7 build step 4: side-stream overlap of C1 with the
+        aggregation kernels)."""
+        import torch.distributed as dist
+
+        C = min(self.overlap_chunks, shard) or 1
+        bounds = [(shard * c) // C for c in range(C + 1)]
+        recvs, works = [], []
+
+        def issue(c: int) -> None:
+            lo, hi = bounds[c], bounds[c + 1]
+            send = blocks[:, :, lo:hi].reshape(world * n_local, hi - lo).contiguous()
+            recv = torch.empty_like(send)
+            works.append(dist.all_to_all_single(recv, send, async_op=True))
+            recvs.append(recv)
+
+        issue(0)
+        outs = []
+        for c in range(C):
+            if c + 1 < C:
+                issue(c + 1)
+            works[c].wait()  # compute stream waits on the RCCL stream event
+            lo, hi = bounds[c], bounds[c + 1]
+            X_c = recvs[c].reshape(world, n_local, hi - lo).reshape(
+                world * n_local, hi - lo
+            )
+            outs.append(self.aggregate_fn(X_c))
+        return torch.cat(outs)
 
 
 def trimmed_mean_aggregate(f: int):

@@ -99,6 +99,12 @@ def _body_rccl_ps(rank):
     assert out.shape == (d,)
     # n=4 grads: {1, 5, 2, 1e6}; trimmed f=1 -> mean(2, 5) = 3.5
     assert torch.allclose(out, torch.full((d,), 3.5), atol=1e-4), out[:3]
+    # chunked comm/compute overlap pipeline gives the same result
+    ps2 = RcclParameterServer(
+        fns, trimmed_mean_aggregate(1), gather_result=True, overlap_chunks=3
+    )
+    out2 = ps2.round()
+    assert torch.allclose(out2, out, atol=1e-5)
 
 
 def _body_rccl_p2p(rank):
