@@ -42,7 +42,9 @@ def main() -> None:
         return -2.0 * torch.randn(d, generator=gen).to(device)
 
     fns = [honest, (signflip if rank < 3 else honest)]
-    ps = RcclParameterServer(fns, trimmed_mean_aggregate(f=3))
+    # overlap_chunks pipelines chunk i+1's all-to-all under chunk i's
+    # aggregation kernels (coordinate-wise ops are chunkable)
+    ps = RcclParameterServer(fns, trimmed_mean_aggregate(f=3), overlap_chunks=4)
 
     for _ in range(3):
         ps.round()  # warmup
