@@ -31,7 +31,6 @@ class _FeatureChunkedAggregator(Aggregator):
 
     def __init__(self, *, chunk_size: int | None = None) -> None:
         self.chunk_size = int(chunk_size or self.default_chunk_size)
-        self._pending = None  # (like, handles)
 
     def _chunk_fn(self):
         raise NotImplementedError
@@ -44,7 +43,7 @@ class _FeatureChunkedAggregator(Aggregator):
         ref, X, like, handles = self._matrix_ref(ctx, gradients)
         if X.is_cuda:
             return []  # GPU: single-kernel compute path
-        self._pending = (like, handles)
+        ctx.metadata["_op_pending"] = (like, handles)
         d = X.shape[1]
         chunk = select_adaptive_chunk_size(d, ctx.pool_size, self.chunk_size)
         fn = self._chunk_fn()
@@ -55,8 +54,7 @@ class _FeatureChunkedAggregator(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        like, handles = self._pending
-        self._pending = None
+        like, handles = ctx.metadata.pop("_op_pending")
         try:
             out = torch.cat([torch.as_tensor(r) for r in results])
             return to_like(out, like)

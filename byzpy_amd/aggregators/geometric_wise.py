@@ -36,7 +36,6 @@ class MultiKrum(Aggregator):
             raise ValueError("need f >= 0 and q >= 1")
         self.f, self.q = int(f), int(q)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.multi_krum(X, self.f, self.q)
@@ -46,7 +45,7 @@ class MultiKrum(Aggregator):
         ref, X, like, handles = self._matrix_ref(ctx, gradients)
         if X.is_cuda:
             return []
-        self._pending = (X, like, handles)
+        ctx.metadata["_op_pending"] = (X, like, handles)
         n = X.shape[0]
         chunk = max(1, min(self.chunk_size, n))
         return [
@@ -55,8 +54,7 @@ class MultiKrum(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             G = torch.cat(results, dim=0)
             norms = torch.diagonal(G)
@@ -84,8 +82,7 @@ class Krum(MultiKrum):
         return D.krum(X, self.f)
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             G = torch.cat(results, dim=0)
             norms = torch.diagonal(G)
@@ -186,7 +183,6 @@ class MinimumDiameterAveraging(Aggregator):
             raise ValueError("f must be >= 0")
         self.f = int(f)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.minimum_diameter_averaging(X, self.f)
@@ -206,7 +202,7 @@ class MinimumDiameterAveraging(Aggregator):
             handles.append(ref)
         else:
             ref = D2
-        self._pending = (X, like, handles)
+        ctx.metadata["_op_pending"] = (X, like, handles)
         combos = list(itertools.combinations(range(n), m))
         batch = max(1, self.chunk_size)
         return [
@@ -215,8 +211,7 @@ class MinimumDiameterAveraging(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             # min diameter, then lexicographic subset (canonical tie-break)
             best_diam, best = min(results, key=lambda r: (r[0], r[1]))
@@ -240,7 +235,6 @@ class MoNNA(Aggregator):
         self.f = int(f)
         self.reference_index = int(reference_index)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.monna(X, self.f, self.reference_index)
@@ -250,7 +244,7 @@ class MoNNA(Aggregator):
         ref, X, like, handles = self._matrix_ref(ctx, gradients)
         if X.is_cuda:
             return []
-        self._pending = (X, like, handles)
+        ctx.metadata["_op_pending"] = (X, like, handles)
         n = X.shape[0]
         chunk = max(1, min(self.chunk_size, n))
         return [
@@ -261,8 +255,7 @@ class MoNNA(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             d2 = torch.cat(results)
             d2[self.reference_index] = -1.0  # reference-first tiebreak
@@ -287,7 +280,6 @@ class SMEA(Aggregator):
             raise ValueError("f must be >= 0")
         self.f = int(f)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.smea(X, self.f)
@@ -308,7 +300,7 @@ class SMEA(Aggregator):
             handles.append(ref)
         else:
             ref = G
-        self._pending = (X, like, handles)
+        ctx.metadata["_op_pending"] = (X, like, handles)
         combos = list(itertools.combinations(range(n), m))
         batch = max(1, self.chunk_size)
         return [
@@ -317,8 +309,7 @@ class SMEA(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             best_ev, best = min(results, key=lambda r: r[0])
             idx = torch.tensor(best, dtype=torch.long)

@@ -107,7 +107,6 @@ class ComparativeGradientElimination(Aggregator):
             raise ValueError("f must be >= 0")
         self.f = int(f)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.cge(X, self.f)
@@ -117,7 +116,7 @@ class ComparativeGradientElimination(Aggregator):
         ref, X, like, handles = self._matrix_ref(ctx, gradients)
         if X.is_cuda:
             return []
-        self._pending = (X, like, handles)
+        ctx.metadata["_op_pending"] = (X, like, handles)
         d = X.shape[1]
         chunk = select_adaptive_chunk_size(d, ctx.pool_size, self.chunk_size)
         return [
@@ -126,8 +125,7 @@ class ComparativeGradientElimination(Aggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        X, like, handles = self._pending
-        self._pending = None
+        X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             norms = sum(results)
             k = X.shape[0] - self.f

@@ -265,8 +265,6 @@ def monna(X: torch.Tensor, f: int, reference_index: int = 0) -> torch.Tensor:
     n = X.shape[0]
     k = n - f
     if _gpu(X):
-        Xf = X
-        ref = X[reference_index : reference_index + 1]
         # ||x - r||^2 = ||x||^2 + ||r||^2 - 2 x.r ; reuse the norm kernel
         norms = row_sqnorms(X)
         dots = (X.float() @ X.float()[reference_index]).flatten()

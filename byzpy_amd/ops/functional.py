@@ -31,10 +31,6 @@ import torch
 # ---------------------------------------------------------------------------
 
 
-def _as_float(x: torch.Tensor) -> torch.Tensor:
-    return x if x.is_floating_point() else x.float()
-
-
 def pairwise_sq_dists(X: torch.Tensor) -> torch.Tensor:
     """Full n x n matrix of squared euclidean distances via the Gram trick
     ``||a||^2 + ||b||^2 - 2 a.b`` with f32 accumulation."""

@@ -33,7 +33,6 @@ class Clipping(PreAggregator):
             raise ValueError("threshold must be >= 0")
         self.threshold = float(threshold)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _pre_aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.clip_rows(X, self.threshold)
@@ -43,7 +42,7 @@ class Clipping(PreAggregator):
         ref, X, like, handles = self._matrix_ref(ctx, vectors)
         if X.is_cuda:
             return []
-        self._pending = (like, handles)
+        ctx.metadata["_op_pending"] = (like, handles)
         n = X.shape[0]
         chunk = max(1, min(self.chunk_size, n))
         return [
@@ -52,8 +51,7 @@ class Clipping(PreAggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        like, handles = self._pending
-        self._pending = None
+        like, handles = ctx.metadata.pop("_op_pending")
         try:
             out = torch.cat(results, dim=0)
             return [to_like(row, like) for row in out]
@@ -82,7 +80,6 @@ class Bucketing(PreAggregator):
         self.feature_chunk_size = int(feature_chunk_size)
         self.perm = None if perm is None else [int(i) for i in perm]
         self.rng = rng or random.Random()
-        self._pending = None
 
     def _draw_perm(self, n: int) -> List[int]:
         if self.perm is not None:
@@ -99,7 +96,7 @@ class Bucketing(PreAggregator):
         ref, X, like, handles = self._matrix_ref(ctx, vectors)
         if X.is_cuda:
             return []
-        self._pending = (like, handles)
+        ctx.metadata["_op_pending"] = (like, handles)
         perm = self._draw_perm(X.shape[0])
         n = X.shape[0]
         buckets = [
@@ -108,8 +105,7 @@ class Bucketing(PreAggregator):
         return [SubTask(fn=SF.bucket_mean_chunk, args=(ref, b)) for b in buckets]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        like, handles = self._pending
-        self._pending = None
+        like, handles = ctx.metadata.pop("_op_pending")
         try:
             return [to_like(r, like) for r in results]
         finally:
@@ -130,7 +126,6 @@ class NearestNeighborMixing(PreAggregator):
             raise ValueError("f must be >= 0")
         self.f = int(f)
         self.feature_chunk_size = int(feature_chunk_size)
-        self._pending = None
 
     def _pre_aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.nnm(X, self.f)
@@ -140,7 +135,7 @@ class NearestNeighborMixing(PreAggregator):
         ref, X, like, handles = self._matrix_ref(ctx, vectors)
         if X.is_cuda:
             return []
-        self._pending = (ref, X, like, handles)
+        ctx.metadata["_op_pending"] = (ref, X, like, handles)
         d = X.shape[1]
         chunk = select_adaptive_chunk_size(d, ctx.pool_size, self.feature_chunk_size)
         return [
@@ -149,8 +144,7 @@ class NearestNeighborMixing(PreAggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        ref, X, like, handles = self._pending
-        self._pending = None
+        ref, X, like, handles = ctx.metadata.pop("_op_pending")
         try:
             G = sum(results)
             norms = torch.diagonal(G)
@@ -177,7 +171,6 @@ class ARC(PreAggregator):
             raise ValueError("f must be >= 0")
         self.f = int(f)
         self.chunk_size = int(chunk_size)
-        self._pending = None
 
     def _pre_aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.arc_clip(X, self.f)
@@ -196,7 +189,7 @@ class ARC(PreAggregator):
             order = torch.argsort(norms, descending=True)
             threshold = norms[order[k]]
             scale = torch.clamp(threshold / norms.clamp_min(1e-20), max=1.0)
-        self._pending = (like, handles)
+        ctx.metadata["_op_pending"] = (like, handles)
         chunk = max(1, min(self.chunk_size, n))
         return [
             SubTask(
@@ -206,8 +199,7 @@ class ARC(PreAggregator):
         ]
 
     def reduce_subtasks(self, ctx: OpContext, results: List[Any], **inputs: Any) -> Any:
-        like, handles = self._pending
-        self._pending = None
+        like, handles = ctx.metadata.pop("_op_pending")
         try:
             out = torch.cat(results, dim=0)
             return [to_like(row, like) for row in out]
