@@ -117,3 +117,35 @@ def test_bf16_inputs_roundtrip(data):
     out = CoordinateWiseMedian().aggregate(grads)
     assert out.dtype == torch.bfloat16
     assert out.shape == (257,)
+
+
+def test_operator_instance_reentrant_across_graph_runs(data):
+    """One operator instance used by two concurrent graph runs must not
+    clobber per-run subtask state (it lives in OpContext, not self)."""
+    import asyncio as aio
+
+    from byzpy_amd.graph.executor import OperatorExecutor
+
+    agg = CoordinateWiseMedian(chunk_size=32)
+    grads_a = list(data)
+    grads_b = [g * 3 for g in data]
+
+    async def main():
+        pool = ActorPool(ActorPoolConfig(backend="thread", count=4))
+        await pool.start()
+        ex1 = OperatorExecutor(agg, pool=pool)
+        ex2 = OperatorExecutor(agg, pool=pool)
+        outs = await aio.gather(
+            ex1.run({"gradients": grads_a}),
+            ex2.run({"gradients": grads_b}),
+            ex1.run({"gradients": grads_a}),
+        )
+        await pool.close()
+        return outs
+
+    a1, b, a2 = _run(main())
+    ref_a = agg.aggregate(grads_a)
+    ref_b = agg.aggregate(grads_b)
+    assert torch.allclose(a1, ref_a, atol=1e-5)
+    assert torch.allclose(a2, ref_a, atol=1e-5)
+    assert torch.allclose(b, ref_b, atol=1e-5)
