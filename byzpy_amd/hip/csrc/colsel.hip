@@ -15,7 +15,6 @@
 // Loads are coalesced: adjacent lanes read adjacent columns, so each
 // row-iteration is one 256 B (f32) / 128 B (bf16) wave transaction.
 #include "common.h"
-#include <cstdlib>
 
 namespace {
 
@@ -312,104 +311,6 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
-// EXPERIMENTAL LDS-tiled packed median (bf16, n <= 64): a 256-thread block
-// stages a [64 rows][256 pair-cols] tile cooperatively — 1 KB contiguous
-// per row per pass instead of 256 B per wave — then each thread sorts its
-// pair-column from LDS. Double-buffered T14-style (issue tile t+1's loads
-// to registers, sort tile t, write t+1 after the barrier) so the HBM
-// stream hides under the sorting VALU. Selected by BYZPY_AMD_MEDIAN_LDS=1
-// (A/B against the register-walk kernel).
-// ---------------------------------------------------------------------------
-
-template <int P>
-__global__ void __launch_bounds__(256, 1)
-colsel_pk_median_lds(const unsigned short* __restrict__ X,
-                     unsigned short* __restrict__ out, int n, long d, int f) {
-  constexpr int TPAIRS = 256;               // pair-columns per tile
-  __shared__ u32 tile[2][64 * TPAIRS];      // 2 x 64 KB... (64*256*4 = 64 KB)
-  const int t = threadIdx.x;
-  const long npairs = d >> 1;
-  const long rowstride = d >> 1;            // u32 units
-  const long tiles = (npairs + TPAIRS - 1) / TPAIRS;
-  const long tiles_per_block = (tiles + gridDim.x - 1) / gridDim.x;
-  const long t_lo = (long)blockIdx.x * tiles_per_block;
-  const long t_hi = min(tiles, t_lo + tiles_per_block);
-  if (t_lo >= t_hi) return;
-  const int n_walk = vecify(n);
-
-  // stage geometry: 64 rows x 256 pairs = 64 rows x 64 16B-chunks;
-  // pass p of 16: chunk id = p*256 + t -> row = id/64, c16 = id%64
-  u32 stage_regs[4][16];  // 16 passes x 16 B = 4 u32 each... (4 per pass)
-  auto stage_load = [&](long tile_idx) {
-    const long pair0 = tile_idx * TPAIRS;
-#pragma unroll
-    for (int p = 0; p < 16; ++p) {
-      const int id = p * 256 + t;
-      const int row = id >> 6;        // 64 chunks per row
-      const int c16 = id & 63;
-      const u32* src = reinterpret_cast<const u32*>(X) +
-                       (long)min(row, n - 1) * rowstride + pair0 + c16 * 4;
-      // clamp the tail tile: stay in bounds, pad later via key
-      const long last_ok = npairs - 4;
-      const long off = min((long)(pair0 + c16 * 4), last_ok < 0 ? 0 : last_ok);
-      const u32* safe = reinterpret_cast<const u32*>(X) +
-                        (long)min(row, n - 1) * rowstride + off;
-      const uint4 w = *reinterpret_cast<const uint4*>(safe);
-      stage_regs[0][p] = w.x; stage_regs[1][p] = w.y;
-      stage_regs[2][p] = w.z; stage_regs[3][p] = w.w;
-      __builtin_amdgcn_sched_barrier(0);
-    }
-  };
-  auto stage_write = [&](int buf) {
-#pragma unroll
-    for (int p = 0; p < 16; ++p) {
-      const int id = p * 256 + t;
-      const int row = id >> 6;
-      const int c16 = id & 63;
-      u32* dst = &tile[buf][row * TPAIRS + c16 * 4];
-      dst[0] = stage_regs[0][p]; dst[1] = stage_regs[1][p];
-      dst[2] = stage_regs[2][p]; dst[3] = stage_regs[3][p];
-    }
-  };
-
-  stage_load(t_lo);
-  stage_write(0);
-  __syncthreads();
-  for (long ti = t_lo; ti < t_hi; ++ti) {
-    if (ti + 1 < t_hi) stage_load(ti + 1);
-    const u32* T = tile[(ti - t_lo) & 1];
-    const long pair = ti * TPAIRS + t;
-    if (pair < npairs) {
-      u32 v[P];
-#pragma unroll
-      for (int i = 0; i < P; ++i)
-        v[i] = pk_key_from_bf16(T[i * TPAIRS + t]);
-#pragma unroll
-      for (int i = 0; i < P; ++i)
-        if (i >= n_walk) v[i] = 0xFFFFFFFFu;
-      bitonic_sort_pk<P>(v);
-      const u32 lo = extract_at_pk<P>(v, vecify((n - 1) >> 1));
-      const u32 hi = extract_at_pk<P>(v, vecify(n >> 1));
-      const float m0 =
-          0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
-      const float m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
-      union { unsigned short s[2]; u32 w; } o;
-      union { unsigned short s; __hip_bfloat16 h; } c0, c1;
-      c0.h = __float2bfloat16(m0);
-      c1.h = __float2bfloat16(m1);
-      o.s[0] = c0.s;
-      o.s[1] = c1.s;
-      reinterpret_cast<u32*>(out)[pair] = o.w;
-    }
-    __syncthreads();
-    if (ti + 1 < t_hi) {
-      stage_write((ti + 1 - t_lo) & 1);
-      __syncthreads();
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
 // LDS variant, 64 < n <= 512: block-COOPERATIVE batched bitonic. One block
 // (16 waves) owns 64 columns staged as an LDS plane [P][64]; every
 // compare-exchange step spreads its P/2 x 64 sites over all 1024 threads
@@ -545,21 +446,6 @@ void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,
 
 void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
-  if (mode == MEDIAN && n <= 64 && (d % 8) == 0 && getenv("BYZPY_AMD_MEDIAN_LDS")) {
-    const long npairs = d >> 1;
-    const long tiles = (npairs + 255) / 256;
-    const int grid = (int)(tiles < 1024 ? tiles : 1024);
-#define PK_LDS_LAUNCH(P)                                                      \
-  hipLaunchKernelGGL((colsel_pk_median_lds<P>), dim3(grid), dim3(256), 0,     \
-                     stream, reinterpret_cast<const unsigned short*>(X),      \
-                     reinterpret_cast<unsigned short*>(out), n, d, f)
-    if (n <= 8) PK_LDS_LAUNCH(8);
-    else if (n <= 16) PK_LDS_LAUNCH(16);
-    else if (n <= 32) PK_LDS_LAUNCH(32);
-    else PK_LDS_LAUNCH(64);
-#undef PK_LDS_LAUNCH
-    return;
-  }
   if (mode == MEDIAN && n <= 64 && (d % 2) == 0) {
     const int block = 256;
     // A/B'd on MI355X: QUADS (8 B/lane) ties the single-pair variant
