@@ -42,7 +42,7 @@ def main():
     rows = []
 
     def add(name, secs, bytes_gb, extra=""):
-        rows.append((name, secs * 1e3, bytes_gb / secs / 1e3, extra))
+        rows.append((name, secs * 1e3, bytes_gb / secs, extra))
 
     add("median (packed-key bitonic, bf16)", timeit(lambda: D.median(X)), gb)
     add("trimmed_mean f=16 (bf16)", timeit(lambda: D.trimmed_mean(X, 16)), gb)

@@ -237,7 +237,9 @@ DEV u32 extract_at_pk(const u32 (&v)[P], int pos) {
 // QUADS=true: each thread owns TWO adjacent packed pairs (4 columns,
 // 8 B/lane row reads — wider DRAM bursts; ~150 VGPRs at P=64) with two
 // independent sorting networks; QUADS=false: one pair (4 B/lane).
-template <int P, bool QUADS>
+// MODE: MEDIAN or TRIMMED (order statistics read straight off the sorted
+// keys; TRIMMED unpacks the kept range and sums in f32).
+template <int P, bool QUADS, int MODE = MEDIAN>
 __global__ void
 // QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
 // so the allocator doesn't cap at 128 and spill
@@ -286,19 +288,33 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
     bitonic_sort_pk<P>(v);
     if (QUADS) bitonic_sort_pk<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]));
     const int plo = vecify((n - 1) >> 1), phi = vecify(n >> 1);
+    const int fv = vecify(f), nfv = vecify(n - f);
     u32* outw = reinterpret_cast<u32*>(out);
 #pragma unroll
     for (int half = 0; half < (QUADS ? 2 : 1); ++half) {
       const u32* arr = half ? v2 : v;
-      u32 lo = 0, hi = 0;
+      float m0, m1;
+      if (MODE == MEDIAN) {
+        u32 lo = 0, hi = 0;
 #pragma unroll
-      for (int i = 0; i < P; ++i) {
-        if (i == plo) lo = arr[i];
-        if (i == phi) hi = arr[i];
+        for (int i = 0; i < P; ++i) {
+          if (i == plo) lo = arr[i];
+          if (i == phi) hi = arr[i];
+        }
+        m0 = 0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
+        m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
+      } else {  // TRIMMED: unpack the kept range of the sorted keys and sum
+        float s0 = 0.0f, s1 = 0.0f;
+#pragma unroll
+        for (int i = 0; i < P; ++i)
+          if (i >= fv && i < nfv) {
+            s0 += key_to_float(arr[i] & 0xFFFFu);
+            s1 += key_to_float(arr[i] >> 16);
+          }
+        const float inv = 1.0f / (float)(n - 2 * f);
+        m0 = s0 * inv;
+        m1 = s1 * inv;
       }
-      const float m0 =
-          0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
-      const float m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
       union { unsigned short s[2]; u32 w; } o;
       union { unsigned short s; __hip_bfloat16 h; } c0, c1;
       c0.h = __float2bfloat16(m0);
@@ -446,25 +462,24 @@ void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,
 
 void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
-  if (mode == MEDIAN && n <= 64 && (d % 2) == 0) {
+  if ((mode == MEDIAN || mode == TRIMMED) && n <= 64 && (d % 2) == 0) {
     const int block = 256;
     // A/B'd on MI355X: QUADS (8 B/lane) ties the single-pair variant
     // (3.9 vs 3.8 ms at 64 x 125M) — the occupancy drop offsets the wider
     // bursts. Keep single-pair as the shipping path.
-    const bool quads = false;
-    const long units = quads ? (d >> 2) : (d >> 1);
+    const long units = d >> 1;
     const long want = (units + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
     const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
     unsigned short* Ou = reinterpret_cast<unsigned short*>(out);
 #define PK_LAUNCH(P)                                                          \
   do {                                                                        \
-    if (quads)                                                                \
-      hipLaunchKernelGGL((colsel_pk_median_bf16<P, true>), dim3(grid),        \
-                         dim3(block), 0, stream, Xu, Ou, n, d, f);            \
+    if (mode == MEDIAN)                                                       \
+      hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, MEDIAN>),           \
+                         dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
     else                                                                      \
-      hipLaunchKernelGGL((colsel_pk_median_bf16<P, false>), dim3(grid),       \
-                         dim3(block), 0, stream, Xu, Ou, n, d, f);            \
+      hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, TRIMMED>),          \
+                         dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
   } while (0)
     if (n <= 8) PK_LAUNCH(8);
     else if (n <= 16) PK_LAUNCH(16);
