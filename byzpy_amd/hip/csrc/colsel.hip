@@ -243,7 +243,7 @@ template <int P, bool QUADS, int MODE = MEDIAN>
 __global__ void
 // QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
 // so the allocator doesn't cap at 128 and spill
-__launch_bounds__(256, (QUADS && P >= 64) ? 3 : 4)
+__launch_bounds__(256, ((QUADS || MODE == 2) && P >= 64) ? 3 : 4)
 colsel_pk_median_bf16(const unsigned short* __restrict__ X,
                                       unsigned short* __restrict__ out, int n,
                                       long d, int f) {
@@ -303,7 +303,8 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
         }
         m0 = 0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
         m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
-      } else {  // TRIMMED: unpack the kept range of the sorted keys and sum
+      } else if (MODE == TRIMMED) {
+        // unpack the kept range of the sorted keys and sum
         float s0 = 0.0f, s1 = 0.0f;
 #pragma unroll
         for (int i = 0; i < P; ++i)
@@ -314,6 +315,43 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
         const float inv = 1.0f / (float)(n - 2 * f);
         m0 = s0 * inv;
         m1 = s1 * inv;
+      } else {  // MEAMED: per half, shrink the sorted window [l, r) from
+        // whichever end deviates more from the median; ends are fetched by
+        // predicated extract (registers cannot be indexed dynamically)
+        u32 mlo = 0, mhi = 0;
+#pragma unroll
+        for (int i = 0; i < P; ++i) {
+          if (i == plo) mlo = arr[i];
+          if (i == phi) mhi = arr[i];
+        }
+#pragma unroll
+        for (int half2 = 0; half2 < 2; ++half2) {
+          const int sh = half2 * 16;
+          const float med =
+              0.5f * (key_to_float((mlo >> sh) & 0xFFFFu) +
+                      key_to_float((mhi >> sh) & 0xFFFFu));
+          // running totals: start with the full first-n sum, subtract the
+          // dropped end each step
+          float total = 0.0f;
+#pragma unroll
+          for (int i = 0; i < P; ++i)
+            if (i < nv) total += key_to_float((arr[i] >> sh) & 0xFFFFu);
+          int l = vecify(0), r = nv;
+          for (int kdrop = 0; kdrop < f; ++kdrop) {
+            float vl = 0.0f, vr = 0.0f;
+#pragma unroll
+            for (int i = 0; i < P; ++i) {
+              if (i == l) vl = key_to_float((arr[i] >> sh) & 0xFFFFu);
+              if (i == r - 1) vr = key_to_float((arr[i] >> sh) & 0xFFFFu);
+            }
+            const bool drop_left = (med - vl) > (vr - med);
+            total -= drop_left ? vl : vr;
+            l += drop_left ? 1 : 0;
+            r -= drop_left ? 0 : 1;
+          }
+          const float m = total / (float)(n - f);
+          if (half2 == 0) m0 = m; else m1 = m;
+        }
       }
       union { unsigned short s[2]; u32 w; } o;
       union { unsigned short s; __hip_bfloat16 h; } c0, c1;
@@ -462,7 +500,7 @@ void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,
 
 void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
-  if ((mode == MEDIAN || mode == TRIMMED) && n <= 64 && (d % 2) == 0) {
+  if (n <= 64 && (d % 2) == 0) {
     const int block = 256;
     // A/B'd on MI355X: QUADS (8 B/lane) ties the single-pair variant
     // (3.9 vs 3.8 ms at 64 x 125M) — the occupancy drop offsets the wider
@@ -477,8 +515,11 @@ void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
     if (mode == MEDIAN)                                                       \
       hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, MEDIAN>),           \
                          dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
-    else                                                                      \
+    else if (mode == TRIMMED)                                                 \
       hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, TRIMMED>),          \
+                         dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
+    else                                                                      \
+      hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, MEAMED>),           \
                          dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
   } while (0)
     if (n <= 8) PK_LAUNCH(8);
