@@ -334,11 +334,10 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
 
 typedef __attribute__((ext_vector_type(4))) float f32x4v;
 
-template <bool DIAG>
+template <bool DIAG, int BKF = 64>
 __global__ void __launch_bounds__(1024, 2)
 gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
                     int n, long d, long k_per_block) {
-  constexpr int BKF = 64;                        // k elements per chunk
   constexpr int CHUNK_BYTES = TILE * BKF * 4;    // 16 KB
   __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
   char* const bufA0 = smem;
@@ -357,49 +356,59 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
   const long k_hi = min(d, k_lo + k_per_block);
   const long nchunks = (k_hi - k_lo + BKF - 1) / BKF;
 
-  // stage: thread t owns LDS (row = t>>4, slot = t&15); content of that
-  // slot is global slot (slot ^ (row & 7)) — both-sides swizzle
+  constexpr int SLOTSF = BKF / 4;   // 16-B slots per row
+  constexpr int SUBF = SLOTSF / 16; // stage passes per thread
+  // stage: thread t owns LDS rows (t>>4) at slots (t&15) + u*16; content
+  // of slot s is global slot (s ^ (row & 7)) — both-sides swizzle
   const int st_row = t >> 4;
-  const int st_slot = t & 15;
-  const int src_slot = st_slot ^ (st_row & 7);
   const int a_rows = min(TILE, n - row_base);
   const int b_rows = min(TILE, n - col_base);
-  const float* a_src =
-      X + (long)(row_base + min(st_row, a_rows - 1)) * d + (long)src_slot * 4;
-  const float* b_src =
-      X + (long)(col_base + min(st_row, b_rows - 1)) * d + (long)src_slot * 4;
+  int st_slot[SUBF], src_slot[SUBF];
+#pragma unroll
+  for (int u = 0; u < SUBF; ++u) {
+    st_slot[u] = (t & 15) + u * 16;
+    src_slot[u] = st_slot[u] ^ (st_row & 7);
+  }
+  const float* a_row_src = X + (long)(row_base + min(st_row, a_rows - 1)) * d;
+  const float* b_row_src = X + (long)(col_base + min(st_row, b_rows - 1)) * d;
 
-  f32x4v ra, rb;
+  f32x4v ra[SUBF], rb[SUBF];
   auto stage_load = [&](long c) {
     const long k0 = k_lo + c * BKF;
     const bool full = (k0 + BKF <= k_hi);
-    if (full && a_rows == TILE) {
-      ra = *reinterpret_cast<const f32x4v*>(a_src + k0);
-    } else {
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const long k = k0 + (long)src_slot * 4 + j;
-        ra[j] = (st_row < a_rows && k < k_hi) ? a_src[k0 + j] : 0.0f;
-      }
-    }
-    if (!DIAG) {
-      if (full && b_rows == TILE) {
-        rb = *reinterpret_cast<const f32x4v*>(b_src + k0);
+    for (int u = 0; u < SUBF; ++u) {
+      const long off = k0 + (long)src_slot[u] * 4;
+      if (full && a_rows == TILE) {
+        ra[u] = *reinterpret_cast<const f32x4v*>(a_row_src + off);
       } else {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const long k = k0 + (long)src_slot * 4 + j;
-          rb[j] = (st_row < b_rows && k < k_hi) ? b_src[k0 + j] : 0.0f;
+        for (int j = 0; j < 4; ++j)
+          ra[u][j] =
+              (st_row < a_rows && off + j < k_hi) ? a_row_src[off + j] : 0.0f;
+      }
+      if (!DIAG) {
+        if (full && b_rows == TILE) {
+          rb[u] = *reinterpret_cast<const f32x4v*>(b_row_src + off);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            rb[u][j] =
+                (st_row < b_rows && off + j < k_hi) ? b_row_src[off + j] : 0.0f;
         }
       }
     }
   };
   auto stage_write = [&](int which) {
     char* a = which ? bufA1 : bufA0;
-    *reinterpret_cast<f32x4v*>(a + st_row * 256 + st_slot * 16) = ra;
-    if (!DIAG) {
-      char* b = which ? bufB1 : bufB0;
-      *reinterpret_cast<f32x4v*>(b + st_row * 256 + st_slot * 16) = rb;
+    char* b = which ? bufB1 : bufB0;
+#pragma unroll
+    for (int u = 0; u < SUBF; ++u) {
+      *reinterpret_cast<f32x4v*>(a + st_row * (SLOTSF * 16) + st_slot[u] * 16) =
+          ra[u];
+      if (!DIAG)
+        *reinterpret_cast<f32x4v*>(b + st_row * (SLOTSF * 16) +
+                                   st_slot[u] * 16) = rb[u];
     }
   };
 
@@ -419,13 +428,15 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
     const char* A = (c & 1) ? bufA1 : bufA0;
     const char* B = (c & 1) ? bufB1 : bufB0;
 #pragma unroll
-    for (int step = 0; step < 16; ++step) {
-      const int k = step * 4 + kgrp;  // element column 0..63
+    for (int step = 0; step < BKF / 4; ++step) {
+      const int k = step * 4 + kgrp;  // element column within the chunk
       // element k lives in slot (k>>2), swizzled by row
       const int sa = ((k >> 2) ^ (rowA & 7)) * 16 + (k & 3) * 4;
       const int sb = ((k >> 2) ^ (rowB & 7)) * 16 + (k & 3) * 4;
-      const float a = *reinterpret_cast<const float*>(A + rowA * 256 + sa);
-      const float b = *reinterpret_cast<const float*>(B + rowB * 256 + sb);
+      const float a =
+          *reinterpret_cast<const float*>(A + rowA * (SLOTSF * 16) + sa);
+      const float b =
+          *reinterpret_cast<const float*>(B + rowB * (SLOTSF * 16) + sb);
       if (step & 1)
         acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc1, 0, 0, 0);
       else
@@ -497,13 +508,19 @@ void launch_gram_f32(const float* X, float* G, int n, long d,
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
   dim3 grid(splitk, tiles, tiles);
+  if ((d % 4) == 0 && (kpb % 128) == 0 && tiles == 1) {
+    // single tile: BKF=128 halves the barrier count (2 x 32 KB buffers)
+    hipLaunchKernelGGL((gram_f32_lds_kernel<true, 128>), grid,
+                       dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+    return;
+  }
   if ((d % 4) == 0 && (kpb % 64) == 0) {
     if (tiles == 1)
-      hipLaunchKernelGGL((gram_f32_lds_kernel<true>), grid, dim3(WAVES * 64),
-                         0, stream, X, G, n, d, kpb);
+      hipLaunchKernelGGL((gram_f32_lds_kernel<true, 64>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
     else
-      hipLaunchKernelGGL((gram_f32_lds_kernel<false>), grid, dim3(WAVES * 64),
-                         0, stream, X, G, n, d, kpb);
+      hipLaunchKernelGGL((gram_f32_lds_kernel<false, 64>), grid,
+                         dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
     return;
   }
   hipLaunchKernelGGL(gram_f32_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
