@@ -243,7 +243,7 @@ template <int P, bool QUADS, int MODE = MEDIAN>
 __global__ void
 // QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
 // so the allocator doesn't cap at 128 and spill
-__launch_bounds__(256, ((QUADS || MODE == 2) && P >= 64) ? 3 : 4)
+__launch_bounds__(256, (QUADS && P >= 64) ? 3 : 4)
 colsel_pk_median_bf16(const unsigned short* __restrict__ X,
                                       unsigned short* __restrict__ out, int n,
                                       long d, int f) {
@@ -316,34 +316,47 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
         m0 = s0 * inv;
         m1 = s1 * inv;
       } else {  // MEAMED: per half, shrink the sorted window [l, r) from
-        // whichever end deviates more from the median; ends are fetched by
-        // predicated extract (registers cannot be indexed dynamically)
+        // whichever end deviates more from the median. The walk only ever
+        // touches positions [0, f] and [n-f-1, n), so those 2(f+1) sorted
+        // keys are staged into a per-thread LDS strip and each step reads
+        // its two candidate ends by dynamic ds_read in O(1) — the previous
+        // O(P) predicated extract per step cost ~f*P*4 VALU ops per column
+        // pair and left the kernel 4x slower than MEDIAN.
+        extern __shared__ __attribute__((aligned(16))) u32 pk_lds[];
+        const int stride_l = 2 * (f + 1) + 1;  // odd => spread banks
+        u32* strip = pk_lds + (int)threadIdx.x * stride_l;
+        const int roff = vecify(2 * f + 2 - n);  // maps [n-f-1, n) -> [f+1, 2f+2)
+#pragma unroll
+        for (int i = 0; i < P; ++i) {
+          if (i <= fv) strip[i] = arr[i];
+          if (i >= nv - fv - 1 && i < nv) strip[i + roff] = arr[i];
+        }
         u32 mlo = 0, mhi = 0;
 #pragma unroll
         for (int i = 0; i < P; ++i) {
           if (i == plo) mlo = arr[i];
           if (i == phi) mhi = arr[i];
         }
+        float t0 = 0.0f, t1 = 0.0f;
+#pragma unroll
+        for (int i = 0; i < P; ++i)
+          if (i < nv) {
+            t0 += key_to_float(arr[i] & 0xFFFFu);
+            t1 += key_to_float(arr[i] >> 16);
+          }
 #pragma unroll
         for (int half2 = 0; half2 < 2; ++half2) {
           const int sh = half2 * 16;
           const float med =
               0.5f * (key_to_float((mlo >> sh) & 0xFFFFu) +
                       key_to_float((mhi >> sh) & 0xFFFFu));
-          // running totals: start with the full first-n sum, subtract the
-          // dropped end each step
-          float total = 0.0f;
-#pragma unroll
-          for (int i = 0; i < P; ++i)
-            if (i < nv) total += key_to_float((arr[i] >> sh) & 0xFFFFu);
+          // running total: full first-n sum minus the dropped end each step
+          float total = half2 ? t1 : t0;
           int l = vecify(0), r = nv;
           for (int kdrop = 0; kdrop < f; ++kdrop) {
-            float vl = 0.0f, vr = 0.0f;
-#pragma unroll
-            for (int i = 0; i < P; ++i) {
-              if (i == l) vl = key_to_float((arr[i] >> sh) & 0xFFFFu);
-              if (i == r - 1) vr = key_to_float((arr[i] >> sh) & 0xFFFFu);
-            }
+            const float vl = key_to_float((strip[l] >> sh) & 0xFFFFu);
+            const float vr =
+                key_to_float((strip[r - 1 + roff] >> sh) & 0xFFFFu);
             const bool drop_left = (med - vl) > (vr - med);
             total -= drop_left ? vl : vr;
             l += drop_left ? 1 : 0;
@@ -520,7 +533,9 @@ void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                          dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
     else                                                                      \
       hipLaunchKernelGGL((colsel_pk_median_bf16<P, false, MEAMED>),           \
-                         dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, f);\
+                         dim3(grid), dim3(block),                             \
+                         (size_t)block * (2 * (f + 1) + 1) * sizeof(u32),     \
+                         stream, Xu, Ou, n, d, f);                            \
   } while (0)
     if (n <= 8) PK_LAUNCH(8);
     else if (n <= 16) PK_LAUNCH(16);
