@@ -128,6 +128,8 @@ def geometric_median(
 ) -> torch.Tensor:
     """Weiszfeld fixed point: z <- sum(x_i/d_i) / sum(1/d_i), d_i clamped
     >= eps; stops when ||dz|| <= tol."""
+    if init not in ("median", "mean"):
+        raise ValueError("init must be 'median' or 'mean'")
     Xf = X.float()
     z = median(Xf) if init == "median" else Xf.mean(dim=0)
     z = z.float()
