@@ -78,6 +78,23 @@ def _body_sharded_ops(rank):
     nn_ = sharded.nnm(Xs, 2)
     assert torch.allclose(nn_, F.nnm(X, 2)[:, lo:hi], atol=1e-4)
 
+    tm = sharded.trimmed_mean(Xs, 2)
+    assert torch.allclose(tm, F.trimmed_mean(X, 2)[lo:hi], atol=1e-5)
+
+    mm = sharded.mean_of_medians(Xs, 2)
+    assert torch.allclose(mm, F.mean_of_medians(X, 2)[lo:hi], atol=1e-5)
+
+    kr = sharded.krum(Xs, 2)
+    assert torch.allclose(kr, F.krum(X, 2)[lo:hi], atol=1e-4)
+
+    ar = sharded.arc_clip(Xs, 2)
+    assert torch.allclose(ar, F.arc_clip(X, 2)[:, lo:hi], atol=1e-4)
+
+    bk = sharded.bucketing(Xs, 3, list(range(10)))
+    assert torch.allclose(
+        bk, F.bucketing(X, 3, perm=list(range(10)))[:, lo:hi], atol=1e-5
+    )
+
 
 def _body_rccl_ps(rank):
     from byzpy_amd.engine.parameter_server.rccl import (
