@@ -35,6 +35,12 @@ void launch_weiszfeld_update(const T*, const float*, const float*, float*,
 template <typename T>
 void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
+template <typename T>
+void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
+                       hipStream_t);
+template <typename T>
+void launch_caf_colsum(const T*, const float*, const float*, const float*,
+                       float*, int, long, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
 void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
@@ -425,6 +431,62 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
 
 }  // namespace
 
+// -- CAF fused power-iteration pair --------------------------
+
+torch::Tensor caf_matvec(torch::Tensor X, torch::Tensor mu, torch::Tensor v) {
+  check_matrix(X);
+  const int n = (int)X.size(0);
+  const long d = (long)X.size(1);
+  TORCH_CHECK(mu.is_cuda() && mu.scalar_type() == torch::kFloat32 &&
+              mu.is_contiguous() && mu.numel() == d);
+  TORCH_CHECK(v.is_cuda() && v.scalar_type() == torch::kFloat32 &&
+              v.is_contiguous() && v.numel() == d);
+  auto out = torch::zeros({n}, X.options().dtype(torch::kFloat32));
+  if (X.scalar_type() == torch::kFloat32)
+    launch_caf_matvec<float>(X.data_ptr<float>(), mu.data_ptr<float>(),
+                             v.data_ptr<float>(), out.data_ptr<float>(), n, d,
+                             cur_stream());
+  else
+    launch_caf_matvec<__hip_bfloat16>(bf16_ptr(X), mu.data_ptr<float>(),
+                                      v.data_ptr<float>(),
+                                      out.data_ptr<float>(), n, d,
+                                      cur_stream());
+  return out;
+}
+
+torch::Tensor caf_colsum(torch::Tensor X, torch::Tensor a,
+                         c10::optional<torch::Tensor> mu,
+                         c10::optional<torch::Tensor> scale) {
+  check_matrix(X);
+  const int n = (int)X.size(0);
+  const long d = (long)X.size(1);
+  TORCH_CHECK(n <= kMaxRowsLds, "caf_colsum supports n <= 1024");
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kFloat32 &&
+              a.is_contiguous() && a.numel() == n);
+  const float* mu_p = nullptr;
+  if (mu.has_value()) {
+    TORCH_CHECK(mu->is_cuda() && mu->scalar_type() == torch::kFloat32 &&
+                mu->is_contiguous() && mu->numel() == d);
+    mu_p = mu->data_ptr<float>();
+  }
+  const float* sc_p = nullptr;
+  if (scale.has_value()) {
+    TORCH_CHECK(scale->is_cuda() &&
+                scale->scalar_type() == torch::kFloat32 &&
+                scale->numel() == 1);
+    sc_p = scale->data_ptr<float>();
+  }
+  auto out = torch::empty({d}, X.options().dtype(torch::kFloat32));
+  if (X.scalar_type() == torch::kFloat32)
+    launch_caf_colsum<float>(X.data_ptr<float>(), a.data_ptr<float>(), mu_p,
+                             sc_p, out.data_ptr<float>(), n, d, cur_stream());
+  else
+    launch_caf_colsum<__hip_bfloat16>(bf16_ptr(X), a.data_ptr<float>(), mu_p,
+                                      sc_p, out.data_ptr<float>(), n, d,
+                                      cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsel", &colsel, "columnwise median/trimmed-mean/meamed");
   m.def("row_sqnorms", &row_sqnorms);
@@ -435,6 +497,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_mean", &bucket_mean);
   m.def("gram", &gram);
   m.def("krum_select", &krum_select);
+  m.def("caf_matvec", &caf_matvec);
+  m.def("caf_colsum", &caf_colsum, py::arg("X"), py::arg("a"),
+        py::arg("mu") = c10::nullopt, py::arg("scale") = c10::nullopt);
   m.def("weiszfeld_iter", &weiszfeld_iter);
   m.def("weiszfeld_apply", &weiszfeld_apply);
   m.def("cc_iter", &cc_iter);

@@ -130,7 +130,7 @@ __global__ void row_red_kernel(const T* __restrict__ X,
 // z is read once per group instead of once per row — at small n the f32
 // center dominates traffic (n=8 bf16: z re-reads were 2/3 of all bytes).
 
-constexpr int DIST_GROUP = 8;
+constexpr int DIST_GROUP = 16;
 
 template <typename T, bool VEC>
 __global__ void center_sqdists_group_kernel(const T* __restrict__ X,
@@ -457,6 +457,141 @@ __global__ void cc_update_kernel(const T* __restrict__ X,
   }
 }
 
+// -- CAF fused power-iteration pair (reference caf.py:133-184)
+// s_i = sum_j (X_ij - mu_j) v_j            (caf_matvec_group_kernel)
+// t_j = (*scale) * sum_i a_i (X_ij - mu_j) (caf_colsum_kernel)
+// Neither materializes diffs = X - mu (32 GB f32 at 64 x 125M) and both
+// replace rocblas gemvt, which runs at ~260 GB/s on this skinny shape.
+
+template <typename T, bool VEC>
+__global__ void caf_matvec_group_kernel(const T* __restrict__ X,
+                                        const float* __restrict__ mu,
+                                        const float* __restrict__ v,
+                                        float* __restrict__ out, int n,
+                                        long d) {
+  __shared__ float lds[DIST_GROUP][16];
+  constexpr int V = VecTraits<T>::V;
+  const int g0 = blockIdx.y * DIST_GROUP;
+  const int rows = min(DIST_GROUP, n - g0);
+  float acc[DIST_GROUP] = {0};
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float muv[V], vv[V];
+      {
+        float tmp[4];
+#pragma unroll
+        for (int q4 = 0; q4 < V / 4; ++q4) {
+          VecTraits<float>::load(mu + jv * V + q4 * 4, tmp);
+#pragma unroll
+          for (int c = 0; c < 4; ++c) muv[q4 * 4 + c] = tmp[c];
+          VecTraits<float>::load(v + jv * V + q4 * 4, tmp);
+#pragma unroll
+          for (int c = 0; c < 4; ++c) vv[q4 * 4 + c] = tmp[c];
+        }
+      }
+      int i = 0;
+      for (; i + 4 <= rows; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)(g0 + i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)(g0 + i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)(g0 + i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)(g0 + i + 3) * d + jv * V, x3);
+#pragma unroll
+        for (int c = 0; c < V; ++c) {
+          acc[i + 0] += (x0[c] - muv[c]) * vv[c];
+          acc[i + 1] += (x1[c] - muv[c]) * vv[c];
+          acc[i + 2] += (x2[c] - muv[c]) * vv[c];
+          acc[i + 3] += (x3[c] - muv[c]) * vv[c];
+        }
+      }
+      for (; i < rows; ++i) {
+        float x[V];
+        VecTraits<T>::load(X + (long)(g0 + i) * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[i] += (x[c] - muv[c]) * vv[c];
+      }
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      const float mj = mu[j], vj = v[j];
+      for (int i = 0; i < rows; ++i)
+        acc[i] += (to_f<T>(X[(long)(g0 + i) * d + j]) - mj) * vj;
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < DIST_GROUP; ++i) {
+    if (i < rows) {
+      const float s = block_reduce_sum(acc[i], lds[i]);
+      if (threadIdx.x == 0) atomicAdd(&out[g0 + i], s);
+    }
+    __syncthreads();
+  }
+}
+
+// sum_i a_i X_ij - mu_j * sum_i a_i, scaled: the mu subtraction factors out
+// of the row loop, so the inner walk is a plain weighted column sum.
+// mu may be null (plain weighted sum: computes the weighted mean when
+// a = w and *scale = 1/sum(w)); scale may be null (1.0).
+template <typename T, bool VEC>
+__global__ void caf_colsum_kernel(const T* __restrict__ X,
+                                  const float* __restrict__ a,
+                                  const float* __restrict__ mu,
+                                  const float* __restrict__ scale,
+                                  float* __restrict__ out, int n, long d) {
+  __shared__ float a_lds[MAX_N_LDS];
+  constexpr int V = VecTraits<T>::V;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) a_lds[i] = a[i];
+  __syncthreads();
+  float a_sum = 0.0f;
+  for (int i = 0; i < n; ++i) a_sum += a_lds[i];
+  const float sc = scale ? *scale : 1.0f;
+
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float acc[V] = {0};
+      int i = 0;
+      for (; i + 4 <= n; i += 4) {
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(X + (long)(i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(X + (long)(i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(X + (long)(i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(X + (long)(i + 3) * d + jv * V, x3);
+        const float a0 = a_lds[i], a1 = a_lds[i + 1], a2 = a_lds[i + 2],
+                    a3 = a_lds[i + 3];
+#pragma unroll
+        for (int c = 0; c < V; ++c)
+          acc[c] += (a0 * x0[c] + a1 * x1[c]) + (a2 * x2[c] + a3 * x3[c]);
+      }
+      for (; i < n; ++i) {
+        const float ai = a_lds[i];
+        float x[V];
+        VecTraits<T>::load(X + (long)i * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) acc[c] += ai * x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) {
+        const float m = mu ? mu[jv * V + c] : 0.0f;
+        out[jv * V + c] = (acc[c] - a_sum * m) * sc;
+      }
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      float acc = 0.0f;
+      for (int i = 0; i < n; ++i)
+        acc += a_lds[i] * to_f<T>(X[(long)i * d + j]);
+      const float m = mu ? mu[j] : 0.0f;
+      out[j] = (acc - a_sum * m) * sc;
+    }
+  }
+}
+
 inline int col_grid(long work, int block) {
   const long want = (work + block - 1) / block;
   return (int)(want < 4096 ? (want > 0 ? want : 1) : 4096);
@@ -603,8 +738,44 @@ void launch_cc_update(const T* X, const float* v, const float* dist2,
                        v, dist2, v_new, n, d, c_tau, eps);
 }
 
+template <typename T>
+void launch_caf_matvec(const T* X, const float* mu, const float* v, float* out,
+                       int n, long d, hipStream_t stream) {
+  const int block = 256;
+  const int groups = (n + DIST_GROUP - 1) / DIST_GROUP;
+  dim3 grid(slab_grid(d, groups, block), groups);
+  if (vec_ok<T>(d))
+    hipLaunchKernelGGL((caf_matvec_group_kernel<T, true>), grid, dim3(block),
+                       0, stream, X, mu, v, out, n, d);
+  else
+    hipLaunchKernelGGL((caf_matvec_group_kernel<T, false>), grid, dim3(block),
+                       0, stream, X, mu, v, out, n, d);
+}
+
+template <typename T>
+void launch_caf_colsum(const T* X, const float* a, const float* mu,
+                       const float* scale, float* out, int n, long d,
+                       hipStream_t stream) {
+  const int block = 256;
+  const bool vo = vec_ok<T>(d);
+  const long work = vo ? d / VecTraits<T>::V : d;
+  if (vo)
+    hipLaunchKernelGGL((caf_colsum_kernel<T, true>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       a, mu, scale, out, n, d);
+  else
+    hipLaunchKernelGGL((caf_colsum_kernel<T, false>),
+                       dim3(col_grid(work, block)), dim3(block), 0, stream, X,
+                       a, mu, scale, out, n, d);
+}
+
 // explicit instantiations for bind.cpp
 #define INSTANTIATE(T)                                                         \
+  template void launch_caf_matvec<T>(const T*, const float*, const float*,     \
+                                     float*, int, long, hipStream_t);          \
+  template void launch_caf_colsum<T>(const T*, const float*, const float*,     \
+                                     const float*, float*, int, long,          \
+                                     hipStream_t);                             \
   template void launch_row_sqnorms<T>(const T*, float*, int, long,             \
                                       hipStream_t);                            \
   template void launch_row_center_sqdists<T>(const T*, const float*, float*,   \

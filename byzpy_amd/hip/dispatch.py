@@ -3,9 +3,9 @@
 CPU tensors -> byzpy_amd.ops.functional (pure torch, the parity oracle).
 CUDA(ROCm) tensors -> hand-written gfx950 kernels in byzpy_amd/_hip_ops
 (SURVEY.md §2.7 kernel inventory K1-K14). Ops whose GPU path composes
-library GEMMs / batched eig (SMEA combos, CAF matvecs, attacks) run the
-functional torch path on-device — rocBLAS library GEMMs are the sanctioned
-path for plain GEMM shapes.
+library GEMMs / batched eig (SMEA combos, attacks) run the functional
+torch path on-device — rocBLAS library GEMMs are the sanctioned path for
+plain GEMM shapes. CAF runs on the fused K9 matvec/colsum kernel pair.
 """
 from __future__ import annotations
 
@@ -280,4 +280,52 @@ def smea(X: torch.Tensor, f: int) -> torch.Tensor:
 
 
 def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
-    return F.caf(X, f, power_iters=power_iters)
+    """Covariance-agnostic filter on fused K9 kernels (reference
+    caf.py:133-184): caf_matvec computes s = (X - mu) @ v, caf_colsum
+    computes weighted centered column sums — neither materializes the
+    (n, d) f32 diffs matrix and both replace rocblas gemvt (~260 GB/s on
+    this skinny shape). Same algorithm and seeded-rng contract as the
+    functional oracle."""
+    import math
+
+    n = X.shape[0]
+    if not (_gpu(X) and n <= 1024 and n - 2 * f > 0):
+        return F.caf(X, f, power_iters=power_iters)
+    ext = _hip.require()
+    d = X.shape[1]
+    dev = X.device
+    w = torch.ones(n, device=dev)
+    gen = torch.Generator(device="cpu")
+    gen.manual_seed(0)
+    best_lambda = math.inf
+    best_mu = X.float().mean(dim=0)
+    target = float(n - 2 * f)
+    block_rows = max(1, min(4, (1 << 27) // max(d, 1)))
+    seeds = torch.empty(0)
+    for r in range(n):
+        if r % block_rows == 0:
+            rows = min(block_rows, n - r)
+            seeds = torch.randn(rows, d, generator=gen).to(dev)
+        wsum = w.sum()
+        inv_wsum = torch.reciprocal(wsum.clamp_min(1e-30))
+        mu = ext.caf_colsum(X, w, None, inv_wsum)
+        v = seeds[r % block_rows]
+        v = v / v.norm().clamp_min(1e-20)
+        lam = torch.zeros((), device=dev)
+        for _ in range(max(1, power_iters)):
+            s = ext.caf_matvec(X, mu, v)
+            t = ext.caf_colsum(X, w * s, mu, inv_wsum)
+            lam = t.norm()
+            v = t / lam.clamp_min(1e-20)
+        proj = ext.caf_matvec(X, mu, v) ** 2
+        pmax = proj.max().clamp_min(1e-20)
+        w_next = (w * (1.0 - proj / pmax)).clamp_min(0.0)
+        # one host sync per round: break conditions + best tracking
+        wsum_f, wnext_f, lam_f = torch.stack([wsum, w_next.sum(), lam]).tolist()
+        if lam_f < best_lambda:
+            best_lambda = lam_f
+            best_mu = mu
+        if wsum_f <= target or wnext_f <= 0:
+            break
+        w = w_next
+    return best_mu.to(X.dtype)

@@ -206,3 +206,48 @@ def test_krum_select_kernel(n, f, q):
     scores = torch.topk(D2, k=n - f - 1, dim=1, largest=False).values.sum(dim=1)
     ref = torch.topk(scores, k=q, largest=False).indices
     assert set(idx.cpu().tolist()) == set(ref.cpu().tolist())
+
+
+class TestCafKernels:
+    """K9 fused pair: caf_matvec (s = (X - mu) @ v) and caf_colsum
+    (t = scale * sum_i a_i (x_i - mu)) vs plain torch f32."""
+
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    @pytest.mark.parametrize("n,d", [(8, 4096), (64, 8192), (33, 1000)])
+    def test_matvec_parity(self, n, d, dtype):
+        from byzpy_amd.hip import require
+
+        X = _rand(n, d, dtype, seed=31)
+        mu = _rand(1, d, torch.float32, seed=32)[0]
+        v = _rand(1, d, torch.float32, seed=33)[0]
+        out = require().caf_matvec(X, mu, v)
+        ref = (X.float() - mu) @ v
+        assert torch.allclose(out, ref, atol=1e-2 * d**0.5, rtol=1e-3)
+
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    @pytest.mark.parametrize("n,d", [(8, 4096), (64, 8192), (33, 1000)])
+    def test_colsum_parity(self, n, d, dtype):
+        from byzpy_amd.hip import require
+
+        X = _rand(n, d, dtype, seed=41)
+        mu = _rand(1, d, torch.float32, seed=42)[0]
+        a = _rand(1, n, torch.float32, seed=43)[0]
+        sc = torch.tensor(0.25, device="cuda")
+        out = require().caf_colsum(X, a, mu, sc)
+        ref = 0.25 * (a[:, None] * (X.float() - mu)).sum(dim=0)
+        assert torch.allclose(out, ref, atol=2e-2, rtol=1e-3)
+        # mu=None, scale=None: plain weighted column sum
+        out2 = require().caf_colsum(X, a)
+        ref2 = (a[:, None] * X.float()).sum(dim=0)
+        assert torch.allclose(out2, ref2, atol=2e-2, rtol=1e-3)
+
+    def test_caf_dispatch_matches_oracle(self):
+        X = _rand(24, 4096, torch.float32, seed=51)
+        honest = X.clone()
+        honest[-4:] = 40.0  # 4 clear outlier rows
+        got = D.caf(honest, 4)
+        ref = F.caf(honest.cpu(), 4)
+        # trajectories are float-order sensitive; demand the robust outcome
+        assert (got.cpu() - ref).norm() < 0.5 * ref.norm() + 1e-3
+        clean_mean = honest[:-4].mean(dim=0).cpu()
+        assert (got.cpu() - clean_mean).norm() < (honest.mean(dim=0).cpu() - clean_mean).norm()
