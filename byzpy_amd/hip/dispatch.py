@@ -297,17 +297,11 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     w = torch.ones(n, device=dev)
     gen = torch.Generator(device="cpu")
     gen.manual_seed(0)
-    best_lambda = torch.full((), math.inf, device=dev)
+    best_lambda = math.inf
     best_mu = X.float().mean(dim=0)
     target = float(n - 2 * f)
     block_rows = max(1, min(4, (1 << 27) // max(d, 1)))
     seeds = torch.empty(0)
-    # With the fused kernels a round at small d is ~25 launches of a few us
-    # each, so the per-round break sync dominates: batch the check 4 rounds
-    # deep (inert rounds are gated by `active` and cheap). At large d a
-    # wasted round re-reads X seven times, so sync every round instead.
-    sync_every = 4 if d <= (1 << 22) else 1
-    active = torch.ones((), dtype=torch.bool, device=dev)
     for r in range(n):
         if r % block_rows == 0:
             rows = min(block_rows, n - r)
@@ -326,12 +320,12 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
         proj = ext.caf_matvec(X, mu, v) ** 2
         pmax = proj.max().clamp_min(1e-20)
         w_next = (w * (1.0 - proj / pmax)).clamp_min(0.0)
-        better = active & (lam < best_lambda)
-        best_lambda = torch.where(better, lam, best_lambda)
-        best_mu = torch.where(better, mu, best_mu)
-        cont = active & (wsum > target) & (w_next.sum() > 0)
-        w = torch.where(cont, w_next, w)
-        active = cont
-        if (r + 1) % sync_every == 0 and not bool(active):
+        # one host sync per round: break conditions + best tracking
+        wsum_f, wnext_f, lam_f = torch.stack([wsum, w_next.sum(), lam]).tolist()
+        if lam_f < best_lambda:
+            best_lambda = lam_f
+            best_mu = mu
+        if wsum_f <= target or wnext_f <= 0:
             break
+        w = w_next
     return best_mu.to(X.dtype)
