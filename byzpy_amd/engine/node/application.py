@@ -29,8 +29,20 @@ class NodeApplication:
         self._pipelines: Dict[str, NodePipeline] = {}
 
     def register_pipeline(
-        self, name: str, graph: ComputationGraph, metadata: Optional[dict] = None
+        self,
+        name: str,
+        graph: ComputationGraph,
+        metadata: Optional[dict] = None,
+        *,
+        _internal: bool = False,
     ) -> None:
+        """Register a named pipeline. Names in ``reserved_pipelines`` are
+        wired by the node itself (reference application.py:144-261) and
+        cannot be overridden by application code."""
+        if not _internal and name in self.reserved_pipelines:
+            raise ValueError(
+                f"pipeline name {name!r} is reserved by {type(self).__name__}"
+            )
         self._pipelines[name] = NodePipeline(graph, dict(metadata or {}))
 
     def has_pipeline(self, name: str) -> bool:

@@ -45,6 +45,7 @@ class DistributedHonestNode(HonestNode):
                     )
                 ]
             ),
+            _internal=True,
         )
         self.app.register_pipeline(
             "honest_gradient",
@@ -57,6 +58,7 @@ class DistributedHonestNode(HonestNode):
                     )
                 ]
             ),
+            _internal=True,
         )
 
     # -- overridables -------------------------------------------------------
@@ -113,6 +115,7 @@ class DistributedByzantineNode(ByzantineNode):
                         )
                     ]
                 ),
+                _internal=True,
             )
             self._attack_keys = keys
         else:
@@ -136,6 +139,7 @@ class DistributedByzantineNode(ByzantineNode):
             self.app.register_pipeline(
                 "attack",
                 ComputationGraph([GraphNode("attack", attack, inputs)]),
+                _internal=True,
             )
             self._attack_keys = keys
 
