@@ -1,0 +1,79 @@
+"""BASELINE config 4: P2P ring gossip with NNM pre-aggregation +
+GeometricMedian, 8 nodes.
+
+Single process = all 8 node states resident on one GPU; one round = every
+node aggregates NNM(GeoMedian) over [self] + its ring neighbors' theta-half
+vectors. The multi-rank variant runs the identical aggregate per rank after
+an RCCL all-gather (engine/peer_to_peer/rccl.py; gloo-ws2 covered by
+tests/test_dist_gloo.py::test_rccl_peer_to_peer). Reports ms per full gossip
+round (all 8 node updates).
+
+  python benchmarks/config4_p2p_ring.py [--rounds 10] [--d 25610152]
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import torch
+
+from byzpy_amd.hip import dispatch as D
+from byzpy_amd.engine.peer_to_peer.topology import Topology
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--rounds", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--nodes", type=int, default=8)
+    p.add_argument("--k", type=int, default=2, help="ring degree (each side)")
+    p.add_argument("--d", type=int, default=25_610_152)
+    p.add_argument("--f", type=int, default=1)
+    p.add_argument("--device", default="cuda" if torch.cuda.is_available() else "cpu")
+    args = p.parse_args()
+    dev = torch.device(args.device)
+    dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
+    if dev.type == "cpu":
+        args.d = min(args.d, 200_000)
+
+    topo = Topology.ring(args.nodes, args.k)
+    g = torch.Generator(device=dev).manual_seed(0)
+    theta = torch.empty(args.nodes, args.d, dtype=dtype, device=dev).normal_(
+        generator=g
+    )
+
+    def round_once():
+        new = torch.empty_like(theta)
+        for i in range(args.nodes):
+            group = [i] + topo.in_neighbors(i)
+            X = theta[group]  # (1+2k, d) view-gather
+            mixed = D.nnm(X, args.f)
+            new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+        theta.copy_(new)
+
+    def sync():
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        round_once()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.rounds):
+        round_once()
+    sync()
+    ms = (time.perf_counter() - t0) / args.rounds * 1e3
+    spread = float((theta.float().std(dim=0)).mean())
+    print(
+        f"config4 P2P ring({args.nodes},{args.k}) NNM+GeoMedian d={args.d} "
+        f"{dtype}: {ms:.3f} ms/round (all {args.nodes} node updates); "
+        f"consensus spread {spread:.4f}"
+    )
+
+
+if __name__ == "__main__":
+    main()
