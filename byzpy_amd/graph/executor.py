@@ -8,7 +8,6 @@ from __future__ import annotations
 
 from typing import Any, Dict, Optional, Sequence
 
-from byzpy_amd.graph.graph import ComputationGraph
 from byzpy_amd.graph.ops import make_single_operator_graph
 from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
 from byzpy_amd.graph.scheduler import NodeScheduler
