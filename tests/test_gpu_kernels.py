@@ -251,3 +251,74 @@ class TestCafKernels:
         assert (got.cpu() - ref).norm() < 0.5 * ref.norm() + 1e-3
         clean_mean = honest[:-4].mean(dim=0).cpu()
         assert (got.cpu() - clean_mean).norm() < (honest.mean(dim=0).cpu() - clean_mean).norm()
+
+
+AGG_GPU_CASES = None
+
+
+def _all_agg_instances():
+    from byzpy_amd.aggregators import (
+        CAF,
+        CenteredClipping,
+        ComparativeGradientElimination,
+        CoordinateWiseMedian,
+        CoordinateWiseTrimmedMean,
+        GeometricMedian,
+        Krum,
+        MeanOfMedians,
+        MinimumDiameterAveraging,
+        MoNNA,
+        MultiKrum,
+        SMEA,
+    )
+
+    return [
+        CoordinateWiseMedian(),
+        CoordinateWiseTrimmedMean(3),
+        MeanOfMedians(3),
+        MultiKrum(3, 4),
+        Krum(3),
+        GeometricMedian(),
+        MinimumDiameterAveraging(3),
+        MoNNA(3),
+        SMEA(3),
+        CenteredClipping(c_tau=0.7),
+        ComparativeGradientElimination(3),
+        CAF(3),
+    ]
+
+
+@pytest.mark.parametrize(
+    "agg", _all_agg_instances(), ids=[a.name for a in _all_agg_instances()]
+)
+def test_every_aggregator_on_device_matches_cpu_oracle(agg):
+    """Every aggregator class, bf16 CUDA tensors in -> device result out,
+    within bf16 tolerance of the fp32 CPU oracle on the same values."""
+    X = _rand(16, 4097, torch.bfloat16, seed=77)  # odd d: vector fallback
+    grads = list(X)
+    out = agg.aggregate(grads)
+    assert out.is_cuda and out.dtype == torch.bfloat16 and out.shape == (4097,)
+    ref = agg.aggregate([g.cpu().float() for g in grads])
+    tol = 0.05 if agg.name in ("caf",) else 0.02
+    num = (out.cpu().float() - ref).norm()
+    den = ref.norm().clamp_min(1e-6)
+    assert num / den < tol, f"{agg.name}: rel err {num / den:.4f}"
+
+
+@pytest.mark.parametrize("preagg_name", ["clipping", "bucketing", "nnm", "arc"])
+def test_every_preagg_on_device(preagg_name):
+    from byzpy_amd.pre_aggregators import ARC, Bucketing, Clipping, NearestNeighborMixing
+
+    pre = {
+        "clipping": Clipping(1.0),
+        "bucketing": Bucketing(3, perm=list(range(16))),
+        "nnm": NearestNeighborMixing(3),
+        "arc": ARC(3),
+    }[preagg_name]
+    X = _rand(16, 2048, torch.bfloat16, seed=78)
+    out = pre.pre_aggregate(list(X))
+    assert all(v.is_cuda and v.dtype == torch.bfloat16 for v in out)
+    ref = pre.pre_aggregate([g.cpu().float() for g in X])
+    assert len(out) == len(ref)
+    for a, b in zip(out, ref):
+        assert (a.cpu().float() - b).norm() / b.norm().clamp_min(1e-6) < 0.02
