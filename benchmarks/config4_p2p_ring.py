@@ -33,6 +33,8 @@ def main() -> None:
     p.add_argument("--k", type=int, default=2, help="ring degree (each side)")
     p.add_argument("--d", type=int, default=25_610_152)
     p.add_argument("--f", type=int, default=1)
+    p.add_argument("--streams", action="store_true",
+                   help="one HIP stream per node update (overlap)")
     p.add_argument("--device", default="cuda" if torch.cuda.is_available() else "cpu")
     args = p.parse_args()
     dev = torch.device(args.device)
@@ -46,13 +48,34 @@ def main() -> None:
         generator=g
     )
 
+    streams = (
+        [torch.cuda.Stream() for _ in range(args.nodes)]
+        if (args.streams and dev.type == "cuda")
+        else None
+    )
+
     def round_once():
         new = torch.empty_like(theta)
-        for i in range(args.nodes):
-            group = [i] + topo.in_neighbors(i)
-            X = theta[group]  # (1+2k, d) view-gather
-            mixed = D.nnm(X, args.f)
-            new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+        if streams is None:
+            for i in range(args.nodes):
+                group = [i] + topo.in_neighbors(i)
+                X = theta[group]  # (1+2k, d) view-gather
+                mixed = D.nnm(X, args.f)
+                new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+        else:
+            # node updates are independent: one HIP stream each, so the
+            # launch/sync-bound small-kernel chains overlap on the chip
+            cur = torch.cuda.current_stream()
+            for s in streams:
+                s.wait_stream(cur)
+            for i in range(args.nodes):
+                with torch.cuda.stream(streams[i]):
+                    group = [i] + topo.in_neighbors(i)
+                    X = theta[group]
+                    mixed = D.nnm(X, args.f)
+                    new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+            for s in streams:
+                cur.wait_stream(s)
         theta.copy_(new)
 
     def sync():
