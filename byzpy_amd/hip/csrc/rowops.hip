@@ -130,7 +130,7 @@ __global__ void row_red_kernel(const T* __restrict__ X,
 // z is read once per group instead of once per row — at small n the f32
 // center dominates traffic (n=8 bf16: z re-reads were 2/3 of all bytes).
 
-constexpr int DIST_GROUP = 16;
+constexpr int DIST_GROUP = 32;
 
 template <typename T, bool VEC>
 __global__ void center_sqdists_group_kernel(const T* __restrict__ X,
