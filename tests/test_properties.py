@@ -105,3 +105,90 @@ def test_nnm_rows_in_convex_hull_bounds(p):
     f = max(0, n // 4)
     out = F.nnm(X, f)
     assert (out.max() <= X.max() + 1e-5) and (out.min() >= X.min() - 1e-5)
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_trimmed_mean_within_envelope(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = (n - 1) // 2
+    out = F.trimmed_mean(X, f)
+    lo, hi = X.min(dim=0).values, X.max(dim=0).values
+    assert (out >= lo - 1e-5).all() and (out <= hi + 1e-5).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_meamed_within_envelope(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = max(0, n // 3)
+    out = F.mean_of_medians(X, f)
+    lo, hi = X.min(dim=0).values, X.max(dim=0).values
+    assert (out >= lo - 1e-5).all() and (out <= hi + 1e-5).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_arc_never_grows_norms(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = max(0, n // 4)
+    out = F.arc_clip(X, f)
+    assert (out.norm(dim=1) <= X.norm(dim=1) + 1e-4).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_nnm_is_variance_contraction(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = max(0, n // 4)
+    out = F.nnm(X, f)
+    assert float(out.var(dim=0).sum()) <= float(X.var(dim=0).sum()) + 1e-4
+
+
+@settings(max_examples=15, deadline=None)
+@given(matrix_params)
+def test_geomed_optimality_vs_mean_and_rows(p):
+    # the geometric median minimizes sum of distances: it must beat both
+    # the mean and every input row on that objective
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    gm = F.geometric_median(X, tol=1e-9, max_iter=500)
+
+    def obj(z):
+        return float((X - z[None, :]).norm(dim=1).sum())
+
+    best_row = min(obj(X[i]) for i in range(n))
+    assert obj(gm) <= obj(X.mean(dim=0)) + 1e-3
+    assert obj(gm) <= best_row + 1e-3
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_cge_subset_of_rows_mean(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = max(0, n // 3)
+    out = F.cge(X, f)
+    norms = (X * X).sum(dim=1)
+    keep = torch.argsort(norms, stable=True)[: n - f]
+    assert torch.allclose(out, X[keep].mean(dim=0), atol=1e-4)
+
+
+@settings(max_examples=20, deadline=None)
+@given(matrix_params)
+def test_multi_krum_q_equals_n_minus_f_is_selection_mean(p):
+    n, d, seed = p
+    X = _matrix(n, d, seed)
+    f = (n - 2) // 2
+    if n - f - 1 < 1:
+        return
+    q = max(1, n - f)
+    scores = F.multi_krum_scores(X, f)
+    keep = torch.topk(scores, q, largest=False).indices
+    assert torch.allclose(
+        F.multi_krum(X, f, q), X[keep].mean(dim=0), atol=1e-4
+    )
