@@ -56,10 +56,17 @@ def _worker(conn) -> None:
                 name, payload = msg[1]
                 mailboxes.setdefault(name, stdlib_queue.Queue()).put(payload)
                 conn.send(("ok", None))
-            elif op == "chan_get":
+            elif op == "chan_get_try":
+                # NON-blocking: a blocking q.get() here would wedge the
+                # request loop while the parent holds _io_lock, so the
+                # chan_deliver that fills the mailbox could never arrive
+                # (get-before-put deadlock). The parent polls instead.
                 name = msg[1]
-                payload = mailboxes.setdefault(name, stdlib_queue.Queue()).get()
-                conn.send(("ok", payload))
+                q = mailboxes.setdefault(name, stdlib_queue.Queue())
+                try:
+                    conn.send(("ok", ("item", q.get_nowait())))
+                except stdlib_queue.Empty:
+                    conn.send(("ok", ("empty", None)))
             elif op == "close":
                 conn.send(("ok", None))
                 break
@@ -153,4 +160,8 @@ class ProcessActorBackend:
         raise RuntimeError(f"no route to endpoint {endpoint!r}")
 
     async def chan_get(self, name: str) -> Any:
-        return unwrap_payload(await self._request(("chan_get", name)))
+        while True:
+            kind, payload = await self._request(("chan_get_try", name))
+            if kind == "item":
+                return unwrap_payload(payload)
+            await asyncio.sleep(0.005)  # lock released between tries

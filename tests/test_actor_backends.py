@@ -120,3 +120,33 @@ def test_remote_tcp_actor_loopback():
             await server.stop()
 
     asyncio.run(main())
+
+
+def test_process_chan_get_before_put_no_deadlock():
+    """Regression: a chan_get on an empty mailbox must not wedge the child's
+    request loop — the deliver that fills it arrives over the same pipe."""
+    import asyncio
+
+    from byzpy_amd.actor.backends.process import ProcessActorBackend
+
+    class Obj:
+        pass
+
+    async def main():
+        be = ProcessActorBackend()
+        await be.start()
+        await be.construct(Obj)
+        await be.chan_open("m")
+
+        async def getter():
+            return await be.chan_get("m")
+
+        task = asyncio.create_task(getter())
+        await asyncio.sleep(0.2)  # getter is polling an empty mailbox
+        await be.chan_put(be.get_endpoint(), "m", {"x": 7})
+        out = await asyncio.wait_for(task, timeout=10)
+        await be.close()
+        return out
+
+    out = asyncio.run(main())
+    assert out["x"] == 7
