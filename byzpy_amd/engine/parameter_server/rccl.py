@@ -50,6 +50,14 @@ class RcclParameterServer:
         # RCCL's stream (coordinate-chunkable aggregators only: every
         # coordinate-wise op qualifies; Krum-style global scoring does not).
         self.overlap_chunks = max(1, int(overlap_chunks))
+        if self.overlap_chunks > 1 and getattr(
+            aggregate_fn, "coordinate_chunkable", True
+        ) is False:
+            raise ValueError(
+                "aggregate_fn is not coordinate-chunkable; overlap_chunks "
+                "must be 1 (chunked pipelining would aggregate each chunk "
+                "independently)"
+            )
 
     def round(self) -> torch.Tensor:
         world = pdist.get_world_size()
@@ -130,7 +138,12 @@ def median_aggregate():
 
 
 def multi_krum_aggregate(f: int, q: int):
+    """NOT coordinate-chunkable: Krum scores are a global function of the
+    full shard (Gram all-reduce), so use overlap_chunks=1 with this one —
+    chunked pipelining would score each chunk independently."""
+
     def _agg(X_shard: torch.Tensor) -> torch.Tensor:
         return sharded.multi_krum(X_shard, f, q)
 
+    _agg.coordinate_chunkable = False
     return _agg
