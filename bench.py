@@ -152,6 +152,10 @@ def main():
                 }
             )
         )
+    if pdist.is_initialized():
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
 
 
 if __name__ == "__main__":
