@@ -27,26 +27,46 @@ class _P2PNodeShell(DecentralizedNode):
         self.obj = obj
         self.is_byzantine = is_byzantine
         self._round_vectors: Dict[int, List[torch.Tensor]] = {}
-        self._round_events: Dict[int, asyncio.Event] = {}
+        self._round_conds: Dict[int, asyncio.Condition] = {}
         self._expected = 0
         self.register_handler("gradient", self._on_gradient)
 
     def set_expected(self, count: int) -> None:
         self._expected = count
 
+    def _cond(self, rnd: int) -> asyncio.Condition:
+        return self._round_conds.setdefault(rnd, asyncio.Condition())
+
     async def _on_gradient(self, msg: dict) -> None:
         rnd = int(msg["round"])
-        self._round_vectors.setdefault(rnd, []).append(msg["vector"])
-        evt = self._round_events.setdefault(rnd, asyncio.Event())
-        if len(self._round_vectors[rnd]) >= self._expected:
-            evt.set()
+        cond = self._cond(rnd)
+        async with cond:
+            self._round_vectors.setdefault(rnd, []).append(msg["vector"])
+            cond.notify_all()
 
-    async def wait_round(self, rnd: int, timeout: float = 30.0) -> List[torch.Tensor]:
-        evt = self._round_events.setdefault(rnd, asyncio.Event())
-        if self._expected == 0:
+    async def wait_round(
+        self, rnd: int, count: Optional[int] = None, timeout: float = 30.0
+    ) -> List[torch.Tensor]:
+        """Await `count` vectors for round `rnd` (default: all in-neighbors).
+        Byzantine turns pass the HONEST in-neighbor count: waiting for the
+        full neighborhood would deadlock two byzantine neighbors, each
+        waiting for the other's not-yet-broadcast vector."""
+        want = self._expected if count is None else count
+        if want <= 0:
             return []
-        await asyncio.wait_for(evt.wait(), timeout)
-        return self._round_vectors.pop(rnd)
+        cond = self._cond(rnd)
+        async with cond:
+            await asyncio.wait_for(
+                cond.wait_for(
+                    lambda: len(self._round_vectors.get(rnd, ())) >= want
+                ),
+                timeout,
+            )
+            return list(self._round_vectors.get(rnd, ()))[:want] if count is not None else self._round_vectors.get(rnd, [])
+
+    def drop_round(self, rnd: int) -> None:
+        self._round_vectors.pop(rnd, None)
+        self._round_conds.pop(rnd, None)
 
 
 class DecentralizedPeerToPeer:
@@ -105,9 +125,15 @@ class DecentralizedPeerToPeer:
         )
 
         # 3. byzantine nodes attack the honest vectors they can see, then
-        # broadcast the malicious vector
+        # broadcast the malicious vector. They wait only for HONEST
+        # in-neighbors (honest indices are < n_honest): waiting for the
+        # full neighborhood would deadlock two byzantine neighbors.
         async def byz_turn(s: _P2PNodeShell) -> None:
-            received = await s.wait_round(rnd) if s._expected else []
+            idx = self.shells.index(s)
+            h_in = sum(
+                1 for j in self.topology.in_neighbors(idx) if j < self.n_honest
+            )
+            received = await s.wait_round(rnd, count=h_in) if h_in else []
             honest_vecs = received or list(halves.values())
             vec = s.obj.p2p_broadcast_vector(honest_vecs)
             await s.broadcast_message("gradient", {"vector": vec, "round": rnd})
@@ -121,6 +147,8 @@ class DecentralizedPeerToPeer:
             s.obj.p2p_aggregate_and_set(vectors, self.aggregator, self.pre_aggregator)
 
         await asyncio.gather(*(honest_aggregate(s) for s in honest))
+        for s in self.shells:
+            s.drop_round(rnd)
 
     async def shutdown(self) -> None:
         await self.cluster.shutdown_all()

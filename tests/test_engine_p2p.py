@@ -107,3 +107,60 @@ def test_router_topology_enforcement():
         await cluster2.shutdown_all()
 
     asyncio.run(main())
+
+
+def test_p2p_with_preaggregator():
+    """Gossip round with NNM pre-aggregation before the robust aggregate
+    (BASELINE config 4 op pair, actor engine flavor)."""
+    from byzpy_amd.aggregators import GeometricMedian
+    from byzpy_amd.pre_aggregators import NearestNeighborMixing
+
+    async def main():
+        honest = [TinyHonest(s) for s in range(4)]
+        p2p = PeerToPeer(
+            honest,
+            [],
+            GeometricMedian(tol=1e-7),
+            pre_aggregator=NearestNeighborMixing(1),
+            lr=0.05,
+        )
+        await p2p.bootstrap()
+        for _ in range(25):
+            await p2p.round()
+        await p2p.shutdown()
+        w = honest[0].model.weight.detach()
+        err = (w - honest[0].w_true).norm()
+        assert err < 0.6, f"NNM+GeoMedian gossip did not converge: {err}"
+
+    asyncio.run(main())
+
+
+def test_p2p_repeated_attack_rounds():
+    """Byzantine pressure applied EVERY round must not prevent honest
+    convergence under a robust aggregator."""
+    async def main():
+        honest = [TinyHonest(s) for s in range(4)]
+        byz = [TinyByz(), TinyByz()]  # 2 byzantine vs 4 honest, complete topo
+        p2p = PeerToPeer(honest, byz, CoordinateWiseMedian(), lr=0.05)
+        await p2p.bootstrap()
+        for _ in range(60):
+            await p2p.round()
+        await p2p.shutdown()
+        w = honest[0].model.weight.detach()
+        assert (w - honest[0].w_true).norm() < 0.5
+
+    asyncio.run(main())
+
+
+def test_p2p_facade_kwargs_compat():
+    """PeerToPeer facade accepts the reference's keyword surface."""
+    honest = [TinyHonest(0)]
+    p2p = PeerToPeer(
+        honest,
+        [],
+        CoordinateWiseMedian(),
+        topology=Topology.complete(1),
+        pre_aggregator=None,
+        lr=0.123,
+    )
+    assert p2p._runner.lr == 0.123
