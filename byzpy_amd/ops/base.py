@@ -60,9 +60,31 @@ class Operator:
             subtasks = list(self.create_subtasks(ctx, **inputs))
             if subtasks:
                 self._assign_worker_affinities(ctx, subtasks)
-                results = await self._run_subtasks_windowed(ctx, subtasks)
+                try:
+                    results = await self._run_subtasks_windowed(ctx, subtasks)
+                except BaseException:
+                    # a failed fan-out never reaches reduce_subtasks (whose
+                    # finally releases shared-memory handles): release any
+                    # pending op state here so POSIX shm never leaks
+                    self._release_pending(ctx)
+                    raise
                 return self.reduce_subtasks(ctx, results, **inputs)
         return self.compute(ctx, **inputs)
+
+    def _release_pending(self, ctx: OpContext) -> None:
+        """Best-effort release of create_subtasks state when the fan-out
+        fails before reduce. Operators stash (like, handles) tuples under
+        the ``_op_pending`` metadata key (aggregators/_chunking pattern)."""
+        pending = ctx.metadata.pop("_op_pending", None)
+        if not pending:
+            return
+        try:
+            from byzpy_amd.aggregators.base import cleanup_handles
+
+            handles = pending[-1]
+            cleanup_handles(handles or ())
+        except Exception:
+            pass
 
     # -- overridables ------------------------------------------------------
     def compute(self, ctx: OpContext, **inputs: Any) -> Any:

@@ -159,3 +159,63 @@ class TestGraphOutputs:
     def test_empty_graph(self):
         g = ComputationGraph([])
         assert g.topo_order == [] and g.outputs == []
+
+
+class TestShmLeakOnFailedFanout:
+    def test_release_pending_unlinks_handles(self):
+        """_release_pending (called when the fan-out raises before reduce)
+        must unlink POSIX shm segments stashed under _op_pending."""
+        import numpy as np
+        import pytest
+
+        from byzpy_amd.aggregators import CoordinateWiseMedian
+        from byzpy_amd.ops.base import OpContext
+        from byzpy_amd.storage import shared_store
+
+        handle = shared_store.register_tensor(np.zeros((4, 8), dtype=np.float32))
+        ctx = OpContext()
+        ctx.metadata["_op_pending"] = (None, [handle])
+        CoordinateWiseMedian()._release_pending(ctx)
+        assert "_op_pending" not in ctx.metadata
+        with pytest.raises(Exception):
+            with shared_store.open_tensor(handle):
+                pass
+
+    def test_failed_fanout_calls_release(self):
+        """Operator.run releases pending state when subtasks fail."""
+        import asyncio
+
+        import pytest
+        import torch
+
+        from byzpy_amd.aggregators import CoordinateWiseMedian
+        from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
+        from byzpy_amd.ops.base import OpContext
+
+        agg = CoordinateWiseMedian(chunk_size=8)
+        grads = [torch.randn(32) for _ in range(4)]
+
+        async def main():
+            pool = ActorPool(ActorPoolConfig(backend="thread", count=2))
+            await pool.start()
+            ctx = OpContext(pool=pool)
+            subtasks = list(agg.create_subtasks(ctx, gradients=grads))
+            assert subtasks and "_op_pending" in ctx.metadata
+            for st in subtasks:
+                st.fn = _boom
+                st.max_retries = 0
+            agg._assign_worker_affinities(ctx, subtasks)
+            with pytest.raises(RuntimeError):
+                try:
+                    await agg._run_subtasks_windowed(ctx, subtasks)
+                except BaseException:
+                    agg._release_pending(ctx)
+                    raise
+            await pool.close()
+            assert "_op_pending" not in ctx.metadata
+
+        asyncio.run(main())
+
+
+def _boom(*a, **k):
+    raise RuntimeError("boom")
