@@ -41,10 +41,18 @@ def main() -> None:
     def signflip():
         return -2.0 * torch.randn(d, generator=gen).to(device)
 
-    fns = [honest, (signflip if rank < 3 else honest)]
+    # 11 workers spread over the ranks: with fewer ranks each hosts more
+    # workers; f scales so n > 2f holds at any world size
+    per_rank = max(2, (11 + world - 1) // world)
+    n_byz_left = 3
+    fns = []
+    for w in range(per_rank):
+        is_byz = rank < 3 and w == 0 and n_byz_left > 0
+        fns.append(signflip if is_byz else honest)
+    f = min(3, (per_rank * world - 1) // 2)
     # overlap_chunks pipelines chunk i+1's all-to-all under chunk i's
     # aggregation kernels (coordinate-wise ops are chunkable)
-    ps = RcclParameterServer(fns, trimmed_mean_aggregate(f=3), overlap_chunks=4)
+    ps = RcclParameterServer(fns, trimmed_mean_aggregate(f=f), overlap_chunks=4)
 
     for _ in range(3):
         ps.round()  # warmup
@@ -60,7 +68,7 @@ def main() -> None:
     pdist.barrier()
     dt = (time.perf_counter() - t0) / steps
     if rank == 0:
-        n = 2 * world
+        n = per_rank * world
         print(
             f"RCCL PS: {n} workers x d={d} on {world} rank(s): "
             f"{dt * 1000:.2f} ms/round ({n / dt:.0f} aggregated-grads/s)"
