@@ -132,3 +132,56 @@ def test_shared_handle_gradients():
     finally:
         for h in handles:
             cleanup_tensor(h)
+
+
+class TestAttackExecutorSweep:
+    """Every attack routed through run_operator with explicit input keys."""
+
+    @staticmethod
+    def _honest(n=6, d=9, seed=2):
+        g = torch.Generator().manual_seed(seed)
+        return [torch.randn(d, generator=g) for _ in range(n)]
+
+    @pytest.mark.parametrize(
+        "atk_name",
+        ["empire", "little", "gaussian", "inf", "mimic"],
+    )
+    def test_honest_grads_attacks(self, atk_name):
+        from byzpy_amd.attacks import (
+            EmpireAttack,
+            GaussianAttack,
+            InfAttack,
+            LittleAttack,
+            MimicAttack,
+        )
+
+        atk = {
+            "empire": EmpireAttack(scale=-2.0),
+            "little": LittleAttack(f=1),
+            "gaussian": GaussianAttack(seed=5),
+            "inf": InfAttack(),
+            "mimic": MimicAttack(epsilon=2),
+        }[atk_name]
+        honest = self._honest()
+        direct = atk.apply(honest_grads=honest)
+
+        async def run():
+            return await run_operator(atk, {"honest_grads": honest})
+
+        out = asyncio.run(run())
+        assert out.shape == direct.shape
+        if atk_name != "gaussian":  # gaussian reseeds per call? seeded: equal
+            assert torch.allclose(out, direct, equal_nan=True, atol=1e-5)
+
+    def test_sign_flip_via_executor(self):
+        from byzpy_amd.attacks import SignFlipAttack
+
+        base = torch.arange(5.0)
+
+        async def run():
+            return await run_operator(
+                SignFlipAttack(scale=-1.5), {"base_grad": base}
+            )
+
+        out = asyncio.run(run())
+        assert torch.allclose(out, -1.5 * base)
