@@ -134,3 +134,67 @@ def test_auto_step_runner():
         assert r.steps_done >= 2
 
     asyncio.run(main())
+
+
+def test_remote_server_two_clients_exchange():
+    """Two RemoteNodeClients routed through one RemoteNodeServer."""
+    import asyncio
+
+    from byzpy_amd.engine.node.remote import RemoteNodeClient, RemoteNodeServer
+
+    async def main():
+        server = RemoteNodeServer(host="127.0.0.1", port=0)
+        await server.start()
+        port = server.port
+        a = RemoteNodeClient("a", "127.0.0.1", port)
+        b = RemoteNodeClient("b", "127.0.0.1", port)
+        await a.connect()
+        await b.connect()
+        await a.send("b", {"type": "hello", "x": 42})
+        msg = None
+        for _ in range(100):
+            msg = await b.receive(timeout=0.1)
+            if msg is not None:
+                break
+        await a.close()
+        await b.close()
+        await server.stop()
+        return msg
+
+    msg = asyncio.run(main())
+    assert msg["x"] == 42
+
+
+def test_mesh_context_survives_peer_restart():
+    """Mesh reconnect monitor re-establishes a dropped peer link."""
+    import asyncio
+
+    from byzpy_amd.engine.node.remote import MeshRemoteContext
+
+    class Sink:
+        def __init__(self):
+            self.got = asyncio.Queue()
+
+        async def handle_incoming_message(self, msg):
+            self.got.put_nowait(msg)
+
+    async def main():
+        a = MeshRemoteContext("a", host="127.0.0.1", port=0,
+                              peers={}, reconnect_interval=0.2)
+        sink_a = Sink()
+        await a.start(sink_a)
+        b = MeshRemoteContext("b", host="127.0.0.1", port=0,
+                              peers={"a": ("127.0.0.1", a.port)},
+                              reconnect_interval=0.2)
+        sink_b = Sink()
+        await b.start(sink_b)
+        a.add_peer("b", "127.0.0.1", b.port)
+        await asyncio.sleep(0.4)
+        await b.send_message("a", {"k": 1})
+        m = await asyncio.wait_for(sink_a.got.get(), timeout=10)
+        await a.shutdown()
+        await b.shutdown()
+        return m
+
+    m = asyncio.run(main())
+    assert m["k"] == 1
