@@ -32,14 +32,16 @@ def main() -> None:
     device = torch.device("cuda", rank % max(1, torch.cuda.device_count())) if torch.cuda.is_available() else torch.device("cpu")
     d = D_RESNET50 if torch.cuda.is_available() else 100_000
 
-    # workers per rank: honest everywhere, byzantine on ranks 0..2
-    gen = torch.Generator().manual_seed(1234 + rank)
+    # workers per rank: honest everywhere, byzantine on ranks 0..2.
+    # Gradients are generated ON DEVICE (a real worker's fwd/bwd would be
+    # too): a CPU randn + H2D here costs ~80 ms per 25M-param worker.
+    gen = torch.Generator(device=device).manual_seed(1234 + rank)
 
     def honest():
-        return torch.randn(d, generator=gen).to(device)
+        return torch.randn(d, generator=gen, device=device)
 
     def signflip():
-        return -2.0 * torch.randn(d, generator=gen).to(device)
+        return -2.0 * torch.randn(d, generator=gen, device=device)
 
     # 11 workers spread over the ranks: with fewer ranks each hosts more
     # workers; f scales so n > 2f holds at any world size
