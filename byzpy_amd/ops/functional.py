@@ -250,19 +250,37 @@ def smea(X: torch.Tensor, f: int) -> torch.Tensor:
         raise ValueError(f"need n - f >= 1, got n={n}, f={f}")
     Xf = X.float()
     G = Xf @ Xf.T
-    combos = list(itertools.combinations(range(n), m))
-    idx = torch.tensor(combos, device=X.device, dtype=torch.long)  # (C, m)
-    # batched centered Gram: HGH with H = I - 1/m
-    sub = G[idx[:, :, None], idx[:, None, :]]  # (C, m, m)
-    row_mean = sub.mean(dim=2, keepdim=True)
-    col_mean = sub.mean(dim=1, keepdim=True)
-    all_mean = sub.mean(dim=(1, 2), keepdim=True)
-    centered = sub - row_mean - col_mean + all_mean
-    ev = torch.linalg.eigvalsh(centered.cpu() if X.device.type == "cpu" else centered)
-    max_ev = ev[..., -1]
-    best = int(torch.argmin(max_ev))
-    rows = idx[best]
-    return Xf[rows].mean(dim=0).to(X.dtype)
+    # enumerate subsets in bounded blocks: materializing all C(n, m)
+    # (m, m) Grams at once OOMs for larger n (the search stays exact and
+    # exponential-time by definition — same as the reference smea.py:63-107)
+    BLOCK = 1 << 16
+    best_ev = math.inf
+    best_rows: Optional[torch.Tensor] = None
+    it = itertools.combinations(range(n), m)
+    while True:
+        combos = list(itertools.islice(it, BLOCK))
+        if not combos:
+            break
+        idx = torch.tensor(combos, device=X.device, dtype=torch.long)  # (B, m)
+        # batched centered Gram: HGH with H = I - 1/m
+        sub = G[idx[:, :, None], idx[:, None, :]]  # (B, m, m)
+        row_mean = sub.mean(dim=2, keepdim=True)
+        col_mean = sub.mean(dim=1, keepdim=True)
+        all_mean = sub.mean(dim=(1, 2), keepdim=True)
+        centered = sub - row_mean - col_mean + all_mean
+        ev = torch.linalg.eigvalsh(
+            centered.cpu() if X.device.type == "cpu" else centered
+        )
+        max_ev = ev[..., -1]
+        b = int(torch.argmin(max_ev))
+        v = float(max_ev[b])
+        # ties resolve to the lexicographically-smallest subset (blocks are
+        # generated in lexicographic order, so strict < suffices)
+        if v < best_ev:
+            best_ev = v
+            best_rows = idx[b]
+    assert best_rows is not None
+    return Xf[best_rows].mean(dim=0).to(X.dtype)
 
 
 # ---------------------------------------------------------------------------
