@@ -93,9 +93,14 @@ def main():
             cur.wait_stream(s_med)
             cur.wait_stream(s_krum)
             return outs
-        if args.op in ("both", "median"):
+        if args.op == "both":
+            # fused: Gram + selection first, then one X pass yields both
+            # the median and the winners' mean (sharded.median_multi_krum)
+            outs.extend(sharded.median_multi_krum(X, args.f, args.q))
+            return outs
+        if args.op == "median":
             outs.append(sharded.median(X))
-        if args.op in ("both", "krum"):
+        if args.op == "krum":
             outs.append(sharded.multi_krum(X, args.f, args.q))
         return outs
 
