@@ -93,14 +93,9 @@ def main():
             cur.wait_stream(s_med)
             cur.wait_stream(s_krum)
             return outs
-        if args.op == "both":
-            # fused: Gram + selection first, then one X pass yields both
-            # the median and the winners' mean (sharded.median_multi_krum)
-            outs.extend(sharded.median_multi_krum(X, args.f, args.q))
-            return outs
-        if args.op == "median":
+        if args.op in ("both", "median"):
             outs.append(sharded.median(X))
-        if args.op == "krum":
+        if args.op in ("both", "krum"):
             outs.append(sharded.multi_krum(X, args.f, args.q))
         return outs
 

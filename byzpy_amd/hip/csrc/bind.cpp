@@ -41,9 +41,6 @@ void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
 template <typename T>
 void launch_caf_colsum(const T*, const float*, const float*, const float*,
                        float*, int, long, hipStream_t);
-void launch_colsel_median_gather_bf16(const __hip_bfloat16*, __hip_bfloat16*,
-                                      __hip_bfloat16*, const int*, int, int,
-                                      long, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
 void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
@@ -434,27 +431,6 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
 
 }  // namespace
 
-// -- fused median + Multi-Krum gather (one X pass) --------------------------
-
-std::tuple<torch::Tensor, torch::Tensor> median_gather(torch::Tensor X,
-                                                       torch::Tensor idx) {
-  check_matrix(X);
-  TORCH_CHECK(X.scalar_type() == torch::kBFloat16,
-              "median_gather: bf16 only");
-  const int n = (int)X.size(0);
-  const long d = (long)X.size(1);
-  TORCH_CHECK(n <= 64 && (d % 2) == 0,
-              "median_gather needs n <= 64 and even d");
-  TORCH_CHECK(idx.is_cuda() && idx.scalar_type() == torch::kInt32 &&
-              idx.is_contiguous() && idx.numel() >= 1 && idx.numel() <= 64);
-  auto med = torch::empty({d}, X.options());
-  auto gm = torch::empty({d}, X.options());
-  launch_colsel_median_gather_bf16(
-      bf16_ptr(X), bf16_ptr_mut(med), bf16_ptr_mut(gm), idx.data_ptr<int>(),
-      (int)idx.numel(), n, d, cur_stream());
-  return {med, gm};
-}
-
 // -- CAF fused power-iteration pair --------------------------
 
 torch::Tensor caf_matvec(torch::Tensor X, torch::Tensor mu, torch::Tensor v) {
@@ -521,7 +497,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_mean", &bucket_mean);
   m.def("gram", &gram);
   m.def("krum_select", &krum_select);
-  m.def("median_gather", &median_gather);
   m.def("caf_matvec", &caf_matvec);
   m.def("caf_colsum", &caf_colsum, py::arg("X"), py::arg("a"),
         py::arg("mu") = c10::nullopt, py::arg("scale") = c10::nullopt);

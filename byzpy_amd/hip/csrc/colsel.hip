@@ -239,30 +239,14 @@ DEV u32 extract_at_pk(const u32 (&v)[P], int pos) {
 // independent sorting networks; QUADS=false: one pair (4 B/lane).
 // MODE: MEDIAN or TRIMMED (order statistics read straight off the sorted
 // keys; TRIMMED unpacks the kept range and sums in f32).
-// GATHER (MEDIAN only): the same X pass also accumulates the f32 mean of
-// a selected row subset (Multi-Krum winners, known before launch from the
-// (n, n) Gram) — the bench's separate gather-mean pass re-read all of X
-// for 12 rows' worth of output. idx/q build a 64-bit row mask per thread.
-template <int P, bool QUADS, int MODE = MEDIAN, bool GATHER = false>
+template <int P, bool QUADS, int MODE = MEDIAN>
 __global__ void
 // QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
 // so the allocator doesn't cap at 128 and spill
 __launch_bounds__(256, (QUADS && P >= 64) ? 3 : 4)
 colsel_pk_median_bf16(const unsigned short* __restrict__ X,
                                       unsigned short* __restrict__ out, int n,
-                                      long d, int f,
-                                      const int* __restrict__ gidx = nullptr,
-                                      int gq = 0,
-                                      unsigned short* __restrict__ gout = nullptr) {
-  u32 sel_lo = 0, sel_hi = 0;
-  float inv_q = 0.0f;
-  if (GATHER) {
-    unsigned long long sel = 0ull;
-    for (int j = 0; j < gq; ++j) sel |= 1ull << gidx[j];
-    sel_lo = vecify((u32)(sel & 0xFFFFFFFFull));
-    sel_hi = vecify((u32)(sel >> 32));
-    inv_q = 1.0f / (float)(gq > 0 ? gq : 1);
-  }
+                                      long d, int f) {
   const long npairs = d >> 1;
   const long nunits = QUADS ? (npairs >> 1) : npairs;
   const long unit0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -288,18 +272,6 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
       }
       p += (i + 1 < n_walk) ? rowstride : 0;
       __builtin_amdgcn_sched_barrier(0);
-    }
-    float g0 = 0.0f, g1 = 0.0f;
-    if (GATHER) {
-      // accumulate the selected rows BEFORE the key transform destroys
-      // the raw bf16 bits; bf16 -> f32 is bits << 16
-#pragma unroll
-      for (int i = 0; i < P; ++i) {
-        const u32 bit = ((i < 32 ? sel_lo : sel_hi) >> (i & 31)) & 1u;
-        const u32 m = v[i] & (0u - bit);  // zero when unselected
-        g0 += __uint_as_float((m & 0xFFFFu) << 16);
-        g1 += __uint_as_float(m & 0xFFFF0000u);
-      }
     }
     const int nv = vecify(n);
 #pragma unroll
@@ -401,15 +373,6 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
       o.s[0] = c0.s;
       o.s[1] = c1.s;
       outw[pair + half] = o.w;
-      if (GATHER && half == 0) {
-        union { unsigned short s[2]; u32 w; } og;
-        union { unsigned short s; __hip_bfloat16 h; } d0, d1;
-        d0.h = __float2bfloat16(g0 * inv_q);
-        d1.h = __float2bfloat16(g1 * inv_q);
-        og.s[0] = d0.s;
-        og.s[1] = d1.s;
-        reinterpret_cast<u32*>(gout)[pair] = og.w;
-      }
     }
   }
 }
@@ -541,25 +504,6 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
       hipLaunchKernelGGL((colsel_lds_kernel<MEAMED, T>), dim3(grid),
                          dim3(LDS_THREADS), lds, stream, X, out, n, d, f, P);
   }
-}
-
-void launch_colsel_median_gather_bf16(const __hip_bfloat16* X,
-                                      __hip_bfloat16* med_out,
-                                      __hip_bfloat16* gather_out,
-                                      const int* idx, int q, int n, long d,
-                                      hipStream_t stream) {
-  // fused median + selected-rows mean in ONE X pass (n <= 64, even d —
-  // enforced by the binding); always the P=64 packed kernel
-  const int block = 256;
-  const long units = d >> 1;
-  const long want = (units + block - 1) / block;
-  const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
-  const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
-  unsigned short* Ou = reinterpret_cast<unsigned short*>(med_out);
-  unsigned short* Gu = reinterpret_cast<unsigned short*>(gather_out);
-  hipLaunchKernelGGL((colsel_pk_median_bf16<64, false, MEDIAN, true>),
-                     dim3(grid), dim3(block), 0, stream, Xu, Ou, n, d, 0, idx,
-                     q, Gu);
 }
 
 void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,

@@ -59,28 +59,6 @@ def multi_krum(X_local: torch.Tensor, f: int, q: int) -> torch.Tensor:
     return D.mean_rows(X_local, winners.to(torch.int32))
 
 
-def median_multi_krum(X_local: torch.Tensor, f: int, q: int):
-    """Fused bench step (BASELINE config 2): Gram + winner selection run
-    FIRST (they only need the (n, n) statistics), then ONE pass over X
-    produces both the coordinate median and the winners' mean — the
-    separate gather-mean pass used to re-read all of X for q rows of
-    output. Returns (median_shard, multi_krum_shard)."""
-    n, d = X_local.shape
-    if (
-        X_local.is_cuda
-        and X_local.dtype == torch.bfloat16
-        and n <= 64
-        and d % 2 == 0
-    ):
-        from byzpy_amd.hip import require
-
-        ext = require()
-        G = _global_gram(X_local)
-        winners = ext.krum_select(G, int(f), int(q))
-        return ext.median_gather(X_local.contiguous(), winners)
-    return median(X_local), multi_krum(X_local, f, q)
-
-
 def krum(X_local: torch.Tensor, f: int) -> torch.Tensor:
     n = X_local.shape[0]
     G = _global_gram(X_local)

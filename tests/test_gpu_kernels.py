@@ -322,25 +322,3 @@ def test_every_preagg_on_device(preagg_name):
     assert len(out) == len(ref)
     for a, b in zip(out, ref):
         assert (a.cpu().float() - b).norm() / b.norm().clamp_min(1e-6) < 0.02
-
-
-class TestMedianGatherFused:
-    @pytest.mark.parametrize("n,d,q", [(64, 8192, 12), (16, 4096, 5), (7, 1000, 3)])
-    def test_matches_separate_paths(self, n, d, q):
-        from byzpy_amd.hip import require
-
-        X = _rand(n, d, torch.bfloat16, seed=91)
-        idx = torch.randperm(n, device="cuda")[:q].to(torch.int32).contiguous()
-        med, gm = require().median_gather(X, idx)
-        assert torch.equal(med, D.median(X))
-        ref = X.float()[idx.long()].mean(dim=0)
-        assert torch.allclose(gm.float(), ref, atol=3e-2, rtol=1e-2)
-
-    def test_sharded_fused_step(self):
-        from byzpy_amd.parallel import sharded
-
-        X = _rand(64, 16384, torch.bfloat16, seed=92)
-        med, mk = sharded.median_multi_krum(X, 16, 12)
-        assert torch.equal(med, D.median(X))
-        ref_mk = D.multi_krum(X, 16, 12)
-        assert torch.allclose(mk.float(), ref_mk.float(), atol=3e-2, rtol=1e-2)
