@@ -388,7 +388,10 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
 // ---------------------------------------------------------------------------
 
 constexpr int LDS_COLS = 64;
-constexpr int LDS_THREADS = 1024;
+// 512 threads (8 waves): halves barrier cost vs 16 waves and lets 2-4
+// blocks co-reside per CU at small P (LDS is the binding resource at
+// P=512: 128 KB -> 1 block)
+constexpr int LDS_THREADS = 512;
 
 template <int MODE, typename T>
 __global__ void __launch_bounds__(LDS_THREADS)
@@ -411,17 +414,21 @@ colsel_lds_kernel(const T* __restrict__ X, T* __restrict__ out,
   }
   __syncthreads();
 
-  // batched bitonic: P/2 compare sites x LDS_COLS columns per substep
+  // batched bitonic: P/2 compare sites x LDS_COLS columns per substep.
+  // j is always a power of two, so the site decomposition is pure shifts
+  // (an integer division per site per substage dominated this kernel).
   for (int k = 2; k <= P; k <<= 1) {
-    for (int j = k >> 1; j > 0; j >>= 1) {
+    for (int jl = 31 - __clz(k >> 1); jl >= 0; --jl) {
+      const int j = 1 << jl;
+      const int k_ = k;
       for (int idx = t; idx < (P >> 1) * LDS_COLS; idx += LDS_THREADS) {
-        const int site = idx / LDS_COLS;
-        const int c = idx % LDS_COLS;
+        const int site = idx >> 6;          // LDS_COLS == 64
+        const int c = idx & 63;
         // site s enumerates pairs (i, i^j) with i^j > i:
-        // i = (s / j) * 2j + (s % j)
-        const int i = ((site / j) * (j << 1)) + (site % j);
+        // i = (s >> jl) << (jl+1) | (s & (j-1))
+        const int i = ((site >> jl) << (jl + 1)) | (site & (j - 1));
         const int l = i ^ j;
-        const bool asc = (i & k) == 0;
+        const bool asc = (i & k_) == 0;
         float* a = &buf[i * LDS_COLS + c];
         float* b = &buf[l * LDS_COLS + c];
         const float av = *a, bv = *b;
