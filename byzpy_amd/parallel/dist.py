@@ -29,12 +29,18 @@ def get_world_size() -> int:
 
 def init_from_env(timeout_s: float = 600.0) -> int:
     """Initialize from torchrun-style env vars; returns local rank. Uses
-    RCCL when a GPU is visible, gloo otherwise. Safe to call twice."""
+    the composite backend (gloo for CPU tensors, RCCL for CUDA tensors)
+    when a GPU is visible, plain gloo otherwise — a GPU-visible host can
+    still all-reduce CPU tensors (a bare "nccl" group cannot). Override
+    with BYZPY_DIST_BACKEND. Safe to call twice."""
     if is_initialized():
         return int(os.environ.get("LOCAL_RANK", 0))
     if "RANK" not in os.environ:
         return 0  # single-process mode: no process group
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get(
+        "BYZPY_DIST_BACKEND",
+        "cpu:gloo,cuda:nccl" if torch.cuda.is_available() else "gloo",
+    )
     local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", 0)))
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
