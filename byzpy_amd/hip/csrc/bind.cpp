@@ -35,6 +35,8 @@ void launch_weiszfeld_update(const T*, const float*, const float*, float*,
 template <typename T>
 void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
+void launch_colsel_median_radix_bf16(const __hip_bfloat16*, __hip_bfloat16*,
+                                     unsigned int*, int, long, hipStream_t);
 template <typename T>
 void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
                        hipStream_t);
@@ -71,9 +73,25 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
   check_matrix(X);
   const int n = (int)X.size(0);
   const long d = (long)X.size(1);
-  TORCH_CHECK(n >= 1 && n <= 512, "colsel supports 1 <= n <= 512, got ", n);
+  // measured crossover vs the cooperative LDS path: radix wins for
+  // n > ~192 (and is the ONLY path past 512)
+  const bool radix_ok =
+      mode == 0 && n > 192 && n <= 65535 && X.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(n >= 1 && (n <= 512 || radix_ok),
+              "colsel supports 1 <= n <= 512 (bf16 MEDIAN: n <= 65535), got ",
+              n);
   TORCH_CHECK(f >= 0 && 2 * f < n, "bad f for colsel");
   auto out = torch::empty({(long)d}, X.options());
+  if (radix_ok) {
+    // 2-pass streaming radix select (see colsel.hip rsel_* kernels)
+    auto mark = torch::empty({(long)d * 2},
+                             X.options().dtype(torch::kInt32));
+    launch_colsel_median_radix_bf16(
+        bf16_ptr(X), bf16_ptr_mut(out),
+        reinterpret_cast<unsigned int*>(mark.data_ptr<int>()), n, d,
+        cur_stream());
+    return out;
+  }
   if (X.scalar_type() == torch::kFloat32)
     launch_colsel_f32(X.data_ptr<float>(), out.data_ptr<float>(), n, d,
                       (int)mode, (int)f, cur_stream());

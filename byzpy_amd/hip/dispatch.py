@@ -33,7 +33,12 @@ def _gpu(X: torch.Tensor) -> bool:
 
 
 def median(X: torch.Tensor) -> torch.Tensor:
-    if _gpu(X) and X.shape[0] <= COLSEL_MAX_N:
+    n = X.shape[0]
+    if _gpu(X) and (
+        n <= COLSEL_MAX_N or (X.dtype == torch.bfloat16 and n <= 65535)
+    ):
+        # n > 64 bf16 routes to the 2-pass streaming radix select inside
+        # the extension (no upper cap below 65536 rows)
         return _hip.require().colsel(X.contiguous(), _COLSEL_MEDIAN, 0)
     return F.median(X)
 

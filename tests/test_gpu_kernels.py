@@ -322,3 +322,38 @@ def test_every_preagg_on_device(preagg_name):
     assert len(out) == len(ref)
     for a, b in zip(out, ref):
         assert (a.cpu().float() - b).norm() / b.norm().clamp_min(1e-6) < 0.02
+
+
+class TestRadixMedianLargeN:
+    """bf16 median for n > 64: 2-pass streaming radix select vs fp32 oracle.
+    bf16 keys are exact (finite bf16 set), so the median of bf16 inputs
+    equals the oracle's median of the same values exactly."""
+
+    @pytest.mark.parametrize("n", [65, 100, 128, 257, 512, 700, 4096])
+    def test_parity(self, n):
+        X = _rand(n, 3001, torch.bfloat16, seed=n)
+        out = D.median(X)
+        ref = F.median(X.float().cpu()).bfloat16()
+        assert torch.equal(out.cpu(), ref), f"n={n}"
+
+    def test_with_inf_rows(self):
+        X = _rand(100, 2048, torch.bfloat16, seed=3)
+        X[7] = float("inf")
+        X[13] = float("-inf")
+        out = D.median(X)
+        ref = F.median(X.float().cpu()).bfloat16()
+        assert torch.equal(out.cpu(), ref)
+
+    def test_ties_heavy(self):
+        # constant-ish columns: every bucket collapses to one value
+        X = torch.ones(300, 1024, dtype=torch.bfloat16, device="cuda")
+        X[:150] = 2.0
+        out = D.median(X)
+        ref = F.median(X.float().cpu()).bfloat16()
+        assert torch.equal(out.cpu(), ref)
+
+    def test_odd_d_and_small_d(self):
+        X = _rand(129, 77, torch.bfloat16, seed=5)
+        out = D.median(X)
+        ref = F.median(X.float().cpu()).bfloat16()
+        assert torch.equal(out.cpu(), ref)
