@@ -83,7 +83,7 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
   TORCH_CHECK(f >= 0 && 2 * f < n, "bad f for colsel");
   auto out = torch::empty({(long)d}, X.options());
   if (radix_ok) {
-    // 2-pass streaming radix select (see colsel.hip rsel_* kernels)
+    // streaming radix select (see colsel.hip rsel_* kernels)
     auto mark = torch::empty({(long)d * 2},
                              X.options().dtype(torch::kInt32));
     launch_colsel_median_radix_bf16(

@@ -357,3 +357,4 @@ class TestRadixMedianLargeN:
         out = D.median(X)
         ref = F.median(X.float().cpu()).bfloat16()
         assert torch.equal(out.cpu(), ref)
+
