@@ -270,6 +270,10 @@ class MeshRemoteContext(NodeContext):
     async def shutdown(self) -> None:
         if self._monitor is not None:
             self._monitor.cancel()
+            try:
+                await self._monitor
+            except asyncio.CancelledError:
+                pass
         for w in self._out.values():
             w.close()
         if self._server is not None:
