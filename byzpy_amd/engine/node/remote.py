@@ -139,6 +139,10 @@ class RemoteNodeClient:
     async def close(self) -> None:
         if self._recv_task is not None:
             self._recv_task.cancel()
+            try:
+                await self._recv_task
+            except asyncio.CancelledError:
+                pass
         if self._writer is not None:
             self._writer.close()
 
@@ -192,6 +196,10 @@ class RemoteContext(NodeContext):
     async def shutdown(self) -> None:
         if self._pump is not None:
             self._pump.cancel()
+            try:
+                await self._pump
+            except asyncio.CancelledError:
+                pass
         await self.client.close()
 
 

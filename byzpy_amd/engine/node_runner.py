@@ -57,6 +57,10 @@ class NodeRunner:
     async def stop(self) -> None:
         if self._auto_task is not None:
             self._auto_task.cancel()
+            try:
+                await self._auto_task
+            except asyncio.CancelledError:
+                pass
         await self.transport.stop()
 
 
