@@ -50,7 +50,7 @@ class LinearWorker:
             self.model.weight.add_(g.reshape(self.model.weight.shape), alpha=-self.lr)
 
     def error(self):
-        return float((self.model.weight - self.w_true).norm())
+        return float((self.model.weight.detach() - self.w_true).norm())
 
 
 class Saboteur:
