@@ -192,3 +192,33 @@ def test_multi_krum_q_equals_n_minus_f_is_selection_mean(p):
     assert torch.allclose(
         F.multi_krum(X, f, q), X[keep].mean(dim=0), atol=1e-4
     )
+
+
+@settings(max_examples=12, deadline=None)
+@given(matrix_params)
+def test_pooled_median_fuzz_parity(p):
+    """Pooled-vs-direct parity under randomized shapes AND chunk sizes
+    (regression net for the chunk-range arithmetic)."""
+    import asyncio
+
+    from byzpy_amd.graph.executor import OperatorExecutor
+    from byzpy_amd.graph.pool import ActorPool, ActorPoolConfig
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+
+    n, d, seed = p
+    import random as _r
+
+    chunk = _r.Random(seed).choice([1, 3, 17, 4096])
+    grads = list(_matrix(n, d, seed))
+    agg = CoordinateWiseMedian(chunk_size=chunk)
+    direct = agg.aggregate(grads)
+
+    async def main():
+        pool = ActorPool(ActorPoolConfig(backend="thread", count=2))
+        await pool.start()
+        out = await OperatorExecutor(agg, pool=pool).run({"gradients": grads})
+        await pool.close()
+        return out
+
+    out = asyncio.run(main())
+    assert torch.allclose(out, direct, atol=1e-4)
