@@ -94,3 +94,91 @@ def test_robust_aggregator_survives_attack(agg):
 def test_plain_mean_is_destroyed():
     err = _train(MeanAggregator(), rounds=10)
     assert err > 1.0, f"attack unexpectedly harmless to plain mean: err={err}"
+
+
+def test_ps_random_config_sweep():
+    """Seeded mini-sweep of PS configurations (aggregator x pre-agg x
+    attack): every combination must produce a finite, correctly-shaped
+    aggregate. (A 25-config version of this sweep runs ad hoc; this seeded
+    5-config subset guards regressions.)"""
+    import asyncio
+    import random
+
+    import torch
+
+    from byzpy_amd.aggregators import (
+        CenteredClipping,
+        CoordinateWiseMedian,
+        GeometricMedian,
+        MeanOfMedians,
+        MultiKrum,
+    )
+    from byzpy_amd.attacks import EmpireAttack, GaussianAttack, SignFlipAttack
+    from byzpy_amd.engine.node.actors import ByzantineNodeActor, HonestNodeActor
+    from byzpy_amd.engine.node.base import ByzantineNode, HonestNode
+    from byzpy_amd.engine.parameter_server.ps import ParameterServer
+    from byzpy_amd.pre_aggregators import ARC, Clipping
+
+    class H(HonestNode):
+        def __init__(self, seed, d):
+            g = torch.Generator().manual_seed(seed)
+            self.g0 = torch.randn(d, generator=g)
+
+        def next_batch(self):
+            return None, None
+
+        def honest_gradient(self, x, y):
+            return self.g0
+
+        def apply_server_gradient(self, grad):
+            self.last = grad
+
+    class B(ByzantineNode):
+        def __init__(self, atk):
+            self.atk = atk
+
+        def next_batch(self):
+            return None, None
+
+        def byzantine_gradient(self, x, y, honest_grads=None):
+            if self.atk.uses_honest_grads:
+                return self.atk.apply(honest_grads=honest_grads)
+            return self.atk.apply(base_grad=honest_grads[0])
+
+        def apply_server_gradient(self, grad):
+            pass
+
+    rng = random.Random(3)
+
+    async def run_one(i):
+        d = rng.randint(8, 64)
+        agg = rng.choice(
+            [
+                CoordinateWiseMedian(),
+                MeanOfMedians(1),
+                MultiKrum(1, 2),
+                GeometricMedian(),
+                CenteredClipping(c_tau=1.0),
+            ]
+        )
+        pre = rng.choice([None, Clipping(5.0), ARC(1)])
+        atk = rng.choice([EmpireAttack(scale=-5.0), SignFlipAttack(), GaussianAttack(seed=1)])
+        honest = [
+            await HonestNodeActor.spawn(H, 10 + j, d, backend="thread")
+            for j in range(4)
+        ]
+        byz = [await ByzantineNodeActor.spawn(B, atk, backend="thread")]
+        ps = ParameterServer(honest, byz, agg, pre_aggregator=pre)
+        try:
+            out = await ps.round()
+            assert out.shape == (d,)
+            assert torch.isfinite(out).all()
+        finally:
+            for nd in honest + byz:
+                await nd.close()
+
+    async def main():
+        for i in range(5):
+            await run_one(i)
+
+    asyncio.run(main())
