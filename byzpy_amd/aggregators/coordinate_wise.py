@@ -63,6 +63,11 @@ class _FeatureChunkedAggregator(Aggregator):
 
 
 class CoordinateWiseMedian(_FeatureChunkedAggregator):
+    """Per-coordinate median over the worker axis; robust to
+    floor((n-1)/2) byzantine workers (reference median.py:28). Both the
+    direct and pooled paths compute the TRUE median (even n: mean of the
+    two middle values)."""
+
     name = "coordinate-wise-median"
     default_chunk_size = 8192
 
@@ -74,6 +79,9 @@ class CoordinateWiseMedian(_FeatureChunkedAggregator):
 
 
 class CoordinateWiseTrimmedMean(_FeatureChunkedAggregator):
+    """Per-coordinate sort, drop f from each end, mean the middle n-2f
+    (Yin et al. 2018; reference trimmed_mean.py:27)."""
+
     name = "coordinate-wise-trimmed-mean"
     default_chunk_size = 4096
 
