@@ -58,4 +58,15 @@ class P2PByzantineMixin:
     attack: Any
 
     def p2p_broadcast_vector(self, neighbor_vectors: Sequence[torch.Tensor]) -> torch.Tensor:
-        return self.attack.apply(honest_grads=list(neighbor_vectors))
+        vecs = list(neighbor_vectors)
+        # derive the attack's inputs from its declared uses_* flags
+        # (reference mixin.py:93-105): base_grad attacks see the mean of
+        # the vectors they observed
+        kwargs = {}
+        if getattr(self.attack, "uses_honest_grads", False):
+            kwargs["honest_grads"] = vecs
+        if getattr(self.attack, "uses_base_grad", False):
+            kwargs["base_grad"] = torch.stack(vecs).float().mean(dim=0).to(vecs[0].dtype)
+        if not kwargs:
+            kwargs["honest_grads"] = vecs
+        return self.attack.apply(**kwargs)

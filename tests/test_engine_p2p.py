@@ -164,3 +164,31 @@ def test_p2p_facade_kwargs_compat():
         lr=0.123,
     )
     assert p2p._runner.lr == 0.123
+
+
+def test_p2p_base_grad_attack():
+    """Regression: base_grad attacks (SignFlip) in P2P gossip derive their
+    input from the mixin's uses_* flags instead of always honest_grads."""
+    from byzpy_amd.attacks import SignFlipAttack
+    from byzpy_amd.engine.peer_to_peer.mixin import P2PByzantineMixin
+
+    class Byz(P2PByzantineMixin):
+        def __init__(self):
+            self.attack = SignFlipAttack(scale=-2.0)
+
+    vecs = [torch.ones(4), 3 * torch.ones(4)]
+    out = Byz().p2p_broadcast_vector(vecs)
+    # base_grad = mean(vecs) = 2; scaled by -2 -> -4
+    assert torch.allclose(out, -4.0 * torch.ones(4))
+
+    async def gossip():
+        honest = [TinyHonest(s) for s in range(3)]
+        p2p = PeerToPeer(honest, [Byz()], CoordinateWiseMedian(), lr=0.05)
+        await p2p.bootstrap()
+        for _ in range(2):
+            await p2p.round()
+        await p2p.shutdown()
+        for h in honest:
+            assert torch.isfinite(h.p2p_flat_params()).all()
+
+    asyncio.run(gossip())
