@@ -44,8 +44,16 @@ class RcclPeerToPeer:
         theta_half = self.half_step_fn().reshape(-1)
         if self.attack is not None:
             # byzantine rank: broadcast the attack vector instead (it sees
-            # its own honest half-step as context)
-            theta_half = self.attack.apply(honest_grads=[theta_half]).reshape(-1)
+            # its own honest half-step as context); inputs derive from the
+            # attack's uses_* flags (base_grad attacks get theta-half)
+            kwargs = {}
+            if getattr(self.attack, "uses_honest_grads", False):
+                kwargs["honest_grads"] = [theta_half]
+            if getattr(self.attack, "uses_base_grad", False):
+                kwargs["base_grad"] = theta_half
+            if not kwargs:
+                kwargs["honest_grads"] = [theta_half]
+            theta_half = self.attack.apply(**kwargs).reshape(-1)
         all_vecs = pdist.all_gather_rows(theta_half.reshape(1, -1))  # (world, d)
         neighbors = self.topology.in_neighbors(rank)
         rows = [rank] + [j for j in neighbors if j != rank]
