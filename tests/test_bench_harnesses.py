@@ -91,3 +91,20 @@ def test_kernels_bench_requires_gpu_gracefully():
 def test_graft_entry_build_and_importable():
     r = _run(["-c", "import __graft_entry__; __graft_entry__.build()"], timeout=900)
     assert r.returncode == 0, r.stderr[-800:]
+
+
+def test_bench_metric_matches_baseline_contract():
+    """bench.py's metric string must describe BASELINE.json's metric
+    (aggregated-grads/sec for Median & Krum) and the config must name the
+    BASELINE config-2 shape."""
+    import json
+
+    r = _run(["bench.py", "--steps", "1", "--warmup", "0", "--d", "10000"])
+    assert r.returncode == 0, r.stderr[-500:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    assert "aggregated-grads/sec" in d["metric"]
+    assert "Median" in d["metric"] and "Krum" in d["metric"]
+    assert d["higher_is_better"] is True
+    assert d["dtype"] in ("bf16", "f32")
+    assert "Multi-Krum" in d["config"]["model"]
+    assert d["config"]["n_workers"] == 64
