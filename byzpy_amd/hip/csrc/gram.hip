@@ -5,10 +5,12 @@
 // design goal is exactly-once HBM traffic + enough blocks to fill 256 CUs:
 //   - one workgroup = one 64x64 output tile x one K-slab (split-K across
 //     the grid; f32 atomicAdd combine into the tiny n x n output),
-//   - 16 waves per block (4x4 of 16x16 MFMA tiles); A-rows and B-rows of
-//     the same tile are shared through L2; a __syncthreads() every few
-//     K-steps keeps the waves' streams inside one L2 window so HBM traffic
-//     stays ~1x X,
+//   - 16 waves per block (4x4 of 16x16 MFMA tiles). SHIPPING PATH: the
+//     K-chunk is STAGED THROUGH LDS (gram_*_lds_kernel: 16 KB double
+//     buffers, XOR-swizzled slots, cooperative T14 load/write split,
+//     single image for the diagonal tile) so 16 waves re-reading the same
+//     tile rows cost LDS bandwidth, not 8x-amplified L2/HBM traffic. The
+//     direct kernels below the LDS ones are the alignment fallback,
 //   - bf16 path: v_mfma_f32_16x16x32_bf16 (both fragments are 8 contiguous
 //     k-elements of a row of X = one 16 B load);
 //     f32 path: v_mfma_f32_16x16x4_f32 (exact f32 at the vector rate,
