@@ -24,6 +24,20 @@ _ids = itertools.count()
 
 
 def _worker(conn) -> None:
+    # cap intra-op threads: N actor children each defaulting to all cores
+    # oversubscribe catastrophically (13 MLP workers on 8 cores: 1020 ->
+    # 94 ms/round with a cap). Override with BYZPY_AMD_ACTOR_TORCH_THREADS.
+    try:
+        import os as _os
+
+        import torch as _torch
+
+        _want = _os.environ.get("BYZPY_AMD_ACTOR_TORCH_THREADS")
+        _torch.set_num_threads(
+            max(1, int(_want) if _want else (_os.cpu_count() or 4) // 4)
+        )
+    except Exception:  # noqa: BLE001 — never block actor startup on this
+        pass
     obj = None
     mailboxes: Dict[str, stdlib_queue.Queue] = {}
     while True:
