@@ -266,6 +266,11 @@ class ActorPool:
                         await self.workers[idx].respawn()
                     except Exception:  # noqa: BLE001 — stays broken, others serve
                         pass
+            except (asyncio.CancelledError, KeyboardInterrupt, SystemExit):
+                # cancellation/shutdown must propagate immediately, never be
+                # recorded as a retryable worker failure (the finally clause
+                # still releases the worker)
+                raise
             except BaseException as e:  # noqa: BLE001
                 last_err = e
             finally:

@@ -76,8 +76,10 @@ class MessageAwareNodeScheduler(NodeScheduler):
         self._waiters: Dict[str, List[asyncio.Future]] = {}
 
     def deliver_message(self, message_type: str, payload: Any) -> None:
+        # skip and prune futures that are already done (e.g. a wait_for
+        # timed out): deliver to the first still-live waiter
         waiters = self._waiters.get(message_type)
-        if waiters:
+        while waiters:
             fut = waiters.pop(0)
             if not fut.done():
                 fut.set_result(payload)

@@ -185,7 +185,9 @@ def geometric_median(
     eps: float = 1e-12,
     init: str = "median",
 ) -> torch.Tensor:
-    if not _gpu(X):
+    if not _gpu(X) or X.shape[0] > 1024:
+        # extension's weiszfeld_iter TORCH_CHECKs n <= 1024; the torch
+        # functional path runs fine on-device for larger n
         return F.geometric_median(X, tol=tol, max_iter=max_iter, eps=eps, init=init)
     ext = _hip.require()
     Xc = X.contiguous()
@@ -200,7 +202,10 @@ def geometric_median(
         for _ in range(steps):
             z = ext.weiszfeld_iter(Xc, z, float(eps), shift)
         it += steps
-        if float(shift) <= tol:  # one sync per `poll` iterations
+        # `shift` holds the accumulated SQUARED step ||dz||^2 (rowops.hip
+        # weiszfeld_iter atomicAdd), so compare against tol^2 to match the
+        # CPU oracle's ||dz|| <= tol and parallel/sharded.py
+        if float(shift) <= tol * tol:  # one sync per `poll` iterations
             break
     return z.to(X.dtype)
 
@@ -213,7 +218,8 @@ def centered_clipping(
     eps: float = 1e-12,
     init: str = "mean",
 ) -> torch.Tensor:
-    if not _gpu(X):
+    if not _gpu(X) or X.shape[0] > 1024:
+        # extension's cc_iter TORCH_CHECKs n <= 1024; fall back on-device
         return F.centered_clipping(X, c_tau=c_tau, M=M, eps=eps, init=init)
     ext = _hip.require()
     Xc = X.contiguous()

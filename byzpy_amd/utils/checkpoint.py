@@ -32,10 +32,13 @@ def _save_tensors(path: Path, tensors: Dict[str, torch.Tensor]) -> None:
 def _load_tensors(path: Path) -> Dict[str, torch.Tensor]:
     try:
         from safetensors.torch import load_file
-
-        return load_file(str(path))
-    except (ImportError, Exception):
+    except ImportError:
+        # environment without safetensors: files were written by the
+        # torch.save fallback in _save_tensors
         return torch.load(str(path), map_location="cpu", weights_only=True)
+    # safetensors installed -> the file is safetensors-format; let a
+    # corruption/truncation error propagate with its original message
+    return load_file(str(path))
 
 
 def save_checkpoint(
