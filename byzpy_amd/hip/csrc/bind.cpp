@@ -37,6 +37,18 @@ void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
 void launch_colsel_median_radix_bf16(const __hip_bfloat16*, __hip_bfloat16*,
                                      unsigned int*, int, long, hipStream_t);
+// rsel.hip: generic multi-pass radix-select for the other large-n modes
+void launch_rsel_trimmed_bf16(const __hip_bfloat16*, __hip_bfloat16*,
+                              unsigned int*, int, long, int, hipStream_t);
+void launch_rsel_trimmed_f32(const float*, float*, unsigned int*, int, long,
+                             int, hipStream_t);
+void launch_rsel_median_f32(const float*, float*, unsigned int*, int, long,
+                            hipStream_t);
+void launch_rsel_meamed_bf16(const __hip_bfloat16*, __hip_bfloat16*,
+                             unsigned int*, float*, int, long, int,
+                             hipStream_t);
+void launch_rsel_meamed_f32(const float*, float*, unsigned int*, float*, int,
+                            long, int, hipStream_t);
 template <typename T>
 void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
                        hipStream_t);
@@ -73,23 +85,50 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
   check_matrix(X);
   const int n = (int)X.size(0);
   const long d = (long)X.size(1);
-  // measured crossover vs the cooperative LDS path: radix wins for
-  // n > ~192 (and is the ONLY path past 512)
-  const bool radix_ok =
-      mode == 0 && n > 192 && n <= 65535 && X.scalar_type() == torch::kBFloat16;
+  // measured crossover vs the cooperative LDS path: streaming radix wins
+  // for n > ~192 (and is the ONLY path past 512); every mode x dtype now
+  // has a radix path to n <= 65535
+  const bool radix_ok = n > 192 && n <= 65535;
   TORCH_CHECK(n >= 1 && (n <= 512 || radix_ok),
-              "colsel supports 1 <= n <= 512 (bf16 MEDIAN: n <= 65535), got ",
-              n);
+              "colsel supports 1 <= n <= 65535, got ", n);
   TORCH_CHECK(f >= 0 && 2 * f < n, "bad f for colsel");
   auto out = torch::empty({(long)d}, X.options());
   if (radix_ok) {
-    // streaming radix select (see colsel.hip rsel_* kernels)
-    auto mark = torch::empty({(long)d * 2},
-                             X.options().dtype(torch::kInt32));
-    launch_colsel_median_radix_bf16(
-        bf16_ptr(X), bf16_ptr_mut(out),
-        reinterpret_cast<unsigned int*>(mark.data_ptr<int>()), n, d,
-        cur_stream());
+    const bool bf16 = X.scalar_type() == torch::kBFloat16;
+    if (mode == 0 && bf16) {
+      // tuned 2-pass streaming radix select (colsel.hip rsel_pass1/2)
+      auto mark = torch::empty({(long)d * 2},
+                               X.options().dtype(torch::kInt32));
+      launch_colsel_median_radix_bf16(
+          bf16_ptr(X), bf16_ptr_mut(out),
+          reinterpret_cast<unsigned int*>(mark.data_ptr<int>()), n, d,
+          cur_stream());
+      return out;
+    }
+    // generic multi-pass engine (rsel.hip): per-column state scratch
+    auto state = torch::zeros({(long)d * 4}, X.options().dtype(torch::kInt32));
+    auto* st = reinterpret_cast<unsigned int*>(state.data_ptr<int>());
+    if (mode == 0) {  // f32 MEDIAN
+      launch_rsel_median_f32(X.data_ptr<float>(), out.data_ptr<float>(), st,
+                             n, d, cur_stream());
+    } else if (mode == 1) {
+      if (bf16)
+        launch_rsel_trimmed_bf16(bf16_ptr(X), bf16_ptr_mut(out), st, n, d,
+                                 (int)f, cur_stream());
+      else
+        launch_rsel_trimmed_f32(X.data_ptr<float>(), out.data_ptr<float>(),
+                                st, n, d, (int)f, cur_stream());
+    } else {
+      auto med = torch::empty({(long)d}, X.options().dtype(torch::kFloat32));
+      if (bf16)
+        launch_rsel_meamed_bf16(bf16_ptr(X), bf16_ptr_mut(out), st,
+                                med.data_ptr<float>(), n, d, (int)f,
+                                cur_stream());
+      else
+        launch_rsel_meamed_f32(X.data_ptr<float>(), out.data_ptr<float>(), st,
+                               med.data_ptr<float>(), n, d, (int)f,
+                               cur_stream());
+    }
     return out;
   }
   if (X.scalar_type() == torch::kFloat32)

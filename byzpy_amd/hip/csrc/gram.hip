@@ -460,6 +460,300 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Small-n Gram (n <= 32): wave-slab kernels. The 64x64-tile kernels above
+// amplify HBM traffic up to 8x at n < 64 (the tile is mostly padding:
+// r01_shape_sweep measured 937 GB/s at n=8 vs 6,091 at n=64). Here each
+// WAVE owns a disjoint K-slab covering ALL n rows, and — because Gram's A
+// and B operands are the same matrix — each fragment register is fed to
+// the MFMA as BOTH operands, so every X element is loaded exactly once
+// grid-wide with no padding traffic at all:
+//   - n <= 16 (bf16): v_mfma_f32_16x16x32_bf16 with SLAB PACKING: the 16
+//     fragment rows carry P = 16/n copies of the matrix at P different
+//     k-sub-slabs (fragment row p*n+i = X row i at sub-slab p). Output
+//     quadrant (p,p) accumulates sub-slab p's Gram; cross-slab quadrants
+//     are discarded. Load efficiency n*P/16 (100% at n in {8, 16}).
+//   - 16 < n <= 32 (bf16): v_mfma_f32_32x32x16_bf16, one slab per wave.
+//   - f32: same structure on v_mfma_f32_16x16x4_f32 / _32x32x2_f32 (exact
+//     f32 at the vector rate, guide §3).
+// Waves reduce their tiny C into an LDS accumulator; one atomicAdd per
+// element per block into the (n, n) output.
+// ---------------------------------------------------------------------------
+
+constexpr int SMALL_WAVES = 8;  // waves per block (512 threads)
+
+// bf16, n <= 16, packed. Each wave: K range [W*kpw, W*kpw+kpw) split into
+// P sub-slabs of `sub` elements (kpw = P * sub, sub % 8 == 0).
+__global__ void __launch_bounds__(SMALL_WAVES * 64, 4)
+gram_bf16_small16_kernel(const __hip_bfloat16* __restrict__ X,
+                         float* __restrict__ G, int n, long d, long kpw,
+                         int P) {
+  __shared__ float cbuf[16 * 16];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  for (int i = t; i < 16 * 16; i += SMALL_WAVES * 64) cbuf[i] = 0.0f;
+
+  const long W = (long)blockIdx.x * SMALL_WAVES + wave;
+  const long base = W * kpw;
+  const long sub = kpw / P;
+
+  const int frow = lane & 15;          // fragment row
+  const int i = (frow < P * n) ? frow % n : 0;   // X row (clamped)
+  const int p = (frow < P * n) ? frow / n : 0;   // sub-slab index
+  const long frag_k = (long)(lane >> 4) * 8;     // k offset within step
+  const long lane_base = base + (long)p * sub;   // this lane's sub-slab base
+  const __hip_bfloat16* src = X + (long)i * d + lane_base + frag_k;
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  if (base + kpw <= d) {
+    // whole wave fully in range: unguarded 16 B loads, K-unroll x4
+    long k = 0;
+    for (; k + 128 <= sub; k += 128) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        a[u] = *reinterpret_cast<const bf16x8*>(src + k + u * 32);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], a[u], acc, 0, 0, 0);
+    }
+    for (; k + 32 <= sub; k += 32) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(src + k);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, a, acc, 0, 0, 0);
+    }
+  } else if (base < d) {
+    // global-tail wave: per-element guards (k limit differs per sub-slab)
+    const long hi = d - lane_base;     // may be <= 0 for high-p lanes
+    for (long k = 0; k < sub; k += 32) {
+      bf16x8 a;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const long kk = k + frag_k + j;
+        a[j] = (kk < hi && kk < sub) ? *reinterpret_cast<const __bf16*>(src + k + j)
+                                     : (__bf16)0.0f;
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, a, acc, 0, 0, 0);
+    }
+  }
+
+  __syncthreads();
+  // C/D map for 16x16: col = lane&15, row = (lane>>4)*4 + r
+  const int cc = lane & 15;
+  const int pc = (cc < P * n) ? cc / n : -1;
+  const int j = cc - pc * n;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int rr = (lane >> 4) * 4 + r;
+    if (pc >= 0 && rr < P * n && rr / n == pc)
+      atomicAdd(&cbuf[(rr - pc * n) * n + j], acc[r]);
+  }
+  __syncthreads();
+  for (int idx = t; idx < n * n; idx += SMALL_WAVES * 64)
+    atomicAdd(&G[(idx / n) * (long)n + (idx % n)], cbuf[idx]);
+}
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// bf16, 16 < n <= 32: 32x32x16 MFMA, one slab per wave.
+__global__ void __launch_bounds__(SMALL_WAVES * 64, 4)
+gram_bf16_small32_kernel(const __hip_bfloat16* __restrict__ X,
+                         float* __restrict__ G, int n, long d, long kpw) {
+  __shared__ float cbuf[32 * 32];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  for (int i = t; i < 32 * 32; i += SMALL_WAVES * 64) cbuf[i] = 0.0f;
+
+  const long W = (long)blockIdx.x * SMALL_WAVES + wave;
+  const long base = W * kpw;
+
+  const int frow = lane & 31;
+  const int i = min(frow, n - 1);
+  const long frag_k = (long)(lane >> 5) * 8;  // two 8-element halves of K=16
+  const __hip_bfloat16* src = X + (long)i * d + base + frag_k;
+
+  f32x16 acc = {};
+  if (base + kpw <= d) {
+    long k = 0;
+    for (; k + 64 <= kpw; k += 64) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        a[u] = *reinterpret_cast<const bf16x8*>(src + k + u * 16);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[u], a[u], acc, 0, 0, 0);
+    }
+    for (; k + 16 <= kpw; k += 16) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(src + k);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, a, acc, 0, 0, 0);
+    }
+  } else if (base < d) {
+    const long hi = d - base;
+    for (long k = 0; k < kpw && k < hi; k += 16) {
+      bf16x8 a;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const long kk = k + frag_k + j;
+        a[j] = (kk < hi) ? *reinterpret_cast<const __bf16*>(src + k + j)
+                         : (__bf16)0.0f;
+      }
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, a, acc, 0, 0, 0);
+    }
+  }
+
+  __syncthreads();
+  // C/D map for 32x32: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int cc = lane & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rr = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    if (rr < n && cc < n) atomicAdd(&cbuf[rr * n + cc], acc[r]);
+  }
+  __syncthreads();
+  for (int idx = t; idx < n * n; idx += SMALL_WAVES * 64)
+    atomicAdd(&G[idx], cbuf[idx]);
+}
+
+// f32, n <= 16, packed: v_mfma_f32_16x16x4_f32 (1 float per lane per op).
+__global__ void __launch_bounds__(SMALL_WAVES * 64, 4)
+gram_f32_small16_kernel(const float* __restrict__ X, float* __restrict__ G,
+                        int n, long d, long kpw, int P) {
+  __shared__ float cbuf[16 * 16];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  for (int i = t; i < 16 * 16; i += SMALL_WAVES * 64) cbuf[i] = 0.0f;
+
+  const long W = (long)blockIdx.x * SMALL_WAVES + wave;
+  const long base = W * kpw;
+  const long sub = kpw / P;
+
+  const int frow = lane & 15;
+  const int i = (frow < P * n) ? frow % n : 0;
+  const int p = (frow < P * n) ? frow / n : 0;
+  const long lane_k = lane >> 4;  // k = k0 + lane_k, K-step 4
+  const long lane_base = base + (long)p * sub;
+  const float* src = X + (long)i * d + lane_base + lane_k;
+
+  // two accumulators break the 32-cycle dependent-accumulator chain
+  f32x4 acc0 = {0.0f, 0.0f, 0.0f, 0.0f};
+  f32x4 acc1 = {0.0f, 0.0f, 0.0f, 0.0f};
+  if (base + kpw <= d) {
+    long k = 0;
+    for (; k + 32 <= sub; k += 32) {
+      float a[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) a[u] = src[k + u * 4];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        if (u & 1)
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a[u], a[u], acc1, 0, 0, 0);
+        else
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a[u], a[u], acc0, 0, 0, 0);
+      }
+    }
+    for (; k + 4 <= sub; k += 4) {
+      const float a = src[k];
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, a, acc0, 0, 0, 0);
+    }
+  } else if (base < d) {
+    const long hi = d - lane_base;
+    for (long k = 0; k < sub; k += 4) {
+      const long kk = k + lane_k;
+      const float a = (kk < hi && kk < sub) ? src[k] : 0.0f;
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, a, acc0, 0, 0, 0);
+    }
+  }
+
+  __syncthreads();
+  const int cc = lane & 15;
+  const int pc = (cc < P * n) ? cc / n : -1;
+  const int j = cc - pc * n;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int rr = (lane >> 4) * 4 + r;
+    if (pc >= 0 && rr < P * n && rr / n == pc)
+      atomicAdd(&cbuf[(rr - pc * n) * n + j], acc0[r] + acc1[r]);
+  }
+  __syncthreads();
+  for (int idx = t; idx < n * n; idx += SMALL_WAVES * 64)
+    atomicAdd(&G[idx], cbuf[idx]);
+}
+
+// f32, 16 < n <= 32: v_mfma_f32_32x32x2_f32.
+__global__ void __launch_bounds__(SMALL_WAVES * 64, 4)
+gram_f32_small32_kernel(const float* __restrict__ X, float* __restrict__ G,
+                        int n, long d, long kpw) {
+  __shared__ float cbuf[32 * 32];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  for (int i = t; i < 32 * 32; i += SMALL_WAVES * 64) cbuf[i] = 0.0f;
+
+  const long W = (long)blockIdx.x * SMALL_WAVES + wave;
+  const long base = W * kpw;
+
+  const int frow = lane & 31;
+  const int i = min(frow, n - 1);
+  const long lane_k = lane >> 5;  // k = k0 + lane_k, K-step 2
+  const float* src = X + (long)i * d + base + lane_k;
+
+  f32x16 acc0 = {};
+  f32x16 acc1 = {};
+  if (base + kpw <= d) {
+    long k = 0;
+    for (; k + 16 <= kpw; k += 16) {
+      float a[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) a[u] = src[k + u * 2];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        if (u & 1)
+          acc1 = __builtin_amdgcn_mfma_f32_32x32x2f32(a[u], a[u], acc1, 0, 0, 0);
+        else
+          acc0 = __builtin_amdgcn_mfma_f32_32x32x2f32(a[u], a[u], acc0, 0, 0, 0);
+      }
+    }
+    for (; k + 2 <= kpw; k += 2) {
+      const float a = src[k];
+      acc0 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, a, acc0, 0, 0, 0);
+    }
+  } else if (base < d) {
+    const long hi = d - base;
+    for (long k = 0; k < kpw && k < hi; k += 2) {
+      const long kk = k + lane_k;
+      const float a = (kk < hi) ? src[k] : 0.0f;
+      acc0 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, a, acc0, 0, 0, 0);
+    }
+  }
+
+  __syncthreads();
+  // C/D map for 32x32: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int cc = lane & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rr = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    if (rr < n && cc < n) atomicAdd(&cbuf[rr * n + cc], acc0[r] + acc1[r]);
+  }
+  __syncthreads();
+  for (int idx = t; idx < n * n; idx += SMALL_WAVES * 64)
+    atomicAdd(&G[idx], cbuf[idx]);
+}
+
+// kpw chosen so each wave has >= ~2048 elements and the grid still fills
+// the chip (8 XCDs x 32 CUs want >= ~512 blocks when d allows).
+inline void small_geometry(long d, int align, long& kpw, long& blocks) {
+  long nw = (d + 2047) / 2048;
+  if (nw < 1) nw = 1;
+  if (nw > 16384) nw = 16384;
+  blocks = (nw + SMALL_WAVES - 1) / SMALL_WAVES;
+  nw = blocks * SMALL_WAVES;
+  kpw = (d + nw - 1) / nw;
+  kpw = ((kpw + align - 1) / align) * align;
+}
+
 inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
   const int tiles = (n + TILE - 1) / TILE;
   const long tile_blocks = (long)tiles * tiles;
@@ -476,6 +770,22 @@ inline void split_geometry(int n, long d, int& splitk, long& k_per_block) {
 
 void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
                       hipStream_t stream) {
+  if (n <= 32 && (d % 8) == 0 && d >= 64) {
+    long kpw, blocks;
+    if (n <= 16) {
+      const int P = 16 / n;
+      // kpw multiple of P*32 so each sub-slab is a whole number of K=32
+      // MFMA steps (the in-range fast loop has no sub-step tail)
+      small_geometry(d, P * 32, kpw, blocks);
+      hipLaunchKernelGGL(gram_bf16_small16_kernel, dim3((unsigned)blocks),
+                         dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw, P);
+    } else {
+      small_geometry(d, 16, kpw, blocks);
+      hipLaunchKernelGGL(gram_bf16_small32_kernel, dim3((unsigned)blocks),
+                         dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw);
+    }
+    return;
+  }
   int splitk; long kpb;
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;
@@ -506,6 +816,20 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
 
 void launch_gram_f32(const float* X, float* G, int n, long d,
                      hipStream_t stream) {
+  if (n <= 32 && d >= 64) {
+    long kpw, blocks;
+    if (n <= 16) {
+      const int P = 16 / n;
+      small_geometry(d, P * 4, kpw, blocks);
+      hipLaunchKernelGGL(gram_f32_small16_kernel, dim3((unsigned)blocks),
+                         dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw, P);
+    } else {
+      small_geometry(d, 16, kpw, blocks);
+      hipLaunchKernelGGL(gram_f32_small32_kernel, dim3((unsigned)blocks),
+                         dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw);
+    }
+    return;
+  }
   int splitk; long kpb;
   split_geometry(n, d, splitk, kpb);
   const int tiles = (n + TILE - 1) / TILE;

@@ -1,0 +1,430 @@
+// Generic streaming radix-select engine for large-n column order
+// statistics (SURVEY.md K1-K3 at n > ~192, where the in-LDS sort
+// re-touches every element log^2 P times and collapses to ~0.1-0.6 TB/s).
+//
+// The specialized 2-pass bf16 MEDIAN kernels live in colsel.hip
+// (rsel_pass1/2). This file generalizes the idea to every other
+// mode x dtype combination as a sequence of STREAMING passes, each
+// HBM-bound and coalesced the same way (64 adjacent columns per block,
+// 4 row slices per column):
+//
+//   level pass   : 256-bucket histogram of one 8-bit digit of a MONOTONE
+//                  u32 key, restricted to elements whose higher digits
+//                  match the per-column prefix resolved so far; a scan
+//                  advances the prefix by 8 bits for 1 or 2 target ranks.
+//                  bf16 value keys need 2 levels, f32 keys 4.
+//   sum pass     : one more read of X accumulating the value sums /
+//                  boundary-tie counts the mode's closed form needs.
+//
+// Pass counts: f32 MEDIAN 4+0, TRIMMED 2+1 (bf16) / 4+1 (f32),
+// MEAMED = median levels + 4 dev-key levels + 1 sum pass (dev = |v - med|
+// is an exact f32, so its key needs all 4 digits — a bf16-rounded dev key
+// would merge near-ties and push boundary elements across med, a
+// 2*rho/(n-f) output error, not an ulp).
+//
+// Per-column selection state in global scratch, 4 u32 per column:
+//   [0] prefix for rank 0   [1] count of keys strictly below prefix0
+//   [2] prefix for rank 1   [3] count below prefix1
+// Ranks are 1-based. Counts are exact after the last level.
+#include "common.h"
+
+namespace {
+
+typedef unsigned int u32;
+
+constexpr int RS_COLS = 64;
+
+// -- monotone key transforms ------------------------------------------------
+
+// bf16 bits -> sortable u16, placed in the TOP 16 bits of the u32 key
+DEV u32 key_from_bf16_bits(u32 bits) {
+  const u32 s = (bits >> 15) & 1u;
+  return ((bits ^ (0x8000u + s * 0x7FFFu)) & 0xFFFFu) << 16;
+}
+
+// f32 -> sortable u32 (sign-magnitude flip; NaN sorts above +inf)
+DEV u32 key_from_f32(float v) {
+  u32 bits = __float_as_uint(v);
+  return (bits >> 31) ? ~bits : (bits | 0x80000000u);
+}
+
+DEV float f32_from_key(u32 key) {
+  const u32 bits = (key & 0x80000000u) ? (key ^ 0x80000000u) : ~key;
+  return __uint_as_float(bits);
+}
+
+DEV float bf16val_from_key(u32 key) {
+  const u32 k16 = key >> 16;
+  unsigned short bits = (k16 & 0x8000u) ? (unsigned short)(k16 ^ 0x8000u)
+                                        : (unsigned short)(~k16 & 0xFFFFu);
+  union { unsigned short s; __hip_bfloat16 h; } c;
+  c.s = bits;
+  return __bfloat162float(c.h);
+}
+
+// |v - med| as a monotone key: non-negative f32 bits are already ordered;
+// NaN (v and med both inf) forced to the top so it is trimmed, not kept
+DEV u32 key_from_dev(float v, float med) {
+  const float dev = fabsf(v - med);
+  if (dev != dev) return 0xFFFFFFFFu;
+  return __float_as_uint(dev);
+}
+
+enum KeyKind { VAL_BF16 = 0, VAL_F32 = 1, DEV_KEY = 2 };
+
+template <typename T, int KK>
+DEV u32 make_key(T raw, float med) {
+  if (KK == VAL_BF16) {
+    union { T t; unsigned short s; } c;
+    c.t = raw;
+    return key_from_bf16_bits((u32)c.s);
+  } else if (KK == VAL_F32) {
+    return key_from_f32(to_f<T>(raw));
+  } else {
+    return key_from_dev(to_f<T>(raw), med);
+  }
+}
+
+template <typename T, int KK>
+DEV float val_from_key(u32 key) {
+  return (KK == VAL_BF16) ? bf16val_from_key(key) : f32_from_key(key);
+}
+
+// -- level pass -------------------------------------------------------------
+
+// TWO=true tracks two ranks with a packed 16/16 histogram (counts <= n
+// <= 65535); TWO=false tracks one rank with plain u32 counts.
+template <typename T, int KK, bool TWO>
+__global__ void __launch_bounds__(256, 2)
+rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
+                  u32* __restrict__ state, int n, long d, int shift,
+                  u32 hi_mask, u32 t0, u32 t1) {
+  extern __shared__ __attribute__((aligned(16))) u32 rs_lds[];
+  u32 (*cnt)[RS_COLS + 1] = reinterpret_cast<u32(*)[RS_COLS + 1]>(rs_lds);
+  // layout: cnt[bin][col] — adjacent threads histogram adjacent columns
+  // into adjacent banks (+1 pad de-conflicts the per-column scan)
+  const int t = threadIdx.x;
+  const int c = t & 63;
+  const int slice = t >> 6;
+  const long col0 = (long)blockIdx.x * RS_COLS;
+  const int cols = (int)min((long)RS_COLS, d - col0);
+  for (int i = t; i < 256 * (RS_COLS + 1); i += 256)
+    reinterpret_cast<u32*>(cnt)[i] = 0;
+  __syncthreads();
+  if (c < cols) {
+    const long col = col0 + c;
+    const u32 p0 = state[col * 4 + 0];
+    const u32 p1 = TWO ? state[col * 4 + 2] : 0;
+    const float m = (KK == DEV_KEY) ? med[col] : 0.0f;
+    const T* xc = X + col;
+    // 4 loads in flight per step (one outstanding request per lane is
+    // latency-bound at this stride)
+    int row = slice;
+    for (; row + 12 < n; row += 16) {
+      const T r0 = xc[(long)(row + 0) * d];
+      const T r1 = xc[(long)(row + 4) * d];
+      const T r2 = xc[(long)(row + 8) * d];
+      const T r3 = xc[(long)(row + 12) * d];
+      const T raw[4] = {r0, r1, r2, r3};
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const u32 key = make_key<T, KK>(raw[q], m);
+        const bool m0 = ((key ^ p0) & hi_mask) == 0;
+        if (TWO) {
+          const bool m1 = ((key ^ p1) & hi_mask) == 0;
+          const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][c], inc);
+        } else if (m0) {
+          atomicAdd(&cnt[(key >> shift) & 0xFFu][c], 1u);
+        }
+      }
+    }
+    for (; row < n; row += 4) {
+      const u32 key = make_key<T, KK>(xc[(long)row * d], m);
+      const bool m0 = ((key ^ p0) & hi_mask) == 0;
+      if (TWO) {
+        const bool m1 = ((key ^ p1) & hi_mask) == 0;
+        const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+        if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][c], inc);
+      } else if (m0) {
+        atomicAdd(&cnt[(key >> shift) & 0xFFu][c], 1u);
+      }
+    }
+  }
+  __syncthreads();
+  // one thread per column: advance prefix by this digit
+  if (t < cols) {
+    const long col = col0 + t;
+    u32 p0 = state[col * 4 + 0], b0 = state[col * 4 + 1];
+    u32 p1 = 0, b1 = 0;
+    if (TWO) { p1 = state[col * 4 + 2]; b1 = state[col * 4 + 3]; }
+    u32 run0 = 0, run1 = 0;
+    bool got0 = false, got1 = !TWO;
+#pragma unroll 4
+    for (int b = 0; b < 256; ++b) {
+      const u32 packed = cnt[b][t];
+      const u32 c0 = TWO ? (packed & 0xFFFFu) : packed;
+      if (!got0 && t0 <= b0 + run0 + c0) {
+        p0 |= (u32)b << shift;
+        b0 += run0;
+        got0 = true;
+      }
+      run0 += c0;
+      if (TWO) {
+        const u32 c1 = packed >> 16;
+        if (!got1 && t1 <= b1 + run1 + c1) {
+          p1 |= (u32)b << shift;
+          b1 += run1;
+          got1 = true;
+        }
+        run1 += c1;
+      }
+    }
+    state[col * 4 + 0] = p0;
+    state[col * 4 + 1] = b0;
+    if (TWO) {
+      state[col * 4 + 2] = p1;
+      state[col * 4 + 3] = b1;
+    }
+  }
+}
+
+// -- finalize kernels -------------------------------------------------------
+
+// median output (or med staging for MEAMED): 0.5 * (v(rank_lo) + v(rank_hi))
+template <typename T, int KK, bool TO_F32>
+__global__ void rsel_median_out_kernel(const u32* __restrict__ state,
+                                       void* __restrict__ out, long d) {
+  const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= d) return;
+  const float lo = val_from_key<T, KK>(state[col * 4 + 0]);
+  const float hi = val_from_key<T, KK>(state[col * 4 + 2]);
+  const float m = 0.5f * (lo + hi);
+  if (TO_F32)
+    reinterpret_cast<float*>(out)[col] = m;
+  else
+    reinterpret_cast<T*>(out)[col] = from_f<T>(m);
+}
+
+// TRIMMED closed form from boundary keys L (rank f+1) and U (rank n-f):
+//   S = sum(L < v < U) + keptL * val(L) + keptU * val(U)
+// with keptL = min(belowL + eqL, n-f) - f and keptU = (n-f) - belowU.
+template <typename T, int KK>
+__global__ void __launch_bounds__(256, 2)
+rsel_trimmed_sum_kernel(const T* __restrict__ X, const u32* __restrict__ state,
+                        T* __restrict__ out, int n, long d, int f) {
+  __shared__ float sBU[RS_COLS], sLeL[RS_COLS];
+  __shared__ u32 cEqL[RS_COLS];
+  const int t = threadIdx.x;
+  const int c = t & 63;
+  const int slice = t >> 6;
+  const long col0 = (long)blockIdx.x * RS_COLS;
+  const int cols = (int)min((long)RS_COLS, d - col0);
+  if (t < RS_COLS) { sBU[t] = 0.0f; sLeL[t] = 0.0f; cEqL[t] = 0; }
+  __syncthreads();
+  u32 keyL = 0, keyU = 0;
+  float my_sBU = 0.0f, my_sLeL = 0.0f;
+  u32 my_cEqL = 0;
+  if (c < cols) {
+    const long col = col0 + c;
+    keyL = state[col * 4 + 0];
+    keyU = state[col * 4 + 2];
+    const T* xc = X + col;
+    for (int row = slice; row < n; row += 4) {
+      const T raw = xc[(long)row * d];
+      const u32 key = make_key<T, KK>(raw, 0.0f);
+      const float v = to_f<T>(raw);
+      if (key < keyU) my_sBU += v;
+      if (key <= keyL) my_sLeL += v;
+      if (key == keyL) ++my_cEqL;
+    }
+    atomicAdd(&sBU[c], my_sBU);
+    atomicAdd(&sLeL[c], my_sLeL);
+    atomicAdd(&cEqL[c], my_cEqL);
+  }
+  __syncthreads();
+  if (t < cols) {
+    const long col = col0 + t;
+    const u32 kL = state[col * 4 + 0], kU = state[col * 4 + 2];
+    const u32 aL = state[col * 4 + 1], aU = state[col * 4 + 3];
+    const float vL = val_from_key<T, KK>(kL);
+    const u32 keep = (u32)(n - f);
+    float S;
+    if (kL == kU) {
+      S = (float)(n - 2 * f) * vL;
+    } else {
+      const float vU = val_from_key<T, KK>(kU);
+      const u32 keptL = min(aL + cEqL[t], keep) - (u32)f;
+      const u32 keptU = keep - aU;
+      S = (sBU[t] - sLeL[t]) + (float)keptL * vL + (float)keptU * vU;
+    }
+    out[col] = from_f<T>(S / (float)(n - 2 * f));
+  }
+}
+
+// MEAMED finish: rho = dev key at rank n-f (state slot 0, ONE-rank run).
+//   S = sum(v : dev < rho) + ties, taking ties below the median first
+// (matches the LDS kernel's shrink-from-the-worse-end policy, which keeps
+// the left end on equal deviation). Tied values sharing an exact dev key
+// are taken at their mean.
+template <typename T, int KK>
+__global__ void __launch_bounds__(256, 2)
+rsel_meamed_sum_kernel(const T* __restrict__ X, const float* __restrict__ med,
+                       const u32* __restrict__ state, T* __restrict__ out,
+                       int n, long d, int f) {
+  __shared__ float sKept[RS_COLS], sEqL[RS_COLS], sEqR[RS_COLS];
+  __shared__ u32 cEqL[RS_COLS], cEqR[RS_COLS];
+  const int t = threadIdx.x;
+  const int c = t & 63;
+  const int slice = t >> 6;
+  const long col0 = (long)blockIdx.x * RS_COLS;
+  const int cols = (int)min((long)RS_COLS, d - col0);
+  if (t < RS_COLS) {
+    sKept[t] = 0.0f; sEqL[t] = 0.0f; sEqR[t] = 0.0f;
+    cEqL[t] = 0; cEqR[t] = 0;
+  }
+  __syncthreads();
+  if (c < cols) {
+    const long col = col0 + c;
+    const u32 rho = state[col * 4 + 0];
+    const float m = med[col];
+    float my_sKept = 0.0f, my_sEqL = 0.0f, my_sEqR = 0.0f;
+    u32 my_cEqL = 0, my_cEqR = 0;
+    const T* xc = X + col;
+    for (int row = slice; row < n; row += 4) {
+      const T raw = xc[(long)row * d];
+      const float v = to_f<T>(raw);
+      const u32 key = key_from_dev(v, m);
+      if (key < rho) {
+        my_sKept += v;
+      } else if (key == rho) {
+        if (v < m) { my_sEqL += v; ++my_cEqL; }
+        else       { my_sEqR += v; ++my_cEqR; }
+      }
+    }
+    atomicAdd(&sKept[c], my_sKept);
+    atomicAdd(&sEqL[c], my_sEqL);
+    atomicAdd(&sEqR[c], my_sEqR);
+    atomicAdd(&cEqL[c], my_cEqL);
+    atomicAdd(&cEqR[c], my_cEqR);
+  }
+  __syncthreads();
+  if (t < cols) {
+    const long col = col0 + t;
+    const u32 below = state[col * 4 + 1];
+    const u32 keep = (u32)(n - f);
+    const u32 ties = keep - below;
+    const u32 takeL = min(ties, cEqL[t]);
+    const u32 takeR = ties - takeL;
+    float S = sKept[t];
+    if (takeL && cEqL[t]) S += (sEqL[t] / (float)cEqL[t]) * (float)takeL;
+    if (takeR && cEqR[t]) S += (sEqR[t] / (float)cEqR[t]) * (float)takeR;
+    out[col] = from_f<T>(S / (float)keep);
+  }
+}
+
+constexpr size_t RS_LDS = (size_t)256 * (RS_COLS + 1) * sizeof(u32);
+
+template <typename T, int KK, bool TWO>
+void run_levels(const T* X, const float* med, u32* state, int n, long d,
+                u32 t0, u32 t1, hipStream_t stream) {
+  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  const int top_shift = 24;
+  const int last_shift = (KK == VAL_BF16) ? 16 : 0;
+  for (int shift = top_shift; shift >= last_shift; shift -= 8) {
+    // digits above `shift` must match the resolved prefix; at the top
+    // level the mask is 0 (everything matches)
+    const u32 hi_mask =
+        (shift + 8 >= 32) ? 0u : (0xFFFFFFFFu << (shift + 8));
+    hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
+                       dim3(256), RS_LDS, stream, X, med, state, n, d, shift,
+                       hi_mask, t0, t1);
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// host-side drivers (called from bind.cpp; all launches enqueue on
+// `stream`, no host syncs). `state` is caller-zeroed d*4 u32 scratch;
+// `med` is a d-float scratch for MEAMED.
+// ---------------------------------------------------------------------------
+
+void launch_rsel_trimmed_bf16(const __hip_bfloat16* X, __hip_bfloat16* out,
+                              unsigned int* state, int n, long d, int f,
+                              hipStream_t stream) {
+  run_levels<__hip_bfloat16, VAL_BF16, true>(X, nullptr, state, n, d,
+                                             (u32)(f + 1), (u32)(n - f),
+                                             stream);
+  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  hipLaunchKernelGGL((rsel_trimmed_sum_kernel<__hip_bfloat16, VAL_BF16>),
+                     dim3((unsigned)grid), dim3(256), 0, stream, X, state, out,
+                     n, d, f);
+}
+
+void launch_rsel_trimmed_f32(const float* X, float* out, unsigned int* state,
+                             int n, long d, int f, hipStream_t stream) {
+  run_levels<float, VAL_F32, true>(X, nullptr, state, n, d, (u32)(f + 1),
+                                   (u32)(n - f), stream);
+  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  hipLaunchKernelGGL((rsel_trimmed_sum_kernel<float, VAL_F32>),
+                     dim3((unsigned)grid), dim3(256), 0, stream, X, state, out,
+                     n, d, f);
+}
+
+void launch_rsel_median_f32(const float* X, float* out, unsigned int* state,
+                            int n, long d, hipStream_t stream) {
+  const u32 t_lo = (u32)((n - 1) >> 1) + 1, t_hi = (u32)(n >> 1) + 1;
+  run_levels<float, VAL_F32, true>(X, nullptr, state, n, d, t_lo, t_hi,
+                                   stream);
+  const long elems = d;
+  const long grid = (elems + 255) / 256;
+  hipLaunchKernelGGL((rsel_median_out_kernel<float, VAL_F32, false>),
+                     dim3((unsigned)grid), dim3(256), 0, stream, state, out,
+                     d);
+}
+
+// MEAMED: median levels -> med buffer -> dev-key rank (n-f) -> sum pass.
+// Caller re-zeroes `state` between the two selections? No — the dev
+// selection reuses slot 0/1 only, and must start from prefix 0: the
+// med_out kernel is followed by a device-side state reset here.
+namespace {
+__global__ void rsel_state_reset_kernel(u32* __restrict__ state, long d) {
+  const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (col < d) {
+    state[col * 4 + 0] = 0;
+    state[col * 4 + 1] = 0;
+  }
+}
+}  // namespace
+
+template <typename T, int KK>
+static void rsel_meamed_impl(const T* X, T* out, unsigned int* state,
+                             float* med, int n, long d, int f,
+                             hipStream_t stream) {
+  const u32 t_lo = (u32)((n - 1) >> 1) + 1, t_hi = (u32)(n >> 1) + 1;
+  run_levels<T, KK, true>(X, nullptr, state, n, d, t_lo, t_hi, stream);
+  const long egrid = (d + 255) / 256;
+  hipLaunchKernelGGL((rsel_median_out_kernel<T, KK, true>),
+                     dim3((unsigned)egrid), dim3(256), 0, stream, state, med,
+                     d);
+  hipLaunchKernelGGL(rsel_state_reset_kernel, dim3((unsigned)egrid), dim3(256),
+                     0, stream, state, d);
+  run_levels<T, DEV_KEY, false>(X, med, state, n, d, (u32)(n - f), 0, stream);
+  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  hipLaunchKernelGGL((rsel_meamed_sum_kernel<T, KK>), dim3((unsigned)grid),
+                     dim3(256), 0, stream, X, med, state, out, n, d, f);
+}
+
+void launch_rsel_meamed_bf16(const __hip_bfloat16* X, __hip_bfloat16* out,
+                             unsigned int* state, float* med, int n, long d,
+                             int f, hipStream_t stream) {
+  rsel_meamed_impl<__hip_bfloat16, VAL_BF16>(X, out, state, med, n, d, f,
+                                             stream);
+}
+
+void launch_rsel_meamed_f32(const float* X, float* out, unsigned int* state,
+                            float* med, int n, long d, int f,
+                            hipStream_t stream) {
+  rsel_meamed_impl<float, VAL_F32>(X, out, state, med, n, d, f, stream);
+}

@@ -20,9 +20,10 @@ _COLSEL_MEDIAN = 0
 _COLSEL_TRIMMED = 1
 _COLSEL_MEAMED = 2
 
-# Largest n the column-sort kernels support: registers to n=64, LDS-staged
-# to n=512. Larger n falls back to a documented torch path.
-COLSEL_MAX_N = 512
+# Column order-statistic caps: registers to n=64, LDS-staged sort to
+# n=512, streaming radix-select (every mode x dtype) to n=65535
+# (colsel.hip + rsel.hip). Larger n falls back to torch.
+COLSEL_MAX_N = 65535
 
 
 def _gpu(X: torch.Tensor) -> bool:
@@ -33,12 +34,7 @@ def _gpu(X: torch.Tensor) -> bool:
 
 
 def median(X: torch.Tensor) -> torch.Tensor:
-    n = X.shape[0]
-    if _gpu(X) and (
-        n <= COLSEL_MAX_N or (X.dtype == torch.bfloat16 and n <= 65535)
-    ):
-        # n > 64 bf16 routes to the 2-pass streaming radix select inside
-        # the extension (no upper cap below 65536 rows)
+    if _gpu(X) and X.shape[0] <= COLSEL_MAX_N:
         return _hip.require().colsel(X.contiguous(), _COLSEL_MEDIAN, 0)
     return F.median(X)
 
