@@ -42,6 +42,7 @@ def main():
             for op, fn in [
                 ("median", lambda X=X: D.median(X)),
                 ("trimmed_mean", lambda X=X, f=f: D.trimmed_mean(X, f)),
+                ("meamed", lambda X=X, f=f: D.mean_of_medians(X, f)),
                 ("gram", lambda X=X: ext.gram(X.contiguous())),
             ]:
                 ms = timeit(fn) * 1e3
