@@ -395,7 +395,9 @@ constexpr int RSEL_COLS = 64;
 // marker layout: (bucket << 16) | count_below
 DEV u32 rsel_mark(u32 bucket, u32 below) { return (bucket << 16) | below; }
 
-__global__ void __launch_bounds__(256, 2)
+constexpr int RSEL_THREADS = 512;  // 8 row slices x 64 columns
+
+__global__ void __launch_bounds__(RSEL_THREADS, 2)
 rsel_pass1_kernel(const unsigned short* __restrict__ X, u32* __restrict__ mark,
                   int n, long d) {
   // dynamic LDS: 64 x 257 u32 (65.8 KB) exceeds the 64 KB static limit
@@ -403,32 +405,29 @@ rsel_pass1_kernel(const unsigned short* __restrict__ X, u32* __restrict__ mark,
   u32 (*cnt)[257] = reinterpret_cast<u32(*)[257]>(rsel_lds);
   const int t = threadIdx.x;
   const int c = t & 63;
-  const int slice = t >> 6;  // 4 row slices
+  const int slice = t >> 6;  // 8 row slices
   const long col0 = (long)blockIdx.x * RSEL_COLS;
   const int cols = (int)min((long)RSEL_COLS, d - col0);
-  for (int i = t; i < RSEL_COLS * 257; i += 256)
+  for (int i = t; i < RSEL_COLS * 257; i += RSEL_THREADS)
     reinterpret_cast<u32*>(cnt)[i] = 0;
   __syncthreads();
   if (c < cols) {
     const unsigned short* xc = X + col0 + c;
     int row = slice;
-    // 4 independent loads in flight per step: a single-load loop leaves
-    // one outstanding request per lane and goes latency-bound
-    for (; row + 12 < n; row += 16) {
-      const u32 b0 = (u32)xc[(long)(row + 0) * d];
-      const u32 b1 = (u32)xc[(long)(row + 4) * d];
-      const u32 b2 = (u32)xc[(long)(row + 8) * d];
-      const u32 b3 = (u32)xc[(long)(row + 12) * d];
-      atomicAdd(&cnt[c][((b0 ^ (0x8000u + ((b0 >> 15) & 1u) * 0x7FFFu)) &
-                         0xFFFFu) >> 8], 1u);
-      atomicAdd(&cnt[c][((b1 ^ (0x8000u + ((b1 >> 15) & 1u) * 0x7FFFu)) &
-                         0xFFFFu) >> 8], 1u);
-      atomicAdd(&cnt[c][((b2 ^ (0x8000u + ((b2 >> 15) & 1u) * 0x7FFFu)) &
-                         0xFFFFu) >> 8], 1u);
-      atomicAdd(&cnt[c][((b3 ^ (0x8000u + ((b3 >> 15) & 1u) * 0x7FFFu)) &
-                         0xFFFFu) >> 8], 1u);
+    // 8 independent loads in flight per step: fewer leaves the kernel
+    // latency-bound at this 2 B/lane stride (a 256-thread 4-deep version
+    // measured ~720 GB/s per pass; streaming peak wants ~12 KB in flight
+    // per CU)
+    for (; row + 56 < n; row += 64) {
+      u32 b[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) b[j] = (u32)xc[(long)(row + 8 * j) * d];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        atomicAdd(&cnt[c][((b[j] ^ (0x8000u + ((b[j] >> 15) & 1u) * 0x7FFFu)) &
+                           0xFFFFu) >> 8], 1u);
     }
-    for (; row < n; row += 4) {
+    for (; row < n; row += 8) {
       const u32 bits = (u32)xc[(long)row * d];
       const u32 s = (bits >> 15) & 1u;
       const u32 key = (bits ^ (0x8000u + s * 0x7FFFu)) & 0xFFFFu;
@@ -454,7 +453,7 @@ rsel_pass1_kernel(const unsigned short* __restrict__ X, u32* __restrict__ mark,
   }
 }
 
-__global__ void __launch_bounds__(256, 2)
+__global__ void __launch_bounds__(RSEL_THREADS, 2)
 rsel_pass2_kernel(const unsigned short* __restrict__ X,
                   const u32* __restrict__ mark,
                   unsigned short* __restrict__ out, int n, long d) {
@@ -467,7 +466,7 @@ rsel_pass2_kernel(const unsigned short* __restrict__ X,
   const int slice = t >> 6;
   const long col0 = (long)blockIdx.x * RSEL_COLS;
   const int cols = (int)min((long)RSEL_COLS, d - col0);
-  for (int i = t; i < RSEL_COLS * 257; i += 256)
+  for (int i = t; i < RSEL_COLS * 257; i += RSEL_THREADS)
     reinterpret_cast<u32*>(cnt)[i] = 0;
   __syncthreads();
   if (c < cols) {
@@ -475,13 +474,12 @@ rsel_pass2_kernel(const unsigned short* __restrict__ X,
     const u32 b_hi = mark[(col0 + c) * 2 + 1] >> 16;
     const unsigned short* xc = X + col0 + c;
     int row = slice;
-    for (; row + 12 < n; row += 16) {
-      const u32 raw[4] = {(u32)xc[(long)(row + 0) * d],
-                          (u32)xc[(long)(row + 4) * d],
-                          (u32)xc[(long)(row + 8) * d],
-                          (u32)xc[(long)(row + 12) * d]};
+    for (; row + 56 < n; row += 64) {
+      u32 raw[8];
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
+      for (int j = 0; j < 8; ++j) raw[j] = (u32)xc[(long)(row + 8 * j) * d];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
         const u32 key =
             (raw[q] ^ (0x8000u + ((raw[q] >> 15) & 1u) * 0x7FFFu)) & 0xFFFFu;
         const u32 hi8 = key >> 8;
@@ -490,7 +488,7 @@ rsel_pass2_kernel(const unsigned short* __restrict__ X,
         if (inc) atomicAdd(&cnt[c][key & 0xFFu], inc);
       }
     }
-    for (; row < n; row += 4) {
+    for (; row < n; row += 8) {
       const u32 bits = (u32)xc[(long)row * d];
       const u32 s = (bits >> 15) & 1u;
       const u32 key = (bits ^ (0x8000u + s * 0x7FFFu)) & 0xFFFFu;
@@ -663,7 +661,7 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
 void launch_colsel_median_radix_bf16(const __hip_bfloat16* X,
                                      __hip_bfloat16* out, unsigned int* mark,
                                      int n, long d, hipStream_t stream) {
-  const int block = 256;
+  const int block = 512;  // must match RSEL_THREADS
   const long grid = (d + RSEL_COLS - 1) / RSEL_COLS;
   const size_t lds = (size_t)RSEL_COLS * 257 * sizeof(u32);
   const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);

@@ -94,8 +94,10 @@ DEV float val_from_key(u32 key) {
 
 // TWO=true tracks two ranks with a packed 16/16 histogram (counts <= n
 // <= 65535); TWO=false tracks one rank with plain u32 counts.
+constexpr int RS_THREADS = 512;  // 8 row slices x 64 columns
+
 template <typename T, int KK, bool TWO>
-__global__ void __launch_bounds__(256, 2)
+__global__ void __launch_bounds__(RS_THREADS, 2)
 rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
                   u32* __restrict__ state, int n, long d, int shift,
                   u32 hi_mask, u32 t0, u32 t1) {
@@ -108,7 +110,7 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
   const int slice = t >> 6;
   const long col0 = (long)blockIdx.x * RS_COLS;
   const int cols = (int)min((long)RS_COLS, d - col0);
-  for (int i = t; i < 256 * (RS_COLS + 1); i += 256)
+  for (int i = t; i < 256 * (RS_COLS + 1); i += RS_THREADS)
     reinterpret_cast<u32*>(cnt)[i] = 0;
   __syncthreads();
   if (c < cols) {
@@ -117,17 +119,15 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     const u32 p1 = TWO ? state[col * 4 + 2] : 0;
     const float m = (KK == DEV_KEY) ? med[col] : 0.0f;
     const T* xc = X + col;
-    // 4 loads in flight per step (one outstanding request per lane is
-    // latency-bound at this stride)
+    // 8 loads in flight per step x 8 slices: fewer is latency-bound at
+    // this strided-2B/4B-per-lane walk
     int row = slice;
-    for (; row + 12 < n; row += 16) {
-      const T r0 = xc[(long)(row + 0) * d];
-      const T r1 = xc[(long)(row + 4) * d];
-      const T r2 = xc[(long)(row + 8) * d];
-      const T r3 = xc[(long)(row + 12) * d];
-      const T raw[4] = {r0, r1, r2, r3};
+    for (; row + 56 < n; row += 64) {
+      T raw[8];
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
+      for (int j = 0; j < 8; ++j) raw[j] = xc[(long)(row + 8 * j) * d];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
         const u32 key = make_key<T, KK>(raw[q], m);
         const bool m0 = ((key ^ p0) & hi_mask) == 0;
         if (TWO) {
@@ -139,7 +139,7 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
         }
       }
     }
-    for (; row < n; row += 4) {
+    for (; row < n; row += 8) {
       const u32 key = make_key<T, KK>(xc[(long)row * d], m);
       const bool m0 = ((key ^ p0) & hi_mask) == 0;
       if (TWO) {
@@ -210,36 +210,47 @@ __global__ void rsel_median_out_kernel(const u32* __restrict__ state,
 //   S = sum(L < v < U) + keptL * val(L) + keptU * val(U)
 // with keptL = min(belowL + eqL, n-f) - f and keptU = (n-f) - belowU.
 template <typename T, int KK>
-__global__ void __launch_bounds__(256, 2)
+__global__ void __launch_bounds__(RS_THREADS, 2)
 rsel_trimmed_sum_kernel(const T* __restrict__ X, const u32* __restrict__ state,
                         T* __restrict__ out, int n, long d, int f) {
-  __shared__ float sBU[RS_COLS], sLeL[RS_COLS];
+  __shared__ float sMid[RS_COLS];
   __shared__ u32 cEqL[RS_COLS];
   const int t = threadIdx.x;
   const int c = t & 63;
   const int slice = t >> 6;
   const long col0 = (long)blockIdx.x * RS_COLS;
   const int cols = (int)min((long)RS_COLS, d - col0);
-  if (t < RS_COLS) { sBU[t] = 0.0f; sLeL[t] = 0.0f; cEqL[t] = 0; }
+  if (t < RS_COLS) { sMid[t] = 0.0f; cEqL[t] = 0; }
   __syncthreads();
-  u32 keyL = 0, keyU = 0;
-  float my_sBU = 0.0f, my_sLeL = 0.0f;
-  u32 my_cEqL = 0;
   if (c < cols) {
     const long col = col0 + c;
-    keyL = state[col * 4 + 0];
-    keyU = state[col * 4 + 2];
+    const u32 keyL = state[col * 4 + 0];
+    const u32 keyU = state[col * 4 + 2];
     const T* xc = X + col;
-    for (int row = slice; row < n; row += 4) {
+    // middle sum accumulated DIRECTLY (strictly between the boundary
+    // keys): a prefix-difference formulation turns -inf rows into
+    // (-inf) - (-inf) = NaN
+    float my_sMid = 0.0f;
+    u32 my_cEqL = 0;
+    int row = slice;
+    for (; row + 24 < n; row += 32) {
+      T raw[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) raw[j] = xc[(long)(row + 8 * j) * d];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const u32 key = make_key<T, KK>(raw[q], 0.0f);
+        if (key > keyL && key < keyU) my_sMid += to_f<T>(raw[q]);
+        if (key == keyL) ++my_cEqL;
+      }
+    }
+    for (; row < n; row += 8) {
       const T raw = xc[(long)row * d];
       const u32 key = make_key<T, KK>(raw, 0.0f);
-      const float v = to_f<T>(raw);
-      if (key < keyU) my_sBU += v;
-      if (key <= keyL) my_sLeL += v;
+      if (key > keyL && key < keyU) my_sMid += to_f<T>(raw);
       if (key == keyL) ++my_cEqL;
     }
-    atomicAdd(&sBU[c], my_sBU);
-    atomicAdd(&sLeL[c], my_sLeL);
+    atomicAdd(&sMid[c], my_sMid);
     atomicAdd(&cEqL[c], my_cEqL);
   }
   __syncthreads();
@@ -254,9 +265,9 @@ rsel_trimmed_sum_kernel(const T* __restrict__ X, const u32* __restrict__ state,
       S = (float)(n - 2 * f) * vL;
     } else {
       const float vU = val_from_key<T, KK>(kU);
-      const u32 keptL = min(aL + cEqL[t], keep) - (u32)f;
-      const u32 keptU = keep - aU;
-      S = (sBU[t] - sLeL[t]) + (float)keptL * vL + (float)keptU * vU;
+      const u32 keptL = min(aL + cEqL[t], keep) - (u32)f;  // provably >= 1
+      const u32 keptU = keep - aU;                          // provably >= 1
+      S = sMid[t] + (float)keptL * vL + (float)keptU * vU;
     }
     out[col] = from_f<T>(S / (float)(n - 2 * f));
   }
@@ -268,7 +279,7 @@ rsel_trimmed_sum_kernel(const T* __restrict__ X, const u32* __restrict__ state,
 // the left end on equal deviation). Tied values sharing an exact dev key
 // are taken at their mean.
 template <typename T, int KK>
-__global__ void __launch_bounds__(256, 2)
+__global__ void __launch_bounds__(RS_THREADS, 2)
 rsel_meamed_sum_kernel(const T* __restrict__ X, const float* __restrict__ med,
                        const u32* __restrict__ state, T* __restrict__ out,
                        int n, long d, int f) {
@@ -291,7 +302,24 @@ rsel_meamed_sum_kernel(const T* __restrict__ X, const float* __restrict__ med,
     float my_sKept = 0.0f, my_sEqL = 0.0f, my_sEqR = 0.0f;
     u32 my_cEqL = 0, my_cEqR = 0;
     const T* xc = X + col;
-    for (int row = slice; row < n; row += 4) {
+    int row = slice;
+    for (; row + 24 < n; row += 32) {
+      T raw[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) raw[j] = xc[(long)(row + 8 * j) * d];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const float v = to_f<T>(raw[q]);
+        const u32 key = key_from_dev(v, m);
+        if (key < rho) {
+          my_sKept += v;
+        } else if (key == rho) {
+          if (v < m) { my_sEqL += v; ++my_cEqL; }
+          else       { my_sEqR += v; ++my_cEqR; }
+        }
+      }
+    }
+    for (; row < n; row += 8) {
       const T raw = xc[(long)row * d];
       const float v = to_f<T>(raw);
       const u32 key = key_from_dev(v, m);
@@ -337,8 +365,8 @@ void run_levels(const T* X, const float* med, u32* state, int n, long d,
     const u32 hi_mask =
         (shift + 8 >= 32) ? 0u : (0xFFFFFFFFu << (shift + 8));
     hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
-                       dim3(256), RS_LDS, stream, X, med, state, n, d, shift,
-                       hi_mask, t0, t1);
+                       dim3(RS_THREADS), RS_LDS, stream, X, med, state, n, d,
+                       shift, hi_mask, t0, t1);
   }
 }
 
@@ -358,8 +386,8 @@ void launch_rsel_trimmed_bf16(const __hip_bfloat16* X, __hip_bfloat16* out,
                                              stream);
   const long grid = (d + RS_COLS - 1) / RS_COLS;
   hipLaunchKernelGGL((rsel_trimmed_sum_kernel<__hip_bfloat16, VAL_BF16>),
-                     dim3((unsigned)grid), dim3(256), 0, stream, X, state, out,
-                     n, d, f);
+                     dim3((unsigned)grid), dim3(RS_THREADS), 0, stream, X,
+                     state, out, n, d, f);
 }
 
 void launch_rsel_trimmed_f32(const float* X, float* out, unsigned int* state,
@@ -368,8 +396,8 @@ void launch_rsel_trimmed_f32(const float* X, float* out, unsigned int* state,
                                    (u32)(n - f), stream);
   const long grid = (d + RS_COLS - 1) / RS_COLS;
   hipLaunchKernelGGL((rsel_trimmed_sum_kernel<float, VAL_F32>),
-                     dim3((unsigned)grid), dim3(256), 0, stream, X, state, out,
-                     n, d, f);
+                     dim3((unsigned)grid), dim3(RS_THREADS), 0, stream, X,
+                     state, out, n, d, f);
 }
 
 void launch_rsel_median_f32(const float* X, float* out, unsigned int* state,
@@ -413,7 +441,7 @@ static void rsel_meamed_impl(const T* X, T* out, unsigned int* state,
   run_levels<T, DEV_KEY, false>(X, med, state, n, d, (u32)(n - f), 0, stream);
   const long grid = (d + RS_COLS - 1) / RS_COLS;
   hipLaunchKernelGGL((rsel_meamed_sum_kernel<T, KK>), dim3((unsigned)grid),
-                     dim3(256), 0, stream, X, med, state, out, n, d, f);
+                     dim3(RS_THREADS), 0, stream, X, med, state, out, n, d, f);
 }
 
 void launch_rsel_meamed_bf16(const __hip_bfloat16* X, __hip_bfloat16* out,
