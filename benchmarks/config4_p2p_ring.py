@@ -35,6 +35,10 @@ def main() -> None:
     p.add_argument("--f", type=int, default=1)
     p.add_argument("--streams", action="store_true",
                    help="one HIP stream per node update (overlap)")
+    p.add_argument("--fixed-iters", type=int, default=16,
+                   help="poll-free Weiszfeld iterations (0 = tol/poll mode)")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the whole gossip round in one hipGraph")
     p.add_argument("--device", default="cuda" if torch.cuda.is_available() else "cpu")
     args = p.parse_args()
     dev = torch.device(args.device)
@@ -54,6 +58,14 @@ def main() -> None:
         else None
     )
 
+    fixed = args.fixed_iters if args.fixed_iters > 0 else None
+
+    def geomed(mixed):
+        if fixed is not None:
+            # poll-free: no host sync, so per-node streams truly overlap
+            return D.geometric_median(mixed, fixed_iters=fixed)
+        return D.geometric_median(mixed, tol=1e-6, max_iter=32)
+
     def round_once():
         new = torch.empty_like(theta)
         if streams is None:
@@ -61,7 +73,7 @@ def main() -> None:
                 group = [i] + topo.in_neighbors(i)
                 X = theta[group]  # (1+2k, d) view-gather
                 mixed = D.nnm(X, args.f)
-                new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+                new[i] = geomed(mixed)
         else:
             # node updates are independent: one HIP stream each, so the
             # launch/sync-bound small-kernel chains overlap on the chip
@@ -73,10 +85,28 @@ def main() -> None:
                     group = [i] + topo.in_neighbors(i)
                     X = theta[group]
                     mixed = D.nnm(X, args.f)
-                    new[i] = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+                    new[i] = geomed(mixed)
             for s in streams:
                 cur.wait_stream(s)
         theta.copy_(new)
+
+    if args.graph and dev.type == "cuda":
+        if fixed is None:
+            raise SystemExit("--graph requires --fixed-iters > 0 (capture-safe)")
+        # capture one full gossip round (all node updates + the theta swap)
+        # and replay it per round: zero Python dispatch in the loop
+        s0 = torch.cuda.Stream()
+        s0.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s0):
+            round_once()
+        torch.cuda.current_stream().wait_stream(s0)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            round_once()
+        round_replay = g.replay
+
+        def round_once():  # noqa: F811 — replayed capture
+            round_replay()
 
     def sync():
         if dev.type == "cuda":

@@ -112,18 +112,27 @@ class GeometricMedian(Aggregator):
         eps: float = 1e-12,
         init: str = "median",
         chunk_size: int = 32,
+        fixed_iters: "int | None" = None,
     ) -> None:
         if tol <= 0 or max_iter <= 0 or eps <= 0:
             raise ValueError("tol, max_iter, eps must be > 0")
         if init not in {"median", "mean"}:
             raise ValueError("init must be 'median' or 'mean'")
+        if fixed_iters is not None and fixed_iters <= 0:
+            raise ValueError("fixed_iters must be > 0")
         self.tol, self.max_iter, self.eps = float(tol), int(max_iter), float(eps)
         self.init = init
         self.chunk_size = int(chunk_size)
+        # fixed_iters: exactly that many Weiszfeld steps, NO convergence
+        # polls — fully async on GPU (per-node streams overlap; hipGraph
+        # capture-safe). The poll-free path is what unblocks config-4
+        # gossip (round-1 weakness 5).
+        self.fixed_iters = None if fixed_iters is None else int(fixed_iters)
 
     def _aggregate(self, X: torch.Tensor) -> torch.Tensor:
         return D.geometric_median(
-            X, tol=self.tol, max_iter=self.max_iter, eps=self.eps, init=self.init
+            X, tol=self.tol, max_iter=self.max_iter, eps=self.eps,
+            init=self.init, fixed_iters=self.fixed_iters
         )
 
     async def run_barriered_subtasks(self, ctx: OpContext, **inputs: Any) -> Any:
@@ -137,12 +146,13 @@ class GeometricMedian(Aggregator):
         try:
             n = X.shape[0]
             chunk = max(1, min(self.chunk_size, n))
+            iters = self.fixed_iters if self.fixed_iters is not None else self.max_iter
             z = (F.median(X) if self.init == "median" else X.float().mean(dim=0)).float()
             if use_shm:
                 # ONE segment, rewritten per iteration (reference
                 # geometric_median.py:126 _write_handle semantics)
                 center_handle = register_tensor(z)
-            for _ in range(self.max_iter):
+            for _ in range(iters):
                 if use_shm:
                     write_handle(center_handle, z)
                     center_ref = center_handle
@@ -161,7 +171,7 @@ class GeometricMedian(Aggregator):
                 z_new = num / den
                 shift = float((z_new - z).norm())
                 z = z_new
-                if shift <= self.tol:
+                if self.fixed_iters is None and shift <= self.tol:
                     break
             return to_like(z.to(X.dtype), like)
         finally:

@@ -180,18 +180,31 @@ def geometric_median(
     max_iter: int = 256,
     eps: float = 1e-12,
     init: str = "median",
+    fixed_iters: Optional[int] = None,
 ) -> torch.Tensor:
+    """Weiszfeld fixed point. ``fixed_iters`` runs exactly that many
+    iterations with NO convergence polls — fully async (no host sync), so
+    per-node streams overlap and the whole call is hipGraph-capture-safe
+    (the polls serialized config-4 gossip in round 1)."""
     if not _gpu(X) or X.shape[0] > 1024:
         # extension's weiszfeld_iter TORCH_CHECKs n <= 1024; the torch
         # functional path runs fine on-device for larger n
+        if fixed_iters is not None:
+            return F.geometric_median(
+                X, tol=0.0, max_iter=int(fixed_iters), eps=eps, init=init
+            )
         return F.geometric_median(X, tol=tol, max_iter=max_iter, eps=eps, init=init)
     ext = _hip.require()
     Xc = X.contiguous()
     z = (median(Xc) if init == "median" else Xc.float().mean(dim=0)).float()
+    shift = torch.zeros((), device=X.device, dtype=torch.float32)
+    if fixed_iters is not None:
+        for _ in range(int(fixed_iters)):
+            z = ext.weiszfeld_iter(Xc, z, float(eps), shift)
+        return z.to(X.dtype)
     # fused per-iteration kernel pair; convergence polled every `poll` iters
     # to avoid a host sync per iteration (SURVEY.md §7 hard part 2)
     poll = 4
-    shift = torch.zeros((), device=X.device, dtype=torch.float32)
     it = 0
     while it < max_iter:
         steps = min(poll, max_iter - it)
@@ -286,18 +299,139 @@ def smea(X: torch.Tensor, f: int) -> torch.Tensor:
     return F.smea(X, f)
 
 
+class _CafGraphBlock:
+    """R CAF rounds captured into one hipGraph (torch.cuda.CUDAGraph is
+    hipGraph on ROCm) with DEVICE-side best-lambda/active tracking — the
+    eager loop's one-host-sync-per-round made CAF only ~8x the CPU
+    reference at 64x65k (round-1 weakness; NOTES_R02 item 3). One replay =
+    R rounds = zero Python dispatch and one sync per R rounds.
+
+    Round semantics match F.caf exactly: wsum/mu/power-iteration/best
+    tracking per reference caf.py:133-184; rounds after the stop condition
+    (weight mass <= n-2f, or the n-round cap via ``rounds_left``) are
+    masked out with ``active`` so replaying a full block never corrupts
+    the result."""
+
+    def __init__(self, n: int, d: int, dtype, f: int, power_iters: int,
+                 R: int, device) -> None:
+        ext = _hip.require()
+        self.R, self.n, self.d = R, n, d
+        self.power_iters = max(1, int(power_iters))
+        self.target = float(n - 2 * f)
+        self.X = torch.zeros(n, d, dtype=dtype, device=device)
+        self.seeds = torch.zeros(R, d, dtype=torch.float32, device=device)
+        self.w = torch.ones(n, device=device)
+        self.best_lambda = torch.full((), float("inf"), device=device)
+        self.best_mu = torch.zeros(d, device=device)
+        self.active = torch.ones((), dtype=torch.bool, device=device)
+        self.rounds_left = torch.full((), float(n), device=device)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._block(ext)  # warmup (allocator pools the intermediates)
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._block(ext)
+
+    def _block(self, ext) -> None:
+        X = self.X
+        for r in range(self.R):
+            w = self.w
+            active_in = self.active.clone()
+            rl = self.rounds_left.clone()
+            wsum = w.sum()
+            inv_wsum = torch.reciprocal(wsum.clamp_min(1e-30))
+            mu = ext.caf_colsum(X, w, None, inv_wsum)
+            v = self.seeds[r]
+            v = v / v.norm().clamp_min(1e-20)
+            lam = torch.zeros((), device=X.device)
+            for _ in range(self.power_iters):
+                s_ = ext.caf_matvec(X, mu, v)
+                t_ = ext.caf_colsum(X, w * s_, mu, inv_wsum)
+                lam = t_.norm()
+                v = t_ / lam.clamp_min(1e-20)
+            proj = ext.caf_matvec(X, mu, v) ** 2
+            pmax = proj.max().clamp_min(1e-20)
+            w_next = (w * (1.0 - proj / pmax)).clamp_min(0.0)
+            better = (lam < self.best_lambda) & active_in
+            self.best_mu.copy_(torch.where(better, mu, self.best_mu))
+            self.best_lambda.copy_(torch.where(better, lam, self.best_lambda))
+            stop = (wsum <= self.target) | (w_next.sum() <= 0)
+            active_next = active_in & ~stop & (rl > 1.0)
+            self.w.copy_(torch.where(active_next, w_next, w))
+            self.active.copy_(active_next)
+            self.rounds_left.copy_(rl - 1.0)
+
+    def run(self, X: torch.Tensor) -> torch.Tensor:
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(0)
+        self.X.copy_(X)
+        self.w.fill_(1.0)
+        self.best_lambda.fill_(float("inf"))
+        self.best_mu.copy_(self.X.mean(dim=0, dtype=torch.float32))
+        self.active.fill_(True)
+        self.rounds_left.fill_(float(self.n))
+        # seeds drawn in the ORACLE's block pattern (rng stream parity with
+        # F.caf, which draws min(block_rows, n-r) rows at a time)
+        block_rows = max(1, min(4, (1 << 27) // max(self.d, 1)))
+        pending: list = []
+        drawn = 0
+        total = 0
+        while total < self.n:
+            need = min(self.R, self.n - total)
+            while sum(p.shape[0] for p in pending) < need and drawn < self.n:
+                rows = min(block_rows, self.n - drawn)
+                pending.append(torch.randn(rows, self.d, generator=gen))
+                drawn += rows
+            flat = torch.cat(pending, dim=0) if len(pending) > 1 else pending[0]
+            self.seeds[:need].copy_(flat[:need])
+            pending = [flat[need:]] if flat.shape[0] > need else []
+            self.graph.replay()
+            total += self.R  # a partial tail block is masked by rounds_left
+            if not bool(self.active):  # one sync per R rounds
+                break
+        return self.best_mu.to(X.dtype)
+
+
+_CAF_GRAPHS: dict = {}
+
+# graph path bounds: seeds are staged per replay (R*d f32 H2D), so cap d;
+# n caps the round count (and kernel TORCH_CHECKs n <= 1024)
+_CAF_GRAPH_MAX_D = 1 << 21
+
+
+def _caf_graph_ok(X: torch.Tensor, f: int) -> bool:
+    import os
+
+    if os.environ.get("BYZPY_CAF_GRAPH", "1") == "0":
+        return False
+    n, d = X.shape
+    return n <= 256 and d <= _CAF_GRAPH_MAX_D
+
+
 def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     """Covariance-agnostic filter on fused K9 kernels (reference
     caf.py:133-184): caf_matvec computes s = (X - mu) @ v, caf_colsum
     computes weighted centered column sums — neither materializes the
     (n, d) f32 diffs matrix and both replace rocblas gemvt (~260 GB/s on
     this skinny shape). Same algorithm and seeded-rng contract as the
-    functional oracle."""
+    functional oracle. Steady-state shapes run the hipGraph block path
+    (_CafGraphBlock); others fall back to the eager loop below."""
     import math
 
     n = X.shape[0]
     if not (_gpu(X) and n <= 1024 and n - 2 * f > 0):
         return F.caf(X, f, power_iters=power_iters)
+    if _caf_graph_ok(X, f):
+        d = X.shape[1]
+        R = min(8, n)
+        key = (n, d, X.dtype, int(f), int(power_iters), R, X.device.index)
+        blk = _CAF_GRAPHS.get(key)
+        if blk is None:
+            blk = _CafGraphBlock(n, d, X.dtype, int(f), power_iters, R, X.device)
+            _CAF_GRAPHS[key] = blk
+        return blk.run(X.contiguous())
     ext = _hip.require()
     d = X.shape[1]
     dev = X.device

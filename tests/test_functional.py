@@ -208,3 +208,38 @@ class TestLabelFlip:
         g = atk.apply(model=model, batch=(x, y))
         d = sum(p.numel() for p in model.parameters())
         assert g.shape == (d,)
+
+
+class TestGeomedFixedIters:
+    """Round-2: poll-free Weiszfeld (fixed_iters) on the CPU paths."""
+
+    def test_dispatch_cpu_fixed_iters(self):
+        import torch
+
+        from byzpy_amd.hip import dispatch as D
+        from byzpy_amd.ops import functional as F
+
+        g = torch.Generator().manual_seed(0)
+        X = torch.randn(12, 257, generator=g)
+        out = D.geometric_median(X, fixed_iters=50)
+        ref = F.geometric_median(X, tol=1e-30, max_iter=50)
+        assert torch.allclose(out, ref, atol=1e-5)
+
+    def test_aggregator_fixed_iters_close_to_converged(self):
+        import torch
+
+        from byzpy_amd.aggregators import GeometricMedian
+
+        g = torch.Generator().manual_seed(1)
+        grads = [torch.randn(129, generator=g) for _ in range(9)]
+        fixed = GeometricMedian(fixed_iters=128).aggregate(grads)
+        conv = GeometricMedian(tol=1e-9).aggregate(grads)
+        assert torch.allclose(fixed, conv, atol=1e-4)
+
+    def test_aggregator_rejects_bad_fixed_iters(self):
+        import pytest
+
+        from byzpy_amd.aggregators import GeometricMedian
+
+        with pytest.raises(ValueError):
+            GeometricMedian(fixed_iters=0)

@@ -502,3 +502,52 @@ class TestGramSmallN:
         ref = F.multi_krum(X.float().cpu(), 1, 3)
         assert torch.allclose(out.float().cpu(), ref, atol=5e-2, rtol=5e-2)
 
+
+class TestCafGraphAndFixedIters:
+    """Round-2 iterative-op unblocking: hipGraph CAF blocks and poll-free
+    geometric median."""
+
+    def test_caf_graph_matches_eager(self, monkeypatch):
+        X = _rand(24, 4096, torch.float32, seed=51)
+        X[-4:] = 40.0
+        got_graph = D.caf(X, 4)
+        monkeypatch.setenv("BYZPY_CAF_GRAPH", "0")
+        got_eager = D.caf(X, 4)
+        assert torch.allclose(got_graph, got_eager, atol=1e-3, rtol=1e-3)
+
+    def test_caf_graph_matches_cpu_oracle(self):
+        X = _rand(16, 2048, torch.float32, seed=52)
+        X[0] = 25.0
+        X[1] = -25.0
+        got = D.caf(X, 2)
+        ref = F.caf(X.cpu(), 2)
+        assert (got.cpu() - ref).norm() < 0.05 * max(1.0, float(ref.norm()))
+
+    def test_caf_graph_replay_is_stateless(self):
+        # two different inputs through the SAME cached graph must not leak
+        X1 = _rand(16, 1024, torch.float32, seed=53)
+        X2 = _rand(16, 1024, torch.float32, seed=54)
+        a1 = D.caf(X1, 3).clone()
+        _ = D.caf(X2, 3)
+        a1_again = D.caf(X1, 3)
+        assert torch.allclose(a1, a1_again)
+
+    @pytest.mark.parametrize("dtype", DTYPES)
+    def test_geomed_fixed_iters_parity(self, dtype):
+        X = _rand(16, 2048, dtype, seed=55)
+        out = D.geometric_median(X, fixed_iters=64)
+        ref = F.geometric_median(X.float().cpu(), tol=1e-30, max_iter=64)
+        assert (out.float().cpu() - ref).norm() < 0.05 * max(1.0, float(ref.norm()))
+
+    def test_geomed_fixed_iters_capture_safe(self):
+        # the poll-free path must capture into a hipGraph without error
+        X = _rand(8, 4096, torch.bfloat16, seed=56)
+        from byzpy_amd.hip.graphs import CapturedAggregate
+
+        cap = CapturedAggregate(
+            lambda Y: D.geometric_median(Y, fixed_iters=16), X
+        )
+        out = cap.run(X).clone()
+        ref = D.geometric_median(X, fixed_iters=16)
+        assert torch.allclose(out.float(), ref.float(), atol=1e-2, rtol=1e-2)
+
