@@ -112,6 +112,42 @@ def all_to_all_rows(local: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def neighbor_exchange(
+    send_to, recv_from, vec: torch.Tensor
+) -> "dict[int, torch.Tensor]":
+    """Batched point-to-point neighbor exchange (ring(k)
+    neighbor send/recv pairs instead of a full all-gather — traffic
+    proportional to the topology degree, not world size). Sends ``vec`` to
+    each rank in ``send_to`` and receives one same-shaped tensor from each
+    rank in ``recv_from``; returns {src_rank: tensor}.
+
+    Uses dist.batch_isend_irecv (ncclGroupStart/End under RCCL) so the
+    p2p pairs ride the xGMI links concurrently and the issue order cannot
+    deadlock. Every rank in the group must post the schedule its peers
+    expect (callers derive it from the shared Topology + byzantine set).
+    """
+    bufs = {int(j): torch.empty_like(vec) for j in recv_from}
+    if not is_initialized():
+        return bufs
+    v = vec.contiguous()
+    ops = [dist.P2POp(dist.isend, v, int(j)) for j in send_to]
+    ops += [dist.P2POp(dist.irecv, bufs[int(j)], int(j)) for j in recv_from]
+    if ops:
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
+    return bufs
+
+
+def all_gather_obj(obj) -> list:
+    """All-gather a small picklable object (control plane, e.g. the
+    byzantine-rank set at engine bring-up)."""
+    if not is_initialized():
+        return [obj]
+    out: list = [None] * get_world_size()
+    dist.all_gather_object(out, obj)
+    return out
+
+
 def column_shard(X: torch.Tensor, rank: Optional[int] = None) -> torch.Tensor:
     """This rank's contiguous d-shard of an (n, d) matrix."""
     world = get_world_size()

@@ -203,3 +203,109 @@ def test_rccl_parameter_server():
 
 def test_rccl_peer_to_peer():
     _run_workers(_body_rccl_p2p)
+
+
+# -- ring(k) neighbor send/recv semantics (world_size=4) --------------------
+
+
+def _run_workers_n(target, world, *args):
+    ctx = tmp.get_context("spawn")
+    port = 29531 + abs(hash(target.__name__ + "n")) % 2000
+    procs = []
+    for rank in range(world):
+        p = ctx.Process(target=_entry_n, args=(target, rank, port, world) + args)
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(120)
+    for p in procs:
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+def _entry_n(target, rank, port, world, *args):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    import torch.distributed as dist
+
+    from byzpy_amd.parallel import dist as pdist
+
+    pdist.init_from_env()
+    try:
+        target(rank, *args)
+    finally:
+        dist.destroy_process_group()
+
+
+def _body_ring_p2p(rank):
+    """ring byzantine rank's Empire output must equal
+    -mean(received neighbor theta-halves), and ring traffic must ride
+    neighbor send/recv pairs (p2p transport), not a full all-gather."""
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+    from byzpy_amd.attacks import EmpireAttack
+    from byzpy_amd.engine.peer_to_peer.rccl import RcclPeerToPeer
+    from byzpy_amd.engine.peer_to_peer.topology import Topology
+
+    d = 6
+    state = {"params": torch.full((d,), float(rank + 1))}
+
+    def half_step():
+        return state["params"].clone()
+
+    def write(params):
+        state["params"] = params
+
+    p2p = RcclPeerToPeer(
+        half_step,
+        write,
+        CoordinateWiseMedian(),
+        topology=Topology.ring(4, 1),
+        attack=EmpireAttack(scale=-1.0) if rank == 3 else None,
+    )
+    assert p2p.transport == "p2p", p2p.transport  # ring(4,1) is sparse
+    assert p2p.byzantine_ranks == frozenset({3})
+    out = p2p.round()
+    if rank == 3:
+        # empire over the RECEIVED honest in-neighbor vectors {1.0, 3.0}
+        assert torch.allclose(out, torch.full((d,), -2.0), atol=1e-6), out
+        # byzantine rank does not write params
+        assert torch.allclose(state["params"], torch.full((d,), 4.0))
+    else:
+        expect = {0: 1.0, 1: 2.0, 2: 2.0}[rank]
+        assert torch.allclose(out, torch.full((d,), expect), atol=1e-6), (
+            rank,
+            out,
+        )
+        assert torch.allclose(state["params"], out)
+
+
+def _body_ring_honest_only(rank):
+    """Pure-honest ring: p2p phase-1 exchange only, no byz edges."""
+    from byzpy_amd.aggregators import CoordinateWiseTrimmedMean
+    from byzpy_amd.engine.peer_to_peer.rccl import RcclPeerToPeer
+    from byzpy_amd.engine.peer_to_peer.topology import Topology
+
+    d = 5
+    theta = torch.full((d,), float(rank * 10))
+    got = {}
+
+    p2p = RcclPeerToPeer(
+        lambda: theta,
+        lambda p: got.__setitem__("p", p),
+        CoordinateWiseTrimmedMean(1),
+        topology=Topology.ring(4, 1),
+    )
+    out = p2p.round()
+    nbrs = sorted({(rank - 1) % 4, (rank + 1) % 4})
+    vals = sorted([rank * 10.0] + [j * 10.0 for j in nbrs])
+    assert torch.allclose(out, torch.full((d,), vals[1]), atol=1e-6)
+
+
+def test_ring_p2p_byzantine_context():
+    _run_workers_n(_body_ring_p2p, 4)
+
+
+def test_ring_p2p_honest():
+    _run_workers_n(_body_ring_honest_only, 4)
