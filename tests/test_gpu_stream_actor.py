@@ -82,3 +82,97 @@ def test_stream_pool_runs_device_subtasks():
         await pool.close()
 
     asyncio.run(main())
+
+
+# -- round-2 depth: windowed in-flight limit as stream depth ---------------
+
+_INFLIGHT = {"cur": 0, "max": 0}
+
+
+def _probe_fn(i):
+    """Tracks concurrent executions (stream workers are in-process, so the
+    module global is shared)."""
+    import threading
+    import time as _time
+
+    lock = getattr(_probe_fn, "_lock", None)
+    if lock is None:
+        lock = threading.Lock()
+        _probe_fn._lock = lock
+    with lock:
+        _INFLIGHT["cur"] += 1
+        _INFLIGHT["max"] = max(_INFLIGHT["max"], _INFLIGHT["cur"])
+    x = torch.full((2048,), float(i), device="cuda")
+    s = float(x.sum())
+    _time.sleep(0.02)
+    with lock:
+        _INFLIGHT["cur"] -= 1
+    return s
+
+
+def test_windowed_inflight_limit_is_stream_depth():
+    """graph/pool.py promises the operator's windowed in-flight limit acts
+    as stream depth for stream workers: with max_subtasks_inflight=-1 on a
+    2-worker stream pool, at most 2 subtasks may ever run concurrently."""
+    from byzpy_amd.ops.base import Operator, OpContext
+
+    class ProbeOp(Operator):
+        name = "probe"
+        input_key = "xs"
+        supports_subtasks = True
+        max_subtasks_inflight = -1  # pool.size * 1
+
+        def create_subtasks(self, ctx, **inputs):
+            return [SubTask(fn=_probe_fn, args=(i,)) for i in range(10)]
+
+        def reduce_subtasks(self, ctx, results, **inputs):
+            return results
+
+    async def main():
+        _INFLIGHT["cur"] = 0
+        _INFLIGHT["max"] = 0
+        pool = ActorPool(ActorPoolConfig(backend="stream:0", count=2, name="g"))
+        await pool.start()
+        ctx = OpContext(pool=pool, metadata={"pool_size": 2})
+        out = await ProbeOp().run(ctx, xs=None)
+        assert out == [2048.0 * i for i in range(10)]
+        assert _INFLIGHT["max"] <= 2, _INFLIGHT["max"]
+        await pool.close()
+
+    asyncio.run(main())
+
+
+def test_stream_pool_runs_hip_kernel_subtasks():
+    """SubTasks that launch the HIP kernels themselves (D.median / D.gram)
+    through stream workers — the kernel path under the pool scheduler."""
+    from byzpy_amd.hip import dispatch as D
+
+    def kmedian(seed):
+        g = torch.Generator().manual_seed(seed)
+        X = torch.randn(16, 4096, generator=g).to("cuda", torch.bfloat16)
+        out = D.median(X)
+        from byzpy_amd.ops import functional as F
+
+        ref = F.median(X.float().cpu())
+        assert torch.allclose(out.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+        return float(out.float().sum())
+
+    def kgram(seed):
+        g = torch.Generator().manual_seed(seed)
+        X = torch.randn(8, 8192, generator=g).to("cuda", torch.bfloat16)
+        G = D.gram(X)
+        Xf = X.float()
+        assert (G - Xf @ Xf.T).abs().max().item() < 0.5
+        return float(G.trace())
+
+    async def main():
+        pool = ActorPool(ActorPoolConfig(backend="stream:0", count=3, name="k"))
+        await pool.start()
+        outs = await asyncio.gather(
+            *(pool.run_subtask(SubTask(fn=kmedian, args=(s,))) for s in range(4)),
+            *(pool.run_subtask(SubTask(fn=kgram, args=(s,))) for s in range(4)),
+        )
+        assert len(outs) == 8 and all(isinstance(o, float) for o in outs)
+        await pool.close()
+
+    asyncio.run(main())
