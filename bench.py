@@ -29,7 +29,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--n", type=int, default=64)
-    p.add_argument("--d", type=int, default=125_000_000)
+    # --dim alias: a bare "--d" after the script path trips torchrun's
+    # argparse abbreviation check (--duplicate-*); --d kept for humans
+    p.add_argument("--d", "--dim", dest="d", type=int, default=125_000_000)
     p.add_argument("--f", type=int, default=16)
     p.add_argument("--q", type=int, default=12)
     p.add_argument("--op", choices=["both", "median", "krum"], default="both")

@@ -66,12 +66,19 @@ def main() -> None:
             return D.geometric_median(mixed, fixed_iters=fixed)
         return D.geometric_median(mixed, tol=1e-6, max_iter=32)
 
+    # index tensors hoisted to device once: a python-list fancy-index does
+    # a pageable H2D copy per call, which is also capture-UNSAFE (the
+    # --graph mode died on hipErrorStreamCaptureUnsupported without this)
+    groups = [
+        torch.tensor([i] + topo.in_neighbors(i), device=dev)
+        for i in range(args.nodes)
+    ]
+
     def round_once():
         new = torch.empty_like(theta)
         if streams is None:
             for i in range(args.nodes):
-                group = [i] + topo.in_neighbors(i)
-                X = theta[group]  # (1+2k, d) view-gather
+                X = theta[groups[i]]  # (1+2k, d) gather
                 mixed = D.nnm(X, args.f)
                 new[i] = geomed(mixed)
         else:
@@ -82,8 +89,7 @@ def main() -> None:
                 s.wait_stream(cur)
             for i in range(args.nodes):
                 with torch.cuda.stream(streams[i]):
-                    group = [i] + topo.in_neighbors(i)
-                    X = theta[group]
+                    X = theta[groups[i]]
                     mixed = D.nnm(X, args.f)
                     new[i] = geomed(mixed)
             for s in streams:

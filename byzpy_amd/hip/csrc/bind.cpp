@@ -44,6 +44,8 @@ void launch_rsel_trimmed_f32(const float*, float*, unsigned int*, int, long,
                              int, hipStream_t);
 void launch_rsel_median_f32(const float*, float*, unsigned int*, int, long,
                             hipStream_t);
+void launch_rsel_median_bf16(const __hip_bfloat16*, __hip_bfloat16*,
+                             unsigned int*, int, long, hipStream_t);
 void launch_rsel_meamed_bf16(const __hip_bfloat16*, __hip_bfloat16*,
                              unsigned int*, float*, int, long, int,
                              hipStream_t);
@@ -95,8 +97,9 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
   auto out = torch::empty({(long)d}, X.options());
   if (radix_ok) {
     const bool bf16 = X.scalar_type() == torch::kBFloat16;
-    if (mode == 0 && bf16) {
-      // tuned 2-pass streaming radix select (colsel.hip rsel_pass1/2)
+    if (mode == 0 && bf16 && (d % 4) != 0) {
+      // unaligned-d fallback: tuned scalar 2-pass radix (colsel.hip);
+      // aligned d routes through the quad-load generic engine below
       auto mark = torch::empty({(long)d * 2},
                                X.options().dtype(torch::kInt32));
       launch_colsel_median_radix_bf16(
@@ -108,7 +111,10 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
     // generic multi-pass engine (rsel.hip): per-column state scratch
     auto state = torch::zeros({(long)d * 4}, X.options().dtype(torch::kInt32));
     auto* st = reinterpret_cast<unsigned int*>(state.data_ptr<int>());
-    if (mode == 0) {  // f32 MEDIAN
+    if (mode == 0 && bf16) {
+      launch_rsel_median_bf16(bf16_ptr(X), bf16_ptr_mut(out), st, n, d,
+                              cur_stream());
+    } else if (mode == 0) {  // f32 MEDIAN
       launch_rsel_median_f32(X.data_ptr<float>(), out.data_ptr<float>(), st,
                              n, d, cur_stream());
     } else if (mode == 1) {

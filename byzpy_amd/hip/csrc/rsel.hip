@@ -189,6 +189,164 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
   }
 }
 
+// -- quad-column level pass -------------------------------------------------
+// Each thread owns FOUR adjacent columns loaded as one 8 B (bf16) / 16 B
+// (f32) vector: 4x fewer memory instructions and 4x the bytes per
+// outstanding request vs the scalar walk above (which measured ~1.9 TB/s
+// per pass at 8-deep — still latency-bound). Requires d % 4 == 0 for
+// vector alignment; the scalar kernel remains the fallback.
+
+template <typename T> struct QuadLoad;
+template <> struct QuadLoad<__hip_bfloat16> {
+  typedef uint2 vec_t;
+  static DEV __hip_bfloat16 get(const vec_t& v, int e) {
+    union { unsigned short s; __hip_bfloat16 h; } c;
+    const u32 w = (e < 2) ? v.x : v.y;
+    c.s = (unsigned short)(w >> ((e & 1) * 16));
+    return c.h;
+  }
+};
+template <> struct QuadLoad<float> {
+  typedef float4 vec_t;
+  static DEV float get(const vec_t& v, int e) {
+    return e == 0 ? v.x : e == 1 ? v.y : e == 2 ? v.z : v.w;
+  }
+};
+
+// shared scan phase: advance each column's prefix by the resolved digit
+template <bool TWO>
+DEV void rsel_scan_phase(u32 (*cnt)[RS_COLS + 1], int t, int cols, long col0,
+                         u32* state, int shift, u32 t0, u32 t1) {
+  if (t < cols) {
+    const long col = col0 + t;
+    u32 p0 = state[col * 4 + 0], b0 = state[col * 4 + 1];
+    u32 p1 = 0, b1 = 0;
+    if (TWO) { p1 = state[col * 4 + 2]; b1 = state[col * 4 + 3]; }
+    u32 run0 = 0, run1 = 0;
+    bool got0 = false, got1 = !TWO;
+#pragma unroll 4
+    for (int b = 0; b < 256; ++b) {
+      const u32 packed = cnt[b][t];
+      const u32 c0 = TWO ? (packed & 0xFFFFu) : packed;
+      if (!got0 && t0 <= b0 + run0 + c0) {
+        p0 |= (u32)b << shift;
+        b0 += run0;
+        got0 = true;
+      }
+      run0 += c0;
+      if (TWO) {
+        const u32 c1 = packed >> 16;
+        if (!got1 && t1 <= b1 + run1 + c1) {
+          p1 |= (u32)b << shift;
+          b1 += run1;
+          got1 = true;
+        }
+        run1 += c1;
+      }
+    }
+    state[col * 4 + 0] = p0;
+    state[col * 4 + 1] = b0;
+    if (TWO) {
+      state[col * 4 + 2] = p1;
+      state[col * 4 + 3] = b1;
+    }
+  }
+}
+
+template <typename T, int KK, bool TWO>
+__global__ void __launch_bounds__(RS_THREADS, 2)
+rsel_level_quad_kernel(const T* __restrict__ X, const float* __restrict__ med,
+                       u32* __restrict__ state, int n, long d, int shift,
+                       u32 hi_mask, u32 t0, u32 t1) {
+  extern __shared__ __attribute__((aligned(16))) u32 rs_lds[];
+  u32 (*cnt)[RS_COLS + 1] = reinterpret_cast<u32(*)[RS_COLS + 1]>(rs_lds);
+  const int t = threadIdx.x;
+  const int qc = (t & 15) * 4;  // first column of this thread's quad
+  const int slice = t >> 4;     // 32 row slices
+  const long col0 = (long)blockIdx.x * RS_COLS;
+  const int cols = (int)min((long)RS_COLS, d - col0);
+  for (int i = t; i < 256 * (RS_COLS + 1); i += RS_THREADS)
+    reinterpret_cast<u32*>(cnt)[i] = 0;
+  __syncthreads();
+  const int nq = min(4, cols - qc);
+  if (nq > 0) {
+    u32 p0q[4], p1q[4];
+    float mq[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long col = col0 + qc + min(u, nq - 1);
+      p0q[u] = state[col * 4 + 0];
+      p1q[u] = TWO ? state[col * 4 + 2] : 0;
+      mq[u] = (KK == DEV_KEY) ? med[col] : 0.0f;
+    }
+    const T* base = X + col0 + qc;
+    typedef typename QuadLoad<T>::vec_t VT;
+    int row = slice;
+    if (nq == 4) {
+      for (; row + 224 < n; row += 256) {  // 8 vector loads in flight
+        VT v[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = *reinterpret_cast<const VT*>(base + (long)(row + 32 * j) * d);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const u32 key = make_key<T, KK>(QuadLoad<T>::get(v[j], u), mq[u]);
+            const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
+            if (TWO) {
+              const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
+              const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+              if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
+            } else if (m0) {
+              atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
+            }
+          }
+        }
+      }
+      for (; row + 96 < n; row += 128) {  // 4-deep mid loop for small n
+        VT v[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          v[j] = *reinterpret_cast<const VT*>(base + (long)(row + 32 * j) * d);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const u32 key = make_key<T, KK>(QuadLoad<T>::get(v[j], u), mq[u]);
+            const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
+            if (TWO) {
+              const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
+              const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+              if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
+            } else if (m0) {
+              atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
+            }
+          }
+        }
+      }
+    }
+    for (; row < n; row += 32) {  // tail / partial-quad: scalar loads
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        if (u >= nq) break;
+        const u32 key =
+            make_key<T, KK>(base[(long)row * d + u], mq[u]);
+        const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
+        if (TWO) {
+          const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
+          const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
+        } else if (m0) {
+          atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  rsel_scan_phase<TWO>(cnt, t, cols, col0, state, shift, t0, t1);
+}
+
 // -- finalize kernels -------------------------------------------------------
 
 // median output (or med staging for MEAMED): 0.5 * (v(rank_lo) + v(rank_hi))
@@ -359,14 +517,20 @@ void run_levels(const T* X, const float* med, u32* state, int n, long d,
   const long grid = (d + RS_COLS - 1) / RS_COLS;
   const int top_shift = 24;
   const int last_shift = (KK == VAL_BF16) ? 16 : 0;
+  const bool quad = (d % 4) == 0;  // vector-load alignment
   for (int shift = top_shift; shift >= last_shift; shift -= 8) {
     // digits above `shift` must match the resolved prefix; at the top
     // level the mask is 0 (everything matches)
     const u32 hi_mask =
         (shift + 8 >= 32) ? 0u : (0xFFFFFFFFu << (shift + 8));
-    hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
-                       dim3(RS_THREADS), RS_LDS, stream, X, med, state, n, d,
-                       shift, hi_mask, t0, t1);
+    if (quad)
+      hipLaunchKernelGGL((rsel_level_quad_kernel<T, KK, TWO>),
+                         dim3((unsigned)grid), dim3(RS_THREADS), RS_LDS,
+                         stream, X, med, state, n, d, shift, hi_mask, t0, t1);
+    else
+      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
+                         dim3(RS_THREADS), RS_LDS, stream, X, med, state, n, d,
+                         shift, hi_mask, t0, t1);
   }
 }
 
@@ -398,6 +562,18 @@ void launch_rsel_trimmed_f32(const float* X, float* out, unsigned int* state,
   hipLaunchKernelGGL((rsel_trimmed_sum_kernel<float, VAL_F32>),
                      dim3((unsigned)grid), dim3(RS_THREADS), 0, stream, X,
                      state, out, n, d, f);
+}
+
+void launch_rsel_median_bf16(const __hip_bfloat16* X, __hip_bfloat16* out,
+                             unsigned int* state, int n, long d,
+                             hipStream_t stream) {
+  const u32 t_lo = (u32)((n - 1) >> 1) + 1, t_hi = (u32)(n >> 1) + 1;
+  run_levels<__hip_bfloat16, VAL_BF16, true>(X, nullptr, state, n, d, t_lo,
+                                             t_hi, stream);
+  const long grid = (d + 255) / 256;
+  hipLaunchKernelGGL((rsel_median_out_kernel<__hip_bfloat16, VAL_BF16, false>),
+                     dim3((unsigned)grid), dim3(256), 0, stream, state, out,
+                     d);
 }
 
 void launch_rsel_median_f32(const float* X, float* out, unsigned int* state,

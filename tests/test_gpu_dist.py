@@ -19,6 +19,9 @@ def _torchrun(args, timeout=240):
     env = dict(os.environ)
     env.setdefault("MASTER_ADDR", "127.0.0.1")
     env["HSA_ENABLE_IPC_MODE_LEGACY"] = env.get("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    # running a script by PATH puts the script's dir (tests/) on sys.path,
+    # not the repo root — the child must still import byzpy_amd in-tree
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     cmd = [
         sys.executable,
         "-m",
@@ -45,17 +48,15 @@ def test_torchrun_single_rank_sharded_bitwise():
 def test_torchrun_bench_contract():
     """bench.py under torchrun ws=1 must emit the contract JSON line and
     report n_gpus=1 — the exact launch shape the driver uses for SCALE."""
+    # --dim (not --d): torchrun's argparse prefix-matches "--d" against
+    # its own --duplicate-* options even after the script path
     r = _torchrun(
         [
             os.path.join(REPO, "bench.py"),
-            "--gpus",
-            "1",
-            "--steps",
-            "3",
-            "--warmup",
-            "1",
-            "--d",
-            "4000000",
+            "--gpus=1",
+            "--steps=3",
+            "--warmup=1",
+            "--dim=4000000",
         ],
         timeout=300,
     )
