@@ -27,6 +27,7 @@
 //   [2] prefix for rank 1   [3] count below prefix1
 // Ranks are 1-based. Counts are exact after the last level.
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -92,25 +93,35 @@ DEV float val_from_key(u32 key) {
 
 // -- level pass -------------------------------------------------------------
 
-// TWO=true tracks two ranks with a packed 16/16 histogram (counts <= n
-// <= 65535); TWO=false tracks one rank with plain u32 counts.
-constexpr int RS_THREADS = 512;  // 8 row slices x 64 columns
+// TWO=true tracks two ranks with a rank-packed 16/16 histogram; TWO=false
+// tracks one rank with a COLUMN-PARITY-packed 16/16 histogram (two
+// adjacent columns share one u32). Counts <= n <= 65535 either way.
+//
+// COLS is the column-strip width per block. These passes are bound by
+// DRAM row TOUCHES (each block touches one 2*COLS/4*COLS-byte segment
+// per row; rows are d elements apart), so wider strips cut the touch
+// count proportionally — a 4-cols-per-thread vector-load variant that
+// only reduced the load-INSTRUCTION count measured flat-to-2x-slower
+// and was removed.
+constexpr int RS_THREADS = 512;
 
-template <typename T, int KK, bool TWO>
-__global__ void __launch_bounds__(RS_THREADS, 2)
+template <typename T, int KK, bool TWO, int COLS>
+__global__ void __launch_bounds__(RS_THREADS, (TWO && COLS > 64) ? 1 : 2)
 rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
                   u32* __restrict__ state, int n, long d, int shift,
                   u32 hi_mask, u32 t0, u32 t1) {
+  // LDS words per bin: TWO -> one u32 per column (rank-packed);
+  // ONE -> one u32 per column PAIR (parity-packed)
+  constexpr int W = TWO ? COLS : COLS / 2;
   extern __shared__ __attribute__((aligned(16))) u32 rs_lds[];
-  u32 (*cnt)[RS_COLS + 1] = reinterpret_cast<u32(*)[RS_COLS + 1]>(rs_lds);
-  // layout: cnt[bin][col] — adjacent threads histogram adjacent columns
-  // into adjacent banks (+1 pad de-conflicts the per-column scan)
+  u32 (*cnt)[W + 1] = reinterpret_cast<u32(*)[W + 1]>(rs_lds);
   const int t = threadIdx.x;
-  const int c = t & 63;
-  const int slice = t >> 6;
-  const long col0 = (long)blockIdx.x * RS_COLS;
-  const int cols = (int)min((long)RS_COLS, d - col0);
-  for (int i = t; i < 256 * (RS_COLS + 1); i += RS_THREADS)
+  const int c = t % COLS;
+  const int slice = t / COLS;
+  constexpr int SLICES = RS_THREADS / COLS;
+  const long col0 = (long)blockIdx.x * COLS;
+  const int cols = (int)min((long)COLS, d - col0);
+  for (int i = t; i < 256 * (W + 1); i += RS_THREADS)
     reinterpret_cast<u32*>(cnt)[i] = 0;
   __syncthreads();
   if (c < cols) {
@@ -118,14 +129,16 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     const u32 p0 = state[col * 4 + 0];
     const u32 p1 = TWO ? state[col * 4 + 2] : 0;
     const float m = (KK == DEV_KEY) ? med[col] : 0.0f;
+    const u32 one = TWO ? 1u : ((c & 1) ? 0x10000u : 1u);
+    const int w = TWO ? c : (c >> 1);
     const T* xc = X + col;
-    // 8 loads in flight per step x 8 slices: fewer is latency-bound at
-    // this strided-2B/4B-per-lane walk
+    // 8 loads in flight per slice thread: fewer is latency-bound at this
+    // strided walk
     int row = slice;
-    for (; row + 56 < n; row += 64) {
+    for (; row + 7 * SLICES < n; row += 8 * SLICES) {
       T raw[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) raw[j] = xc[(long)(row + 8 * j) * d];
+      for (int j = 0; j < 8; ++j) raw[j] = xc[(long)(row + SLICES * j) * d];
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         const u32 key = make_key<T, KK>(raw[q], m);
@@ -133,26 +146,26 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
         if (TWO) {
           const bool m1 = ((key ^ p1) & hi_mask) == 0;
           const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][c], inc);
+          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
         } else if (m0) {
-          atomicAdd(&cnt[(key >> shift) & 0xFFu][c], 1u);
+          atomicAdd(&cnt[(key >> shift) & 0xFFu][w], one);
         }
       }
     }
-    for (; row < n; row += 8) {
+    for (; row < n; row += SLICES) {
       const u32 key = make_key<T, KK>(xc[(long)row * d], m);
       const bool m0 = ((key ^ p0) & hi_mask) == 0;
       if (TWO) {
         const bool m1 = ((key ^ p1) & hi_mask) == 0;
         const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-        if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][c], inc);
+        if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
       } else if (m0) {
-        atomicAdd(&cnt[(key >> shift) & 0xFFu][c], 1u);
+        atomicAdd(&cnt[(key >> shift) & 0xFFu][w], one);
       }
     }
   }
   __syncthreads();
-  // one thread per column: advance prefix by this digit
+  // one thread per column: advance prefix by the resolved digit
   if (t < cols) {
     const long col = col0 + t;
     u32 p0 = state[col * 4 + 0], b0 = state[col * 4 + 1];
@@ -160,10 +173,12 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     if (TWO) { p1 = state[col * 4 + 2]; b1 = state[col * 4 + 3]; }
     u32 run0 = 0, run1 = 0;
     bool got0 = false, got1 = !TWO;
+    const int w = TWO ? t : (t >> 1);
+    const int sh = TWO ? 0 : (t & 1) * 16;
 #pragma unroll 4
     for (int b = 0; b < 256; ++b) {
-      const u32 packed = cnt[b][t];
-      const u32 c0 = TWO ? (packed & 0xFFFFu) : packed;
+      const u32 packed = cnt[b][w];
+      const u32 c0 = TWO ? (packed & 0xFFFFu) : ((packed >> sh) & 0xFFFFu);
       if (!got0 && t0 <= b0 + run0 + c0) {
         p0 |= (u32)b << shift;
         b0 += run0;
@@ -187,164 +202,6 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
       state[col * 4 + 3] = b1;
     }
   }
-}
-
-// -- quad-column level pass -------------------------------------------------
-// Each thread owns FOUR adjacent columns loaded as one 8 B (bf16) / 16 B
-// (f32) vector: 4x fewer memory instructions and 4x the bytes per
-// outstanding request vs the scalar walk above (which measured ~1.9 TB/s
-// per pass at 8-deep — still latency-bound). Requires d % 4 == 0 for
-// vector alignment; the scalar kernel remains the fallback.
-
-template <typename T> struct QuadLoad;
-template <> struct QuadLoad<__hip_bfloat16> {
-  typedef uint2 vec_t;
-  static DEV __hip_bfloat16 get(const vec_t& v, int e) {
-    union { unsigned short s; __hip_bfloat16 h; } c;
-    const u32 w = (e < 2) ? v.x : v.y;
-    c.s = (unsigned short)(w >> ((e & 1) * 16));
-    return c.h;
-  }
-};
-template <> struct QuadLoad<float> {
-  typedef float4 vec_t;
-  static DEV float get(const vec_t& v, int e) {
-    return e == 0 ? v.x : e == 1 ? v.y : e == 2 ? v.z : v.w;
-  }
-};
-
-// shared scan phase: advance each column's prefix by the resolved digit
-template <bool TWO>
-DEV void rsel_scan_phase(u32 (*cnt)[RS_COLS + 1], int t, int cols, long col0,
-                         u32* state, int shift, u32 t0, u32 t1) {
-  if (t < cols) {
-    const long col = col0 + t;
-    u32 p0 = state[col * 4 + 0], b0 = state[col * 4 + 1];
-    u32 p1 = 0, b1 = 0;
-    if (TWO) { p1 = state[col * 4 + 2]; b1 = state[col * 4 + 3]; }
-    u32 run0 = 0, run1 = 0;
-    bool got0 = false, got1 = !TWO;
-#pragma unroll 4
-    for (int b = 0; b < 256; ++b) {
-      const u32 packed = cnt[b][t];
-      const u32 c0 = TWO ? (packed & 0xFFFFu) : packed;
-      if (!got0 && t0 <= b0 + run0 + c0) {
-        p0 |= (u32)b << shift;
-        b0 += run0;
-        got0 = true;
-      }
-      run0 += c0;
-      if (TWO) {
-        const u32 c1 = packed >> 16;
-        if (!got1 && t1 <= b1 + run1 + c1) {
-          p1 |= (u32)b << shift;
-          b1 += run1;
-          got1 = true;
-        }
-        run1 += c1;
-      }
-    }
-    state[col * 4 + 0] = p0;
-    state[col * 4 + 1] = b0;
-    if (TWO) {
-      state[col * 4 + 2] = p1;
-      state[col * 4 + 3] = b1;
-    }
-  }
-}
-
-template <typename T, int KK, bool TWO>
-__global__ void __launch_bounds__(RS_THREADS, 2)
-rsel_level_quad_kernel(const T* __restrict__ X, const float* __restrict__ med,
-                       u32* __restrict__ state, int n, long d, int shift,
-                       u32 hi_mask, u32 t0, u32 t1) {
-  extern __shared__ __attribute__((aligned(16))) u32 rs_lds[];
-  u32 (*cnt)[RS_COLS + 1] = reinterpret_cast<u32(*)[RS_COLS + 1]>(rs_lds);
-  const int t = threadIdx.x;
-  const int qc = (t & 15) * 4;  // first column of this thread's quad
-  const int slice = t >> 4;     // 32 row slices
-  const long col0 = (long)blockIdx.x * RS_COLS;
-  const int cols = (int)min((long)RS_COLS, d - col0);
-  for (int i = t; i < 256 * (RS_COLS + 1); i += RS_THREADS)
-    reinterpret_cast<u32*>(cnt)[i] = 0;
-  __syncthreads();
-  const int nq = min(4, cols - qc);
-  if (nq > 0) {
-    u32 p0q[4], p1q[4];
-    float mq[4];
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      const long col = col0 + qc + min(u, nq - 1);
-      p0q[u] = state[col * 4 + 0];
-      p1q[u] = TWO ? state[col * 4 + 2] : 0;
-      mq[u] = (KK == DEV_KEY) ? med[col] : 0.0f;
-    }
-    const T* base = X + col0 + qc;
-    typedef typename QuadLoad<T>::vec_t VT;
-    int row = slice;
-    if (nq == 4) {
-      for (; row + 224 < n; row += 256) {  // 8 vector loads in flight
-        VT v[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          v[j] = *reinterpret_cast<const VT*>(base + (long)(row + 32 * j) * d);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-#pragma unroll
-          for (int u = 0; u < 4; ++u) {
-            const u32 key = make_key<T, KK>(QuadLoad<T>::get(v[j], u), mq[u]);
-            const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
-            if (TWO) {
-              const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
-              const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-              if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
-            } else if (m0) {
-              atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
-            }
-          }
-        }
-      }
-      for (; row + 96 < n; row += 128) {  // 4-deep mid loop for small n
-        VT v[4];
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          v[j] = *reinterpret_cast<const VT*>(base + (long)(row + 32 * j) * d);
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-#pragma unroll
-          for (int u = 0; u < 4; ++u) {
-            const u32 key = make_key<T, KK>(QuadLoad<T>::get(v[j], u), mq[u]);
-            const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
-            if (TWO) {
-              const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
-              const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-              if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
-            } else if (m0) {
-              atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
-            }
-          }
-        }
-      }
-    }
-    for (; row < n; row += 32) {  // tail / partial-quad: scalar loads
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        if (u >= nq) break;
-        const u32 key =
-            make_key<T, KK>(base[(long)row * d + u], mq[u]);
-        const bool m0 = ((key ^ p0q[u]) & hi_mask) == 0;
-        if (TWO) {
-          const bool m1 = ((key ^ p1q[u]) & hi_mask) == 0;
-          const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], inc);
-        } else if (m0) {
-          atomicAdd(&cnt[(key >> shift) & 0xFFu][qc + u], 1u);
-        }
-      }
-    }
-  }
-  __syncthreads();
-  rsel_scan_phase<TWO>(cnt, t, cols, col0, state, shift, t0, t1);
 }
 
 // -- finalize kernels -------------------------------------------------------
@@ -509,28 +366,37 @@ rsel_meamed_sum_kernel(const T* __restrict__ X, const float* __restrict__ med,
   }
 }
 
-constexpr size_t RS_LDS = (size_t)256 * (RS_COLS + 1) * sizeof(u32);
+inline int rsel_cols() {
+  // A/B toggle for the strip-width experiment (default 128: half the DRAM
+  // row touches of 64; ONE-rank histograms stay at 2 blocks/CU via
+  // parity packing, TWO-rank at 128 runs 1 block/CU with a 132 KB LDS
+  // histogram)
+  const char* e = getenv("BYZPY_RSEL_COLS");
+  return (e && e[0] == '6') ? 64 : 128;
+}
 
 template <typename T, int KK, bool TWO>
 void run_levels(const T* X, const float* med, u32* state, int n, long d,
                 u32 t0, u32 t1, hipStream_t stream) {
-  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  const int cols = rsel_cols();
+  const long grid = (d + cols - 1) / cols;
+  const int w = TWO ? cols : cols / 2;
+  const size_t lds = (size_t)256 * (w + 1) * sizeof(u32);
   const int top_shift = 24;
   const int last_shift = (KK == VAL_BF16) ? 16 : 0;
-  const bool quad = (d % 4) == 0;  // vector-load alignment
   for (int shift = top_shift; shift >= last_shift; shift -= 8) {
     // digits above `shift` must match the resolved prefix; at the top
     // level the mask is 0 (everything matches)
     const u32 hi_mask =
         (shift + 8 >= 32) ? 0u : (0xFFFFFFFFu << (shift + 8));
-    if (quad)
-      hipLaunchKernelGGL((rsel_level_quad_kernel<T, KK, TWO>),
-                         dim3((unsigned)grid), dim3(RS_THREADS), RS_LDS,
-                         stream, X, med, state, n, d, shift, hi_mask, t0, t1);
+    if (cols == 128)
+      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO, 128>),
+                         dim3((unsigned)grid), dim3(RS_THREADS), lds, stream,
+                         X, med, state, n, d, shift, hi_mask, t0, t1);
     else
-      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
-                         dim3(RS_THREADS), RS_LDS, stream, X, med, state, n, d,
-                         shift, hi_mask, t0, t1);
+      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO, 64>),
+                         dim3((unsigned)grid), dim3(RS_THREADS), lds, stream,
+                         X, med, state, n, d, shift, hi_mask, t0, t1);
   }
 }
 
