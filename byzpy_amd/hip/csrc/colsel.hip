@@ -483,9 +483,11 @@ rsel_pass2_kernel(const unsigned short* __restrict__ X,
         const u32 key =
             (raw[q] ^ (0x8000u + ((raw[q] >> 15) & 1u) * 0x7FFFu)) & 0xFFFFu;
         const u32 hi8 = key >> 8;
+        // unconditional zero-capable increment: a per-element `if (inc)`
+        // exec save/restore chain serializes the loop (see rsel.hip)
         const u32 inc =
             (hi8 == b_lo ? 1u : 0u) | (hi8 == b_hi ? 0x10000u : 0u);
-        if (inc) atomicAdd(&cnt[c][key & 0xFFu], inc);
+        atomicAdd(&cnt[c][key & 0xFFu], inc);
       }
     }
     for (; row < n; row += 8) {
@@ -495,7 +497,7 @@ rsel_pass2_kernel(const unsigned short* __restrict__ X,
       const u32 hi8 = key >> 8;
       const u32 inc =
           (hi8 == b_lo ? 1u : 0u) | (hi8 == b_hi ? 0x10000u : 0u);
-      if (inc) atomicAdd(&cnt[c][key & 0xFFu], inc);
+      atomicAdd(&cnt[c][key & 0xFFu], inc);
     }
   }
   __syncthreads();

@@ -142,26 +142,33 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         const u32 key = make_key<T, KK>(raw[q], m);
+        // UNCONDITIONAL atomic with a possibly-zero increment: an
+        // `if (inc)` guard compiles to a per-element exec save/restore
+        // chain that serializes the whole loop (restricted levels
+        // measured as slow as full ones, 7x the sum kernel's time on
+        // the same access pattern); ds_add of 0 pipelines freely
         const bool m0 = ((key ^ p0) & hi_mask) == 0;
+        u32 inc;
         if (TWO) {
           const bool m1 = ((key ^ p1) & hi_mask) == 0;
-          const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-          if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
-        } else if (m0) {
-          atomicAdd(&cnt[(key >> shift) & 0xFFu][w], one);
+          inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+        } else {
+          inc = m0 ? one : 0u;
         }
+        atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
       }
     }
     for (; row < n; row += SLICES) {
       const u32 key = make_key<T, KK>(xc[(long)row * d], m);
       const bool m0 = ((key ^ p0) & hi_mask) == 0;
+      u32 inc;
       if (TWO) {
         const bool m1 = ((key ^ p1) & hi_mask) == 0;
-        const u32 inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
-        if (inc) atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
-      } else if (m0) {
-        atomicAdd(&cnt[(key >> shift) & 0xFFu][w], one);
+        inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
+      } else {
+        inc = m0 ? one : 0u;
       }
+      atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
     }
   }
   __syncthreads();
