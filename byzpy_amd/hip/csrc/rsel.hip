@@ -8,15 +8,19 @@
 // HBM-bound and coalesced the same way (64 adjacent columns per block,
 // 4 row slices per column):
 //
-//   level pass   : 256-bucket histogram of one 8-bit digit of a MONOTONE
+//   level pass   : 64-bucket histogram of one 6-bit digit of a MONOTONE
 //                  u32 key, restricted to elements whose higher digits
 //                  match the per-column prefix resolved so far; a scan
-//                  advances the prefix by 8 bits for 1 or 2 target ranks.
-//                  bf16 value keys need 2 levels, f32 keys 4.
+//                  advances the prefix for 1 or 2 target ranks. bf16
+//                  value keys need 3 levels, f32 keys 6. 64 bins (not
+//                  256) keep the histogram at 16.6 KB LDS so 4 blocks
+//                  (32 waves) co-reside per CU — the 256-bin variant's
+//                  66-132 KB allowed only 8-16 waves and ran 7x slower
+//                  than the identically-striding sum kernels.
 //   sum pass     : one more read of X accumulating the value sums /
 //                  boundary-tie counts the mode's closed form needs.
 //
-// Pass counts: f32 MEDIAN 4+0, TRIMMED 2+1 (bf16) / 4+1 (f32),
+// Pass counts: f32 MEDIAN 6+0, TRIMMED 3+1 (bf16) / 6+1 (f32),
 // MEAMED = median levels + 4 dev-key levels + 1 sum pass (dev = |v - med|
 // is an exact f32, so its key needs all 4 digits — a bf16-rounded dev key
 // would merge near-ties and push boundary elements across med, a
@@ -96,32 +100,30 @@ DEV float val_from_key(u32 key) {
 // TWO=true tracks two ranks with a rank-packed 16/16 histogram; TWO=false
 // tracks one rank with a COLUMN-PARITY-packed 16/16 histogram (two
 // adjacent columns share one u32). Counts <= n <= 65535 either way.
-//
-// COLS is the column-strip width per block. These passes are bound by
-// DRAM row TOUCHES (each block touches one 2*COLS/4*COLS-byte segment
-// per row; rows are d elements apart), so wider strips cut the touch
-// count proportionally — a 4-cols-per-thread vector-load variant that
-// only reduced the load-INSTRUCTION count measured flat-to-2x-slower
-// and was removed.
+// Digits may OVERLAP already-resolved bits (the bin then carries a
+// constant prefix offset inside the matched set — the scan and the
+// prefix OR are unaffected), so a fixed 6-bit digit covers any key
+// width with ceil(bits/6) levels.
 constexpr int RS_THREADS = 512;
+constexpr int RS_BINS = 64;
 
-template <typename T, int KK, bool TWO, int COLS>
-__global__ void __launch_bounds__(RS_THREADS, (TWO && COLS > 64) ? 1 : 2)
+template <typename T, int KK, bool TWO>
+__global__ void __launch_bounds__(RS_THREADS, 4)
 rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
                   u32* __restrict__ state, int n, long d, int shift,
                   u32 hi_mask, u32 t0, u32 t1) {
   // LDS words per bin: TWO -> one u32 per column (rank-packed);
   // ONE -> one u32 per column PAIR (parity-packed)
-  constexpr int W = TWO ? COLS : COLS / 2;
+  constexpr int W = TWO ? RS_COLS : RS_COLS / 2;
   extern __shared__ __attribute__((aligned(16))) u32 rs_lds[];
   u32 (*cnt)[W + 1] = reinterpret_cast<u32(*)[W + 1]>(rs_lds);
   const int t = threadIdx.x;
-  const int c = t % COLS;
-  const int slice = t / COLS;
-  constexpr int SLICES = RS_THREADS / COLS;
-  const long col0 = (long)blockIdx.x * COLS;
-  const int cols = (int)min((long)COLS, d - col0);
-  for (int i = t; i < 256 * (W + 1); i += RS_THREADS)
+  const int c = t & 63;
+  const int slice = t >> 6;
+  constexpr int SLICES = RS_THREADS / RS_COLS;
+  const long col0 = (long)blockIdx.x * RS_COLS;
+  const int cols = (int)min((long)RS_COLS, d - col0);
+  for (int i = t; i < RS_BINS * (W + 1); i += RS_THREADS)
     reinterpret_cast<u32*>(cnt)[i] = 0;
   __syncthreads();
   if (c < cols) {
@@ -132,8 +134,8 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     const u32 one = TWO ? 1u : ((c & 1) ? 0x10000u : 1u);
     const int w = TWO ? c : (c >> 1);
     const T* xc = X + col;
-    // 8 loads in flight per slice thread: fewer is latency-bound at this
-    // strided walk
+    // 8 loads in flight per slice thread; UNCONDITIONAL zero-capable
+    // atomics (an `if (inc)` guard emits per-element exec save/restore)
     int row = slice;
     for (; row + 7 * SLICES < n; row += 8 * SLICES) {
       T raw[8];
@@ -142,11 +144,6 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         const u32 key = make_key<T, KK>(raw[q], m);
-        // UNCONDITIONAL atomic with a possibly-zero increment: an
-        // `if (inc)` guard compiles to a per-element exec save/restore
-        // chain that serializes the whole loop (restricted levels
-        // measured as slow as full ones, 7x the sum kernel's time on
-        // the same access pattern); ds_add of 0 pipelines freely
         const bool m0 = ((key ^ p0) & hi_mask) == 0;
         u32 inc;
         if (TWO) {
@@ -155,7 +152,7 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
         } else {
           inc = m0 ? one : 0u;
         }
-        atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
+        atomicAdd(&cnt[(key >> shift) & (RS_BINS - 1)][w], inc);
       }
     }
     for (; row < n; row += SLICES) {
@@ -168,7 +165,7 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
       } else {
         inc = m0 ? one : 0u;
       }
-      atomicAdd(&cnt[(key >> shift) & 0xFFu][w], inc);
+      atomicAdd(&cnt[(key >> shift) & (RS_BINS - 1)][w], inc);
     }
   }
   __syncthreads();
@@ -183,7 +180,7 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     const int w = TWO ? t : (t >> 1);
     const int sh = TWO ? 0 : (t & 1) * 16;
 #pragma unroll 4
-    for (int b = 0; b < 256; ++b) {
+    for (int b = 0; b < RS_BINS; ++b) {
       const u32 packed = cnt[b][w];
       const u32 c0 = TWO ? (packed & 0xFFFFu) : ((packed >> sh) & 0xFFFFu);
       if (!got0 && t0 <= b0 + run0 + c0) {
@@ -373,37 +370,28 @@ rsel_meamed_sum_kernel(const T* __restrict__ X, const float* __restrict__ med,
   }
 }
 
-inline int rsel_cols() {
-  // A/B toggle for the strip-width experiment (default 128: half the DRAM
-  // row touches of 64; ONE-rank histograms stay at 2 blocks/CU via
-  // parity packing, TWO-rank at 128 runs 1 block/CU with a 132 KB LDS
-  // histogram)
-  const char* e = getenv("BYZPY_RSEL_COLS");
-  return (e && e[0] == '6') ? 64 : 128;
-}
-
 template <typename T, int KK, bool TWO>
 void run_levels(const T* X, const float* med, u32* state, int n, long d,
                 u32 t0, u32 t1, hipStream_t stream) {
-  const int cols = rsel_cols();
-  const long grid = (d + cols - 1) / cols;
-  const int w = TWO ? cols : cols / 2;
-  const size_t lds = (size_t)256 * (w + 1) * sizeof(u32);
-  const int top_shift = 24;
-  const int last_shift = (KK == VAL_BF16) ? 16 : 0;
-  for (int shift = top_shift; shift >= last_shift; shift -= 8) {
-    // digits above `shift` must match the resolved prefix; at the top
-    // level the mask is 0 (everything matches)
+  const long grid = (d + RS_COLS - 1) / RS_COLS;
+  const int w = TWO ? RS_COLS : RS_COLS / 2;
+  const size_t lds = (size_t)RS_BINS * (w + 1) * sizeof(u32);
+  // 6-bit digits, overlapping at the bottom: bf16 keys (top 16 bits)
+  // resolve in 3 levels, f32/dev keys in 6
+  static const int plan16[] = {26, 20, 16};
+  static const int plan32[] = {26, 20, 14, 8, 2, 0};
+  const int* plan = (KK == VAL_BF16) ? plan16 : plan32;
+  const int nlev = (KK == VAL_BF16) ? 3 : 6;
+  int prev_shift = 32;
+  for (int li = 0; li < nlev; ++li) {
+    const int shift = plan[li];
+    // all bits resolved so far must match the per-column prefix
     const u32 hi_mask =
-        (shift + 8 >= 32) ? 0u : (0xFFFFFFFFu << (shift + 8));
-    if (cols == 128)
-      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO, 128>),
-                         dim3((unsigned)grid), dim3(RS_THREADS), lds, stream,
-                         X, med, state, n, d, shift, hi_mask, t0, t1);
-    else
-      hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO, 64>),
-                         dim3((unsigned)grid), dim3(RS_THREADS), lds, stream,
-                         X, med, state, n, d, shift, hi_mask, t0, t1);
+        (prev_shift >= 32) ? 0u : (0xFFFFFFFFu << prev_shift);
+    hipLaunchKernelGGL((rsel_level_kernel<T, KK, TWO>), dim3((unsigned)grid),
+                       dim3(RS_THREADS), lds, stream, X, med, state, n, d,
+                       shift, hi_mask, t0, t1);
+    prev_shift = shift;
   }
 }
 
