@@ -35,8 +35,6 @@ void launch_weiszfeld_update(const T*, const float*, const float*, float*,
 template <typename T>
 void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
-void launch_colsel_median_radix_bf16(const __hip_bfloat16*, __hip_bfloat16*,
-                                     unsigned int*, int, long, hipStream_t);
 // rsel.hip: generic multi-pass radix-select for the other large-n modes
 void launch_rsel_trimmed_bf16(const __hip_bfloat16*, __hip_bfloat16*,
                               unsigned int*, int, long, int, hipStream_t);
@@ -87,27 +85,19 @@ torch::Tensor colsel(torch::Tensor X, int64_t mode, int64_t f) {
   check_matrix(X);
   const int n = (int)X.size(0);
   const long d = (long)X.size(1);
-  // measured crossover vs the cooperative LDS path: streaming radix wins
-  // for n > ~192 (and is the ONLY path past 512); every mode x dtype now
-  // has a radix path to n <= 65535
-  const bool radix_ok = n > 192 && n <= 65535;
+  // measured crossovers: register kernels to n = 64; past that the
+  // 64-bin streaming radix engine (rsel.hip, ~5.3 TB/s per pass) beats
+  // the cooperative LDS sort at every measured shape unless d is tiny
+  // (launch-count-bound) — the LDS path stays for 64 < n <= 512 at
+  // small d, radix is the only path above 512
+  const bool radix_ok =
+      n <= 65535 && (n > 512 || (n > 64 && d >= 32768));
   TORCH_CHECK(n >= 1 && (n <= 512 || radix_ok),
               "colsel supports 1 <= n <= 65535, got ", n);
   TORCH_CHECK(f >= 0 && 2 * f < n, "bad f for colsel");
   auto out = torch::empty({(long)d}, X.options());
   if (radix_ok) {
     const bool bf16 = X.scalar_type() == torch::kBFloat16;
-    if (mode == 0 && bf16 && (d % 4) != 0) {
-      // unaligned-d fallback: tuned scalar 2-pass radix (colsel.hip);
-      // aligned d routes through the quad-load generic engine below
-      auto mark = torch::empty({(long)d * 2},
-                               X.options().dtype(torch::kInt32));
-      launch_colsel_median_radix_bf16(
-          bf16_ptr(X), bf16_ptr_mut(out),
-          reinterpret_cast<unsigned int*>(mark.data_ptr<int>()), n, d,
-          cur_stream());
-      return out;
-    }
     // generic multi-pass engine (rsel.hip): per-column state scratch
     auto state = torch::zeros({(long)d * 4}, X.options().dtype(torch::kInt32));
     auto* st = reinterpret_cast<unsigned int*>(state.data_ptr<int>());

@@ -377,152 +377,9 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
   }
 }
 
-// ---------------------------------------------------------------------------
-// Radix-select MEDIAN for large n (bf16, n > 64, no upper cap): two
-// STREAMING passes over X instead of an in-LDS sort (which re-touches
-// every element log^2 P times and runs at ~0.3 TB/s for n >= 128).
-//   pass 1: per-column 256-bucket histogram of the HIGH byte of the
-//           sortable u16 key; scan locates the bucket holding each median
-//           rank and the count below it -> packed marker per column.
-//   pass 2: histogram of the LOW byte restricted to the marked bucket(s);
-//           scan resolves the exact key(s); reconstruct the bf16 median.
-// Both passes read X coalesced (64 adjacent columns per block, 4 row
-// slices per column) and are HBM-bound.
-// ---------------------------------------------------------------------------
-
-constexpr int RSEL_COLS = 64;
-
-// marker layout: (bucket << 16) | count_below
-DEV u32 rsel_mark(u32 bucket, u32 below) { return (bucket << 16) | below; }
-
-constexpr int RSEL_THREADS = 512;  // 8 row slices x 64 columns
-
-__global__ void __launch_bounds__(RSEL_THREADS, 2)
-rsel_pass1_kernel(const unsigned short* __restrict__ X, u32* __restrict__ mark,
-                  int n, long d) {
-  // dynamic LDS: 64 x 257 u32 (65.8 KB) exceeds the 64 KB static limit
-  extern __shared__ __attribute__((aligned(16))) u32 rsel_lds[];
-  u32 (*cnt)[257] = reinterpret_cast<u32(*)[257]>(rsel_lds);
-  const int t = threadIdx.x;
-  const int c = t & 63;
-  const int slice = t >> 6;  // 8 row slices
-  const long col0 = (long)blockIdx.x * RSEL_COLS;
-  const int cols = (int)min((long)RSEL_COLS, d - col0);
-  for (int i = t; i < RSEL_COLS * 257; i += RSEL_THREADS)
-    reinterpret_cast<u32*>(cnt)[i] = 0;
-  __syncthreads();
-  if (c < cols) {
-    const unsigned short* xc = X + col0 + c;
-    int row = slice;
-    // 8 independent loads in flight per step: fewer leaves the kernel
-    // latency-bound at this 2 B/lane stride (a 256-thread 4-deep version
-    // measured ~720 GB/s per pass; streaming peak wants ~12 KB in flight
-    // per CU)
-    for (; row + 56 < n; row += 64) {
-      u32 b[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) b[j] = (u32)xc[(long)(row + 8 * j) * d];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        atomicAdd(&cnt[c][((b[j] ^ (0x8000u + ((b[j] >> 15) & 1u) * 0x7FFFu)) &
-                           0xFFFFu) >> 8], 1u);
-    }
-    for (; row < n; row += 8) {
-      const u32 bits = (u32)xc[(long)row * d];
-      const u32 s = (bits >> 15) & 1u;
-      const u32 key = (bits ^ (0x8000u + s * 0x7FFFu)) & 0xFFFFu;
-      atomicAdd(&cnt[c][key >> 8], 1u);
-    }
-  }
-  __syncthreads();
-  // one thread per column scans its 256 buckets for both median ranks
-  if (t < cols) {
-    const u32 t_lo = (u32)((n - 1) >> 1) + 1;  // 1-based target ranks
-    const u32 t_hi = (u32)(n >> 1) + 1;
-    u32 run = 0, m_lo = 0, m_hi = 0;
-    bool got_lo = false, got_hi = false;
-#pragma unroll 4
-    for (int b = 0; b < 256; ++b) {
-      const u32 nb = run + cnt[t][b];
-      if (!got_lo && t_lo <= nb) { m_lo = rsel_mark(b, run); got_lo = true; }
-      if (!got_hi && t_hi <= nb) { m_hi = rsel_mark(b, run); got_hi = true; }
-      run = nb;
-    }
-    mark[(col0 + t) * 2 + 0] = m_lo;
-    mark[(col0 + t) * 2 + 1] = m_hi;
-  }
-}
-
-__global__ void __launch_bounds__(RSEL_THREADS, 2)
-rsel_pass2_kernel(const unsigned short* __restrict__ X,
-                  const u32* __restrict__ mark,
-                  unsigned short* __restrict__ out, int n, long d) {
-  // ONE packed histogram serves both median ranks: low 16 bits count the
-  // rank-lo bucket, high 16 the rank-hi bucket (counts <= n <= 65535)
-  extern __shared__ __attribute__((aligned(16))) u32 rsel_lds[];
-  u32 (*cnt)[257] = reinterpret_cast<u32(*)[257]>(rsel_lds);
-  const int t = threadIdx.x;
-  const int c = t & 63;
-  const int slice = t >> 6;
-  const long col0 = (long)blockIdx.x * RSEL_COLS;
-  const int cols = (int)min((long)RSEL_COLS, d - col0);
-  for (int i = t; i < RSEL_COLS * 257; i += RSEL_THREADS)
-    reinterpret_cast<u32*>(cnt)[i] = 0;
-  __syncthreads();
-  if (c < cols) {
-    const u32 b_lo = mark[(col0 + c) * 2 + 0] >> 16;
-    const u32 b_hi = mark[(col0 + c) * 2 + 1] >> 16;
-    const unsigned short* xc = X + col0 + c;
-    int row = slice;
-    for (; row + 56 < n; row += 64) {
-      u32 raw[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) raw[j] = (u32)xc[(long)(row + 8 * j) * d];
-#pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        const u32 key =
-            (raw[q] ^ (0x8000u + ((raw[q] >> 15) & 1u) * 0x7FFFu)) & 0xFFFFu;
-        const u32 hi8 = key >> 8;
-        // unconditional zero-capable increment: a per-element `if (inc)`
-        // exec save/restore chain serializes the loop (see rsel.hip)
-        const u32 inc =
-            (hi8 == b_lo ? 1u : 0u) | (hi8 == b_hi ? 0x10000u : 0u);
-        atomicAdd(&cnt[c][key & 0xFFu], inc);
-      }
-    }
-    for (; row < n; row += 8) {
-      const u32 bits = (u32)xc[(long)row * d];
-      const u32 s = (bits >> 15) & 1u;
-      const u32 key = (bits ^ (0x8000u + s * 0x7FFFu)) & 0xFFFFu;
-      const u32 hi8 = key >> 8;
-      const u32 inc =
-          (hi8 == b_lo ? 1u : 0u) | (hi8 == b_hi ? 0x10000u : 0u);
-      atomicAdd(&cnt[c][key & 0xFFu], inc);
-    }
-  }
-  __syncthreads();
-  if (t < cols) {
-    const u32 t_lo = (u32)((n - 1) >> 1) + 1;
-    const u32 t_hi = (u32)(n >> 1) + 1;
-    const u32 m0 = mark[(col0 + t) * 2 + 0], m1 = mark[(col0 + t) * 2 + 1];
-    u32 need0 = t_lo - (m0 & 0xFFFFu);  // residual rank within the bucket
-    u32 need1 = t_hi - (m1 & 0xFFFFu);
-    u32 key0 = 0, key1 = 0;
-    u32 run0 = 0, run1 = 0;
-    bool got0 = false, got1 = false;
-#pragma unroll 4
-    for (int b = 0; b < 256; ++b) {
-      run0 += cnt[t][b] & 0xFFFFu;
-      run1 += cnt[t][b] >> 16;
-      if (!got0 && need0 <= run0) { key0 = ((m0 >> 16) << 8) | b; got0 = true; }
-      if (!got1 && need1 <= run1) { key1 = ((m1 >> 16) << 8) | b; got1 = true; }
-    }
-    const float med = 0.5f * (key_to_float(key0) + key_to_float(key1));
-    union { unsigned short s; __hip_bfloat16 h; } o;
-    o.h = __float2bfloat16(med);
-    out[col0 + t] = o.s;
-  }
-}
+// (the specialized 2-pass 256-bin bf16 median radix kernels that lived
+// here were retired: the generic 64-bin engine in rsel.hip runs the same
+// selection in 3 passes at 32 waves/CU and measures 2x faster)
 
 // ---------------------------------------------------------------------------
 // LDS variant, 64 < n <= 512: block-COOPERATIVE batched bitonic. One block
@@ -658,20 +515,6 @@ static void launch_colsel_typed(const T* X, T* out, int n, long d, int mode,
       hipLaunchKernelGGL((colsel_lds_kernel<MEAMED, T>), dim3(grid),
                          dim3(LDS_THREADS), lds, stream, X, out, n, d, f, P);
   }
-}
-
-void launch_colsel_median_radix_bf16(const __hip_bfloat16* X,
-                                     __hip_bfloat16* out, unsigned int* mark,
-                                     int n, long d, hipStream_t stream) {
-  const int block = 512;  // must match RSEL_THREADS
-  const long grid = (d + RSEL_COLS - 1) / RSEL_COLS;
-  const size_t lds = (size_t)RSEL_COLS * 257 * sizeof(u32);
-  const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
-  unsigned short* Ou = reinterpret_cast<unsigned short*>(out);
-  hipLaunchKernelGGL(rsel_pass1_kernel, dim3((unsigned)grid), dim3(block),
-                     lds, stream, Xu, mark, n, d);
-  hipLaunchKernelGGL(rsel_pass2_kernel, dim3((unsigned)grid), dim3(block),
-                     lds, stream, Xu, mark, Ou, n, d);
 }
 
 void launch_colsel_f32(const float* X, float* out, int n, long d, int mode,

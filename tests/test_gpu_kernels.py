@@ -551,3 +551,26 @@ class TestCafGraphAndFixedIters:
         ref = D.geometric_median(X, fixed_iters=16)
         assert torch.allclose(out.float(), ref.float(), atol=1e-2, rtol=1e-2)
 
+
+
+class TestRadixLowCrossover:
+    """The radix engine now takes over from n > 64 when d >= 32768."""
+
+    @pytest.mark.parametrize("n", [65, 100, 150, 192])
+    def test_median_exact_both_dtypes(self, n):
+        for dtype in DTYPES:
+            X = _rand(n, 40000, dtype, seed=n)
+            out = D.median(X)
+            ref = F.median(X.float().cpu()).to(dtype)
+            assert torch.equal(out.cpu(), ref), f"n={n} {dtype}"
+
+    def test_trimmed_and_meamed_crossover_shapes(self):
+        X = _rand(100, 40000, torch.bfloat16, seed=9)
+        assert torch.allclose(
+            D.trimmed_mean(X, 10).float().cpu(),
+            F.trimmed_mean(X.float().cpu(), 10),
+            atol=2e-2, rtol=2e-2,
+        )
+        out = D.mean_of_medians(X, 10)
+        ref = _meamed_tie_avg_ref(X.float().cpu(), 10)
+        assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
