@@ -55,6 +55,13 @@ void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
 template <typename T>
 void launch_caf_colsum(const T*, const float*, const float*, const float*,
                        float*, int, long, hipStream_t);
+// subsets.hip: device-side exact subset searches (K11)
+void launch_smea_select(const float*, const int*, int, int, int,
+                        unsigned long long*, hipStream_t);
+void launch_mda_pass1(const float*, const int*, int, int, int, unsigned int*,
+                      hipStream_t);
+void launch_mda_pass2(const float*, const int*, int, int, int, unsigned int*,
+                      int*, int*, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
 void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
@@ -482,6 +489,61 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
   return out;
 }
 
+// SMEA subset selection fully on device (subsets.hip): returns the
+// winning combo INDEX packed in the low 32 bits of a u64 key tensor.
+torch::Tensor smea_select(torch::Tensor G, torch::Tensor combos) {
+  TORCH_CHECK(G.is_cuda() && G.dim() == 2 && G.size(0) == G.size(1) &&
+              G.scalar_type() == torch::kFloat32 && G.is_contiguous());
+  TORCH_CHECK(combos.is_cuda() && combos.dim() == 2 &&
+              combos.scalar_type() == torch::kInt32 && combos.is_contiguous());
+  const int n = (int)G.size(0);
+  const int C = (int)combos.size(0);
+  const int m = (int)combos.size(1);
+  TORCH_CHECK(m >= 1 && m <= 64 && C >= 1);
+  auto best = torch::full({1}, -1, G.options().dtype(torch::kInt64));
+  launch_smea_select(G.data_ptr<float>(), combos.data_ptr<int>(), n, m, C,
+                     reinterpret_cast<unsigned long long*>(
+                         best.data_ptr<int64_t>()),
+                     cur_stream());
+  return best;
+}
+
+// MDA two-pass device search: returns (found[npairs], subsets[npairs, m]);
+// the first found prefix (pair-lex order) holds the lex-smallest optimal
+// subset. No host sync anywhere.
+std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f) {
+  TORCH_CHECK(D2.is_cuda() && D2.dim() == 2 && D2.size(0) == D2.size(1) &&
+              D2.scalar_type() == torch::kFloat32 && D2.is_contiguous());
+  const int n = (int)D2.size(0);
+  const int m = n - (int)f;
+  TORCH_CHECK(n <= 64, "mda_select supports n <= 64");
+  TORCH_CHECK(m >= 2 && m <= n, "mda_select needs 2 <= n - f <= n");
+  // (a, b) prefix pairs in lexicographic order
+  const int npairs = n * (n - 1) / 2;
+  auto pairs_cpu = torch::empty({npairs, 2}, torch::kInt32);
+  {
+    auto acc = pairs_cpu.accessor<int, 2>();
+    int k = 0;
+    for (int a = 0; a < n; ++a)
+      for (int b = a + 1; b < n; ++b) {
+        acc[k][0] = a;
+        acc[k][1] = b;
+        ++k;
+      }
+  }
+  auto pairs = pairs_cpu.to(D2.device());
+  auto best = torch::full({1}, -1, D2.options().dtype(torch::kInt32));
+  auto subsets = torch::zeros({npairs, m}, D2.options().dtype(torch::kInt32));
+  auto found = torch::zeros({npairs}, D2.options().dtype(torch::kInt32));
+  auto* bp = reinterpret_cast<unsigned int*>(best.data_ptr<int>());
+  launch_mda_pass1(D2.data_ptr<float>(), pairs.data_ptr<int>(), n, m, npairs,
+                   bp, cur_stream());
+  launch_mda_pass2(D2.data_ptr<float>(), pairs.data_ptr<int>(), n, m, npairs,
+                   bp, subsets.data_ptr<int>(), found.data_ptr<int>(),
+                   cur_stream());
+  return {found, subsets};
+}
+
 }  // namespace
 
 // -- CAF fused power-iteration pair --------------------------
@@ -558,4 +620,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cc_iter", &cc_iter);
   m.def("cc_apply", &cc_apply);
   m.def("mda_search", &mda_search, "exact min-diameter subset (host DFS)");
+  m.def("smea_select", &smea_select, "device SMEA subset selection (K11)");
+  m.def("mda_select", &mda_select, "device MDA two-pass B&B search (K11)");
 }

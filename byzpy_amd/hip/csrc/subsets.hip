@@ -1,0 +1,278 @@
+// Device-side exact subset searches for gfx950.
+//
+// SMEA: among C(n, m) row subsets, find the one whose centered subset
+// Gram H G_sub H has the smallest max eigenvalue (reference
+// smea.py:63-107 runs batched host eigvalsh). Here: ONE WAVE per subset,
+// the m x m centered Gram staged in LDS, a fixed-sweep cyclic Jacobi
+// eigensolve executed wave-cooperatively, and a packed (eigkey, combo)
+// u64 atomicMin so ties resolve to the lexicographically-smallest combo
+// index exactly like the oracle's block scan.
+//
+// MDA: exact minimum-diameter (n-f)-subset (reference
+// minimum_diameter_average.py:267-386 runs seeded-DFS subtask batches;
+// round 1 copied D2 to the host for a serial C++ branch-and-bound).
+// Here the reference's seed_prefix=2 decomposition maps to workgroups:
+// one workgroup per (a, b) prefix pair runs a bounded DFS over the
+// remaining candidates with per-level incremental max-distance arrays in
+// LDS, sharing the global best diameter through an atomicMin every
+// improvement. A second bounded pass recovers the lexicographically
+// smallest optimal subset (host parity: bind.cpp mda_search's dfs2).
+#include "common.h"
+
+namespace {
+
+typedef unsigned int u32;
+typedef unsigned long long u64;
+
+// non-negative f32 -> monotone u32 (covers tiny negative rounding too)
+DEV u32 f32_key(float v) {
+  u32 b = __float_as_uint(v);
+  return (b >> 31) ? ~b : (b | 0x80000000u);
+}
+
+// ---------------------------------------------------------------------------
+// SMEA
+// ---------------------------------------------------------------------------
+
+constexpr int SMEA_SWEEPS = 10;
+
+// one 64-lane wave per subset; LDS: A[m*m] + rows[m]
+__global__ void __launch_bounds__(64)
+smea_select_kernel(const float* __restrict__ G, const int* __restrict__ combos,
+                   int n, int m, int C, u64* __restrict__ best) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* A = smem;                 // m*m
+  float* rmean = smem + m * m;     // m
+  __shared__ int rows[64];
+  const int lane = threadIdx.x;
+
+  for (int c = blockIdx.x; c < C; c += gridDim.x) {
+    if (lane < m) rows[lane] = combos[(long)c * m + lane];
+    __syncthreads();
+    // subset Gram
+    for (int e = lane; e < m * m; e += 64) {
+      const int i = e / m, j = e % m;
+      A[e] = G[(long)rows[i] * n + rows[j]];
+    }
+    __syncthreads();
+    // center: A <- A - rmean_i - rmean_j + gmean
+    if (lane < m) {
+      float s = 0.0f;
+      for (int j = 0; j < m; ++j) s += A[lane * m + j];
+      rmean[lane] = s / m;
+    }
+    __syncthreads();
+    float gm = 0.0f;
+    for (int j = 0; j < m; ++j) gm += rmean[j];
+    gm /= m;
+    for (int e = lane; e < m * m; e += 64) {
+      const int i = e / m, j = e % m;
+      A[e] = A[e] - rmean[i] - rmean[j] + gm;
+    }
+    __syncthreads();
+
+    // cyclic Jacobi, fixed sweeps (quadratic convergence: ~6 sweeps is
+    // machine precision at m <= 64)
+    for (int sweep = 0; sweep < SMEA_SWEEPS; ++sweep) {
+      for (int p = 0; p < m - 1; ++p) {
+        for (int q = p + 1; q < m; ++q) {
+          const float apq = A[p * m + q];
+          if (fabsf(apq) < 1e-12f) continue;
+          const float app = A[p * m + p];
+          const float aqq = A[q * m + q];
+          const float tau = (aqq - app) / (2.0f * apq);
+          const float t = (tau >= 0.0f ? 1.0f : -1.0f) /
+                          (fabsf(tau) + sqrtf(1.0f + tau * tau));
+          const float cth = rsqrtf(1.0f + t * t);
+          const float sth = t * cth;
+          // rows p, q (each lane owns columns k, k+64, ...)
+          for (int k = lane; k < m; k += 64) {
+            const float akp = A[p * m + k];
+            const float akq = A[q * m + k];
+            A[p * m + k] = cth * akp - sth * akq;
+            A[q * m + k] = sth * akp + cth * akq;
+          }
+          __syncthreads();
+          // columns p, q
+          for (int k = lane; k < m; k += 64) {
+            const float apk = A[k * m + p];
+            const float aqk = A[k * m + q];
+            A[k * m + p] = cth * apk - sth * aqk;
+            A[k * m + q] = sth * apk + cth * aqk;
+          }
+          __syncthreads();
+          // exact 2x2 block (reduces accumulated error)
+          if (lane == 0) {
+            const float d = app - aqq;
+            const float app2 =
+                cth * cth * app - 2.0f * sth * cth * apq + sth * sth * aqq;
+            const float aqq2 =
+                sth * sth * app + 2.0f * sth * cth * apq + cth * cth * aqq;
+            (void)d;
+            A[p * m + p] = app2;
+            A[q * m + q] = aqq2;
+            A[p * m + q] = 0.0f;
+            A[q * m + p] = 0.0f;
+          }
+          __syncthreads();
+        }
+      }
+    }
+    float lam = -3.4e38f;
+    for (int j = lane; j < m; j += 64) lam = fmaxf(lam, A[j * m + j]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      lam = fmaxf(lam, __shfl_down(lam, off, 64));
+    if (lane == 0) {
+      const u64 key = ((u64)f32_key(lam) << 32) | (u32)c;
+      atomicMin(best, key);
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MDA
+// ---------------------------------------------------------------------------
+
+// One workgroup (one wave) per (a, b) prefix pair in lexicographic pair
+// order. LDS: D2[n][n] + maxd[m+1][n] level stack.
+// PASS 1 (FIND=false): branch-and-bound on the shared best diameter
+// (strict <, so tied subsets prune instantly — all-equal-distance inputs
+// finish after one completion).
+// PASS 2 (FIND=true): bounded lex-order DFS; the FIRST completion is the
+// prefix's lex-smallest subset with diameter <= bound; written out with a
+// found flag. The dispatcher picks the first found prefix in pair order,
+// which is the globally lex-smallest optimal subset (host parity:
+// bind.cpp mda_search's dfs2).
+template <bool FIND>
+__global__ void __launch_bounds__(64)
+mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
+               int n, int m, int npairs, u32* __restrict__ best,
+               int* __restrict__ out_subsets, int* __restrict__ out_found) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* D2 = smem;                  // n*n
+  float* maxd = smem + n * n;        // (m+1) * n level stack
+  __shared__ int chosen[66];         // chosen[1+level] = element pushed
+  __shared__ float pdiam[66];        // diameter after level pushes
+  __shared__ int cand[66];           // next candidate index per level
+  const int lane = threadIdx.x;
+
+  for (int e = lane; e < n * n; e += 64) D2[e] = D2g[e];
+  __syncthreads();
+
+  auto read_best = [&]() -> float {
+    // pass 1 reads the live shared bound; pass 2 reads the (now final)
+    // pass-1 result as a fixed bound — no host round-trip between passes
+    const u32 bk = *(volatile u32*)best;
+    const u32 bb = (bk & 0x80000000u) ? (bk ^ 0x80000000u) : ~bk;
+    return __uint_as_float(bb);
+  };
+  const float bound = FIND ? read_best() : 0.0f;
+
+  for (int pi = blockIdx.x; pi < npairs; pi += gridDim.x) {
+    const int a = pairs[pi * 2 + 0];
+    const int b = pairs[pi * 2 + 1];
+    if (FIND && lane == 0) out_found[pi] = 0;
+    const float diam0 = D2[a * n + b];
+    if (m == 2) {
+      if (FIND) {
+        if (lane == 0 && diam0 <= bound) {
+          out_found[pi] = 1;
+          out_subsets[pi * 2 + 0] = a;
+          out_subsets[pi * 2 + 1] = b;
+        }
+      } else if (lane == 0) {
+        atomicMin(best, f32_key(diam0));
+      }
+      __syncthreads();
+      continue;
+    }
+    // prefix-level prune
+    {
+      const float cb = FIND ? bound : read_best();
+      const bool dead = FIND ? (diam0 > cb) : (diam0 >= cb);
+      if (dead) { __syncthreads(); continue; }
+    }
+    for (int k = lane; k < n; k += 64)
+      maxd[0 * n + k] = fmaxf(D2[a * n + k], D2[b * n + k]);
+    pdiam[0] = diam0;
+    cand[0] = b + 1;
+    __syncthreads();
+
+    const int need = m - 2;  // elements beyond the prefix
+    int depth = 0;
+    bool done = false;
+    while (!done && depth >= 0) {
+      const float cur_best = FIND ? bound : read_best();
+      int j = cand[depth];
+      const int maxj = n - (need - depth - 1);
+      bool pushed = false;
+      while (j < maxj) {
+        const float dj = fmaxf(pdiam[depth], maxd[depth * n + j]);
+        const bool ok = FIND ? (dj <= cur_best) : (dj < cur_best);
+        if (!ok) { ++j; continue; }
+        cand[depth] = j + 1;  // resume point (same value from all lanes)
+        if (depth + 1 == need) {
+          if (FIND) {
+            if (lane == 0) {
+              out_found[pi] = 1;
+              out_subsets[(long)pi * m + 0] = a;
+              out_subsets[(long)pi * m + 1] = b;
+              for (int t2 = 0; t2 < need - 1; ++t2)
+                out_subsets[(long)pi * m + 2 + t2] = chosen[1 + t2];
+              out_subsets[(long)pi * m + 2 + need - 1] = j;
+            }
+            done = true;
+            break;
+          }
+          if (lane == 0) atomicMin(best, f32_key(dj));
+          __syncthreads();
+          // keep scanning this level against the tightened bound
+          ++j;
+          continue;
+        }
+        // push j
+        chosen[1 + depth] = j;
+        pdiam[depth + 1] = dj;
+        for (int k = lane; k < n; k += 64)
+          maxd[(depth + 1) * n + k] =
+              fmaxf(maxd[depth * n + k], D2[j * n + k]);
+        __syncthreads();
+        cand[depth + 1] = j + 1;
+        ++depth;
+        pushed = true;
+        break;
+      }
+      if (!done && !pushed) --depth;  // level exhausted: backtrack
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+void launch_smea_select(const float* G, const int* combos, int n, int m,
+                        int C, unsigned long long* best, hipStream_t stream) {
+  const int grid = C < 8192 ? C : 8192;
+  const size_t lds = (size_t)(m * m + m) * sizeof(float);
+  hipLaunchKernelGGL(smea_select_kernel, dim3(grid), dim3(64), lds, stream, G,
+                     combos, n, m, C, best);
+}
+
+void launch_mda_pass1(const float* D2, const int* pairs, int n, int m,
+                      int npairs, unsigned int* best, hipStream_t stream) {
+  const int grid = npairs < 8192 ? npairs : 8192;
+  const size_t lds = (size_t)(n * n + (m + 1) * n) * sizeof(float);
+  hipLaunchKernelGGL(mda_dfs_kernel<false>, dim3(grid), dim3(64), lds, stream,
+                     D2, pairs, n, m, npairs, best, nullptr, nullptr);
+}
+
+void launch_mda_pass2(const float* D2, const int* pairs, int n, int m,
+                      int npairs, unsigned int* best, int* out_subsets,
+                      int* out_found, hipStream_t stream) {
+  const int grid = npairs < 8192 ? npairs : 8192;
+  const size_t lds = (size_t)(n * n + (m + 1) * n) * sizeof(float);
+  hipLaunchKernelGGL(mda_dfs_kernel<true>, dim3(grid), dim3(64), lds, stream,
+                     D2, pairs, n, m, npairs, best, out_subsets, out_found);
+}

@@ -273,8 +273,18 @@ def nnm(X: torch.Tensor, f: int) -> torch.Tensor:
 
 
 def minimum_diameter_averaging(X: torch.Tensor, f: int) -> torch.Tensor:
+    n = X.shape[0]
+    m = n - f
     if _gpu(X):
         D2 = pairwise_sq_dists(X)
+        if n <= 64 and 2 <= m < n:
+            # fully device-side two-pass branch-and-bound (subsets.hip):
+            # no host D2 copy, no host sync — the winning (lex-smallest
+            # optimal) subset is selected with a cumsum one-hot mask
+            found, subsets = _hip.require().mda_select(D2, int(f))
+            first = ((found.cumsum(0) == 1) & (found == 1)).to(subsets.dtype)
+            subset_t = (subsets * first[:, None]).sum(dim=0).to(torch.int32)
+            return mean_rows(X, subset_t)
         subset = F.mda_subset(D2, f)
         idx = torch.tensor(subset, device=X.device, dtype=torch.long)
         return mean_rows(X, idx)
@@ -295,7 +305,40 @@ def monna(X: torch.Tensor, f: int, reference_index: int = 0) -> torch.Tensor:
     return F.monna(X, f, reference_index)
 
 
+_SMEA_COMBOS: dict = {}
+_SMEA_MAX_COMBOS = 1 << 17
+
+
 def smea(X: torch.Tensor, f: int) -> torch.Tensor:
+    import itertools
+    import math
+
+    n = X.shape[0]
+    m = n - f
+    if (
+        _gpu(X)
+        and 1 <= m <= 64
+        and n <= 128
+        and math.comb(n, m) <= _SMEA_MAX_COMBOS
+    ):
+        # device path (subsets.hip): one wave per subset runs a cyclic
+        # Jacobi eigensolve on the centered subset Gram in LDS; the
+        # packed (eig, combo) atomicMin reproduces the oracle's
+        # lex-smallest tie-break. Combos cached per shape.
+        key = (n, m, X.device.index)
+        combos = _SMEA_COMBOS.get(key)
+        if combos is None:
+            combos = torch.tensor(
+                list(itertools.combinations(range(n), m)),
+                dtype=torch.int32,
+                device=X.device,
+            )
+            _SMEA_COMBOS[key] = combos
+        G = gram(X)
+        best = _hip.require().smea_select(G, combos)
+        idx = (best[0] & 0xFFFFFFFF).long()
+        rows = combos[idx].to(torch.int32)
+        return mean_rows(X, rows)
     return F.smea(X, f)
 
 

@@ -574,3 +574,64 @@ class TestRadixLowCrossover:
         out = D.mean_of_medians(X, 10)
         ref = _meamed_tie_avg_ref(X.float().cpu(), 10)
         assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+class TestDeviceSubsetSearch:
+    """K11 device-side MDA/SMEA (subsets.hip): no host D2 copy, parity
+    with the CPU oracles including lex-smallest tie-breaks."""
+
+    @pytest.mark.parametrize("n,f", [(10, 3), (16, 4), (20, 6), (30, 10)])
+    def test_mda_parity(self, n, f):
+        X = _rand(n, 2048, torch.float32, seed=n * 3 + f)
+        out = D.minimum_diameter_averaging(X, f)
+        ref = F.minimum_diameter_averaging(X.cpu(), f)
+        assert torch.allclose(out.cpu(), ref, atol=1e-4), f"n={n} f={f}"
+
+    def test_mda_ties_all_equal(self):
+        # all-equal rows: every subset has diameter 0; lex-smallest wins
+        X = torch.ones(12, 256, device="cuda")
+        out = D.minimum_diameter_averaging(X, 4)
+        assert torch.allclose(out, torch.ones(256, device="cuda"))
+
+    def test_mda_clustered(self):
+        # 4 outliers far away: the tight cluster must be selected exactly
+        g = torch.Generator().manual_seed(5)
+        X = torch.randn(16, 512, generator=g).cuda()
+        X[3] += 100.0
+        X[7] -= 100.0
+        X[11] += 50.0
+        X[12] -= 50.0
+        out = D.minimum_diameter_averaging(X, 4)
+        ref = X[[i for i in range(16) if i not in (3, 7, 11, 12)]].mean(0)
+        assert torch.allclose(out, ref, atol=1e-4)
+
+    def test_mda_bf16(self):
+        X = _rand(14, 1024, torch.bfloat16, seed=77)
+        out = D.minimum_diameter_averaging(X, 4)
+        ref = F.minimum_diameter_averaging(X.float().cpu(), 4)
+        assert torch.allclose(out.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+
+    @pytest.mark.parametrize("n,f", [(10, 2), (12, 3), (16, 3)])
+    def test_smea_parity(self, n, f):
+        X = _rand(n, 1024, torch.float32, seed=n + f)
+        out = D.smea(X, f)
+        ref = F.smea(X.cpu(), f)
+        assert torch.allclose(out.cpu(), ref, atol=1e-3), f"n={n} f={f}"
+
+    def test_smea_outlier_rejection(self):
+        g = torch.Generator().manual_seed(6)
+        X = torch.randn(12, 768, generator=g).cuda()
+        X[2] *= 50.0
+        X[9] *= 50.0
+        out = D.smea(X, 2)
+        ref = F.smea(X.cpu(), 2)
+        assert torch.allclose(out.cpu(), ref, atol=1e-3)
+
+    def test_smea_bf16_device_class(self):
+        from byzpy_amd.aggregators import SMEA
+
+        X = _rand(12, 2048, torch.bfloat16, seed=8)
+        out = SMEA(3).aggregate(list(X))
+        assert out.is_cuda and out.dtype == torch.bfloat16
+        ref = SMEA(3).aggregate([g.cpu().float() for g in X])
+        assert (out.float().cpu() - ref).norm() / ref.norm().clamp_min(1e-6) < 0.02
