@@ -34,3 +34,39 @@ def main():
 
 if __name__ == "__main__":
     main()
+    caf_debug()
+
+def caf_debug():
+    import torch
+    from byzpy_amd.hip import dispatch as D
+    g = torch.Generator().manual_seed(0)
+    Z = torch.randn(64, 65536, generator=g).cuda()
+    out = D.caf(Z, 16)
+    key = (64, 65536, Z.dtype, 16, 3, 8, Z.device.index)
+    blk = D._CAF_GRAPHS.get(key)
+    if blk is not None:
+        print(f"graph replays={blk.last_replays} wsum={float(blk.w.sum()):.2f} "
+              f"best_lam={float(blk.best_lambda):.4f} active={bool(blk.active)} "
+              f"rounds_left={float(blk.rounds_left)}")
+    # eager-equivalent round count with plain torch math
+    Xf = Z.float(); n = 64; target = float(n - 32)
+    w = torch.ones(n, device="cuda")
+    gen = torch.Generator(device="cpu"); gen.manual_seed(0)
+    rounds = 0
+    for r in range(n):
+        if r % 2 == 0:
+            seeds = torch.randn(2, 65536, generator=gen).cuda()
+        wsum = w.sum()
+        mu = (w[:, None] * Xf).sum(0) / wsum
+        diffs = Xf - mu
+        v = seeds[r % 2]; v = v / v.norm()
+        for _ in range(3):
+            t = (w * (diffs @ v)) @ diffs / wsum
+            lam = t.norm(); v = t / lam.clamp_min(1e-20)
+        proj = (diffs @ v) ** 2
+        w_next = (w * (1 - proj / proj.max())).clamp_min(0)
+        rounds += 1
+        if float(wsum) <= target or float(w_next.sum()) <= 0:
+            break
+        w = w_next
+    print(f"eager-equivalent rounds={rounds} final wsum={float(w.sum()):.2f}")

@@ -532,7 +532,10 @@ std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f) {
       }
   }
   auto pairs = pairs_cpu.to(D2.device());
-  auto best = torch::full({1}, -1, D2.options().dtype(torch::kInt32));
+  // init = monotone key of +inf (0xFF800000), NOT 0xFFFFFFFF: the all-ones
+  // pattern decodes to NaN and every `dj < best` comparison goes false
+  auto best = torch::full({1}, (int)0xFF800000,
+                          D2.options().dtype(torch::kInt32));
   auto subsets = torch::zeros({npairs, m}, D2.options().dtype(torch::kInt32));
   auto found = torch::zeros({npairs}, D2.options().dtype(torch::kInt32));
   auto* bp = reinterpret_cast<unsigned int*>(best.data_ptr<int>());

@@ -421,7 +421,9 @@ class _CafGraphBlock:
         pending: list = []
         drawn = 0
         total = 0
+        self.last_replays = 0
         while total < self.n:
+            self.last_replays += 1
             need = min(self.R, self.n - total)
             while sum(p.shape[0] for p in pending) < need and drawn < self.n:
                 rows = min(block_rows, self.n - drawn)
