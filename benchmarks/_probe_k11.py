@@ -34,7 +34,19 @@ def main():
 
 if __name__ == "__main__":
     main()
-    caf_debug()
+
+def caf_small_d():
+    import torch
+    from byzpy_amd.hip import dispatch as D
+    g = torch.Generator().manual_seed(1)
+    for d in (2048, 8192, 32768):
+        Z = torch.randn(32, d, generator=g).cuda()
+        t_graph = timeit(lambda: D.caf(Z, 8))
+        os.environ["BYZPY_CAF_GRAPH"] = "0"
+        t_eager = timeit(lambda: D.caf(Z, 8))
+        del os.environ["BYZPY_CAF_GRAPH"]
+        print(f"caf n=32 d={d}: graph {t_graph:.3f} ms  eager {t_eager:.3f} ms")
+
 
 def caf_debug():
     import torch
@@ -70,3 +82,8 @@ def caf_debug():
             break
         w = w_next
     print(f"eager-equivalent rounds={rounds} final wsum={float(w.sum()):.2f}")
+
+
+if __name__ == "__main__":
+    caf_debug()
+    caf_small_d()

@@ -511,7 +511,8 @@ torch::Tensor smea_select(torch::Tensor G, torch::Tensor combos) {
 // MDA two-pass device search: returns (found[npairs], subsets[npairs, m]);
 // the first found prefix (pair-lex order) holds the lex-smallest optimal
 // subset. No host sync anywhere.
-std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f) {
+std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f,
+                                      c10::optional<torch::Tensor> ub) {
   TORCH_CHECK(D2.is_cuda() && D2.dim() == 2 && D2.size(0) == D2.size(1) &&
               D2.scalar_type() == torch::kFloat32 && D2.is_contiguous());
   const int n = (int)D2.size(0);
@@ -533,9 +534,24 @@ std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f) {
   }
   auto pairs = pairs_cpu.to(D2.device());
   // init = monotone key of +inf (0xFF800000), NOT 0xFFFFFFFF: the all-ones
-  // pattern decodes to NaN and every `dj < best` comparison goes false
-  auto best = torch::full({1}, (int)0xFF800000,
-                          D2.options().dtype(torch::kInt32));
+  // pattern decodes to NaN and every `dj < best` comparison goes false.
+  // An optional caller-provided upper bound (a real achievable diameter)
+  // seeds the key instead: pass 1 then prunes from the start, and since
+  // the bound is achievable pass 2 still finds a subset. The key
+  // transform for non-negative floats is bits | 0x80000000, done with
+  // tensor ops so the whole path stays host-sync-free.
+  torch::Tensor best;
+  if (ub.has_value()) {
+    TORCH_CHECK(ub->is_cuda() && ub->scalar_type() == torch::kFloat32 &&
+                ub->numel() == 1);
+    // nudge up one ulp so pass-1's strict < can re-find an exact tie
+    auto bits = torch::nan_to_num(ub.value(), 0.0, 3.0e38, 0.0)
+                    .view(torch::kInt32);
+    best = (bits + 1).bitwise_or((int)0x80000000).contiguous();
+  } else {
+    best = torch::full({1}, (int)0xFF800000,
+                       D2.options().dtype(torch::kInt32));
+  }
   auto subsets = torch::zeros({npairs, m}, D2.options().dtype(torch::kInt32));
   auto found = torch::zeros({npairs}, D2.options().dtype(torch::kInt32));
   auto* bp = reinterpret_cast<unsigned int*>(best.data_ptr<int>());
@@ -624,5 +640,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cc_apply", &cc_apply);
   m.def("mda_search", &mda_search, "exact min-diameter subset (host DFS)");
   m.def("smea_select", &smea_select, "device SMEA subset selection (K11)");
-  m.def("mda_select", &mda_select, "device MDA two-pass B&B search (K11)");
+  m.def("mda_select", &mda_select, "device MDA two-pass B&B search (K11)",
+        py::arg("D2"), py::arg("f"), py::arg("ub") = c10::nullopt);
 }

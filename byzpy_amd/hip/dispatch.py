@@ -280,8 +280,18 @@ def minimum_diameter_averaging(X: torch.Tensor, f: int) -> torch.Tensor:
         if n <= 64 and 2 <= m < n:
             # fully device-side two-pass branch-and-bound (subsets.hip):
             # no host D2 copy, no host sync — the winning (lex-smallest
-            # optimal) subset is selected with a cumsum one-hot mask
-            found, subsets = _hip.require().mda_select(D2, int(f))
+            # optimal) subset is selected with a cumsum one-hot mask.
+            # Seed the shared bound with a cheap anchor-greedy upper
+            # bound (diameter of each row's m nearest rows) so pass 1
+            # prunes like the host B&B instead of starting from +inf.
+            idxn = torch.topk(D2, k=m, dim=1, largest=False).indices
+            sub = D2[idxn[:, :, None], idxn[:, None, :]]  # (n, m, m)
+            ub = sub.amax(dim=(1, 2)).min()
+            found, subsets = _hip.require().mda_select(D2, int(f), ub)
+            if not bool(found.any()):  # degenerate (e.g. all-inf D2)
+                subset = F.mda_subset(D2, f)
+                idx = torch.tensor(subset, device=X.device, dtype=torch.long)
+                return mean_rows(X, idx)
             first = ((found.cumsum(0) == 1) & (found == 1)).to(subsets.dtype)
             subset_t = (subsets * first[:, None]).sum(dim=0).to(torch.int32)
             return mean_rows(X, subset_t)
