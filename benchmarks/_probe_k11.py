@@ -60,6 +60,22 @@ def caf_debug():
         print(f"graph replays={blk.last_replays} wsum={float(blk.w.sum()):.2f} "
               f"best_lam={float(blk.best_lambda):.4f} active={bool(blk.active)} "
               f"rounds_left={float(blk.rounds_left)}")
+        import time as _t
+        torch.cuda.synchronize(); t0 = _t.perf_counter()
+        for _ in range(10):
+            blk.graph.replay()
+        torch.cuda.synchronize()
+        print(f"replay-only: {(_t.perf_counter()-t0)/10*1e3:.3f} ms per 8-round replay")
+        t0 = _t.perf_counter()
+        for _ in range(10):
+            sd = torch.randn(4, 65536)
+            blk.seeds[:4].copy_(sd)
+        torch.cuda.synchronize()
+        print(f"seed randn+stage (4 rows): {(_t.perf_counter()-t0)/10*1e3:.3f} ms")
+        t0 = _t.perf_counter()
+        for _ in range(10):
+            _ = bool(blk.active)
+        print(f"active sync read: {(_t.perf_counter()-t0)/10*1e3:.3f} ms")
     # eager-equivalent round count with plain torch math
     Xf = Z.float(); n = 64; target = float(n - 32)
     w = torch.ones(n, device="cuda")
