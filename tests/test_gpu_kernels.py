@@ -510,12 +510,14 @@ class TestCafGraphAndFixedIters:
     def test_caf_graph_matches_eager(self, monkeypatch):
         X = _rand(24, 4096, torch.float32, seed=51)
         X[-4:] = 40.0
+        monkeypatch.setenv("BYZPY_CAF_GRAPH", "1")
         got_graph = D.caf(X, 4)
         monkeypatch.setenv("BYZPY_CAF_GRAPH", "0")
         got_eager = D.caf(X, 4)
         assert torch.allclose(got_graph, got_eager, atol=1e-3, rtol=1e-3)
 
-    def test_caf_graph_matches_cpu_oracle(self):
+    def test_caf_graph_matches_cpu_oracle(self, monkeypatch):
+        monkeypatch.setenv("BYZPY_CAF_GRAPH", "1")
         X = _rand(16, 2048, torch.float32, seed=52)
         X[0] = 25.0
         X[1] = -25.0
@@ -523,8 +525,9 @@ class TestCafGraphAndFixedIters:
         ref = F.caf(X.cpu(), 2)
         assert (got.cpu() - ref).norm() < 0.05 * max(1.0, float(ref.norm()))
 
-    def test_caf_graph_replay_is_stateless(self):
+    def test_caf_graph_replay_is_stateless(self, monkeypatch):
         # two different inputs through the SAME cached graph must not leak
+        monkeypatch.setenv("BYZPY_CAF_GRAPH", "1")
         X1 = _rand(16, 1024, torch.float32, seed=53)
         X2 = _rand(16, 1024, torch.float32, seed=54)
         a1 = D.caf(X1, 3).clone()
