@@ -6,7 +6,14 @@ Full config: 32 x 8B bf16 = 512 GB, d-sharded over 8 x 288 GB MI355X
 the default 1e9 coordinates = 64 GB resident) — under torchrun it runs the
 real sharded collectives per iteration.
 
+Single GPU (64 GB shard):
   python benchmarks/config5_cc_bucketing.py --d-local 1000000000
+Full 8-rank launch (8 x 64 GB = the 512 GB config):
+  python -m torch.distributed.run --standalone --local-addr 127.0.0.1 \
+      --nnodes=1 --nproc-per-node 8 benchmarks/config5_cc_bucketing.py \
+      --d-local 1000000000
+With d-sharded checkpoint round-trip (one shard file per rank):
+  ... --checkpoint /tmp/c5ckpt
 """
 from __future__ import annotations
 
@@ -31,6 +38,9 @@ def main() -> None:
     p.add_argument("--M", type=int, default=10)
     p.add_argument("--bucket", type=int, default=4)
     p.add_argument("--repeat", type=int, default=3)
+    p.add_argument("--checkpoint", default=None,
+                   help="directory for a d-sharded aggregate checkpoint "
+                        "round-trip (model.rank{r}.safetensors per rank)")
     args = p.parse_args()
 
     pdist.init_from_env()
@@ -65,6 +75,28 @@ def main() -> None:
         out = step()
     sync()
     dt = (time.perf_counter() - t0) / args.repeat
+    if args.checkpoint:
+        # d-sharded checkpoint: every rank writes its own shard file; no
+        # rank ever funnels the full 512 GB state (utils/checkpoint.py)
+        from byzpy_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+
+        t1 = time.perf_counter()
+        save_checkpoint(
+            args.checkpoint,
+            round_idx=args.repeat,
+            model_state={"aggregate_shard": out},
+            rank=rank,
+            world_size=world,
+            extra_meta={"d_local": args.d_local, "n_workers": args.n},
+        )
+        pdist.barrier()
+        back = load_checkpoint(args.checkpoint, rank=rank)
+        got = back["model_state"]["aggregate_shard"].to(out.device)
+        assert torch.equal(got, out), "checkpoint round-trip mismatch"
+        if rank == 0:
+            print(f"checkpoint round-trip ok: "
+                  f"{(time.perf_counter() - t1):.2f} s for "
+                  f"{out.numel() * out.element_size() / 1e9:.2f} GB/rank")
     if rank == 0:
         reads = (args.M * (args.n // args.bucket) + args.n) * args.d_local * 2 / 1e9
         print(f"bucketing+CC(M={args.M}): {dt*1000:.1f} ms/aggregate "

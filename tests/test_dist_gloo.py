@@ -303,6 +303,50 @@ def _body_ring_honest_only(rank):
     assert torch.allclose(out, torch.full((d,), vals[1]), atol=1e-6)
 
 
+def _body_config5_path(rank, tmpdir):
+    """Config-5 engine path at gloo ws=2: d-sharded bucketing + centered
+    clipping and the per-rank sharded checkpoint round-trip (VERDICT r01
+    item 7)."""
+    from byzpy_amd.parallel import sharded
+    from byzpy_amd.parallel.dist import column_shard
+    from byzpy_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+
+    g = torch.Generator().manual_seed(11)
+    n, d = 8, 96
+    X = torch.randn(n, d, generator=g)  # identical on both ranks
+    Xs = column_shard(X, rank)
+    perm = list(range(n))
+    B = sharded.bucketing(Xs, 2, perm)
+    out = sharded.centered_clipping(B, c_tau=1.0, M=5)
+
+    # reference: full-matrix pipeline, then slice this rank's shard
+    import byzpy_amd.ops.functional as FF
+
+    ref = FF.centered_clipping(FF.bucketing(X, 2, perm), c_tau=1.0, M=5)
+    lo = rank * 48
+    assert torch.allclose(out, ref[lo : lo + 48], atol=1e-5)
+
+    save_checkpoint(
+        tmpdir,
+        round_idx=3,
+        model_state={"aggregate_shard": out},
+        rank=rank,
+        world_size=2,
+        extra_meta={"d_local": 48},
+    )
+    import torch.distributed as dist
+
+    dist.barrier()
+    back = load_checkpoint(tmpdir, rank=rank)
+    assert back["meta"]["round"] == 3
+    assert back["meta"]["world_size"] == 2
+    assert torch.equal(back["model_state"]["aggregate_shard"], out)
+
+
+def test_config5_sharded_path_and_checkpoint(tmp_path):
+    _run_workers_n(_body_config5_path, 2, str(tmp_path))
+
+
 def test_ring_p2p_byzantine_context():
     _run_workers_n(_body_ring_p2p, 4)
 
