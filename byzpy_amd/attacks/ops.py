@@ -13,6 +13,7 @@ import torch
 from torch import nn
 
 from byzpy_amd.attacks.base import Attack
+from byzpy_amd.hip import dispatch as D
 from byzpy_amd.ops import functional as F
 from byzpy_amd.utils.flatten import to_like
 
@@ -65,7 +66,7 @@ class LittleAttack(Attack):
 
     def apply(self, *, honest_grads: Any) -> Any:
         X, like = self._stack(honest_grads)
-        return to_like(F.little(X, self.f, self.N), like)
+        return to_like(D.little(X, self.f, self.N), like)
 
 
 class GaussianAttack(Attack):
@@ -90,7 +91,7 @@ class GaussianAttack(Attack):
 
     def apply(self, *, honest_grads: Any) -> Any:
         X, like = self._stack(honest_grads)
-        return to_like(F.gaussian_attack(X, self.mu, self.sigma, self.seed), like)
+        return to_like(D.gaussian_attack(X, self.mu, self.sigma, self.seed), like)
 
 
 class InfAttack(Attack):

@@ -20,7 +20,7 @@ OUT_SO = PKG_DIR / "_hip_ops.so"
 BUILD_DIR = PKG_DIR.parent / "build" / "hip"
 
 SOURCES = ["colsel.hip", "rsel.hip", "rowops.hip", "gram.hip", "krumsel.hip",
-           "subsets.hip", "bind.cpp"]
+           "subsets.hip", "attacks.hip", "bind.cpp"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

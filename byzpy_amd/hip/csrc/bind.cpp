@@ -55,6 +55,12 @@ void launch_caf_matvec(const T*, const float*, const float*, float*, int, long,
 template <typename T>
 void launch_caf_colsum(const T*, const float*, const float*, const float*,
                        float*, int, long, hipStream_t);
+// attacks.hip: K14 attack math
+template <typename T>
+void launch_little(const T*, T*, int, long, float, hipStream_t);
+template <typename T>
+void launch_gaussian_fill(T*, long, unsigned long long, float, float,
+                          hipStream_t);
 // subsets.hip: device-side exact subset searches (K11)
 void launch_smea_select(const float*, const int*, int, int, int,
                         unsigned long long*, hipStream_t);
@@ -489,6 +495,38 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
   return out;
 }
 
+// Little attack fused: per-column mu + z*sigma in one streaming pass.
+torch::Tensor little_fused(torch::Tensor X, double z) {
+  check_matrix(X);
+  const int n = (int)X.size(0);
+  const long d = (long)X.size(1);
+  auto out = torch::empty({d}, X.options());
+  if (X.scalar_type() == torch::kFloat32)
+    launch_little<float>(X.data_ptr<float>(), out.data_ptr<float>(), n, d,
+                         (float)z, cur_stream());
+  else
+    launch_little<__hip_bfloat16>(bf16_ptr(X), bf16_ptr_mut(out), n, d,
+                                  (float)z, cur_stream());
+  return out;
+}
+
+// Philox4x32-10 + Box-Muller N(mu, sigma^2) fill (no torch RNG).
+torch::Tensor gaussian_fill(int64_t d, int64_t seed, double mu, double sigma,
+                            torch::Tensor like) {
+  TORCH_CHECK(like.is_cuda(), "gaussian_fill needs a device template");
+  TORCH_CHECK(d >= 1);
+  auto out = torch::empty({d}, like.options());
+  if (like.scalar_type() == torch::kFloat32)
+    launch_gaussian_fill<float>(out.data_ptr<float>(), d,
+                                (unsigned long long)seed, (float)mu,
+                                (float)sigma, cur_stream());
+  else
+    launch_gaussian_fill<__hip_bfloat16>(bf16_ptr_mut(out), d,
+                                         (unsigned long long)seed, (float)mu,
+                                         (float)sigma, cur_stream());
+  return out;
+}
+
 // SMEA subset selection fully on device (subsets.hip): returns the
 // winning combo INDEX packed in the low 32 bits of a u64 key tensor.
 torch::Tensor smea_select(torch::Tensor G, torch::Tensor combos) {
@@ -640,6 +678,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cc_apply", &cc_apply);
   m.def("mda_search", &mda_search, "exact min-diameter subset (host DFS)");
   m.def("smea_select", &smea_select, "device SMEA subset selection (K11)");
+  m.def("little_fused", &little_fused, "fused Little attack (K14)");
+  m.def("gaussian_fill", &gaussian_fill, "philox gaussian fill (K14)");
   m.def("mda_select", &mda_select, "device MDA two-pass B&B search (K11)",
         py::arg("D2"), py::arg("f"), py::arg("ub") = c10::nullopt);
 }

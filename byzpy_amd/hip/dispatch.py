@@ -269,6 +269,46 @@ def nnm(X: torch.Tensor, f: int) -> torch.Tensor:
     return F.nnm(X, f)
 
 
+# -- attack math (K14) ------------------------------------------------------
+
+
+def little(X: torch.Tensor, f: int, N: Optional[int] = None) -> torch.Tensor:
+    """'A Little Is Enough': mu + z*sigma, fused single pass on device
+    (reference little.py:219-224 chunked sum/sum-sq decomposition)."""
+    if not _gpu(X):
+        return F.little(X, f, N)
+    import math
+
+    n = X.shape[0]
+    N_total = N if N is not None else n + f
+    s = N_total // 2 + 1 - f
+    phi_arg = (N_total - s) / N_total
+    # inverse normal CDF via erfinv (host scalar math, no torch RNG)
+    z = math.sqrt(2.0) * torch.special.erfinv(
+        torch.tensor(2.0 * phi_arg - 1.0, dtype=torch.float64)
+    ).item()
+    return _hip.require().little_fused(X.contiguous(), float(z))
+
+
+def gaussian_attack(
+    like: torch.Tensor,
+    mu: float = 0.0,
+    sigma: float = 1.0,
+    seed: Optional[int] = None,
+) -> torch.Tensor:
+    """Philox4x32-10 + Box-Muller device fill — no torch RNG on the hot
+    path. Seeded calls are deterministic per (seed, index); the stream is
+    the kernel's own, not torch's CPU sequence (callers who need the CPU
+    sequence use the functional path)."""
+    if not _gpu(like):
+        return F.gaussian_attack(like, mu, sigma, seed)
+    d = like.shape[-1]
+    if seed is None:
+        seed = int(torch.randint(0, 2**62, (1,)).item())
+    return _hip.require().gaussian_fill(int(d), int(seed), float(mu),
+                                        float(sigma), like)
+
+
 # -- remaining ops: torch composition on either device ----------------------
 
 

@@ -638,3 +638,51 @@ class TestDeviceSubsetSearch:
         assert out.is_cuda and out.dtype == torch.bfloat16
         ref = SMEA(3).aggregate([g.cpu().float() for g in X])
         assert (out.float().cpu() - ref).norm() / ref.norm().clamp_min(1e-6) < 0.02
+
+
+class TestAttackKernels:
+    """K14 attack math: fused Little pass + philox Gaussian fill."""
+
+    @pytest.mark.parametrize("dtype", DTYPES)
+    @pytest.mark.parametrize("n,d", [(10, 4097), (33, 1000), (64, 65536)])
+    def test_little_parity(self, n, d, dtype):
+        X = _rand(n, d, dtype, seed=n + d)
+        out = D.little(X, 3)
+        ref = F.little(X.float().cpu(), 3)
+        tol = dict(atol=1e-3, rtol=1e-3) if dtype == torch.float32 else dict(
+            atol=5e-2, rtol=5e-2
+        )
+        assert torch.allclose(out.float().cpu(), ref, **tol)
+
+    def test_little_attack_class_on_device(self):
+        from byzpy_amd.attacks import LittleAttack
+
+        grads = [torch.randn(2048).cuda() for _ in range(12)]
+        out = LittleAttack(3).apply(honest_grads=grads)
+        assert out.is_cuda
+        ref = F.little(torch.stack([g.cpu() for g in grads]), 3)
+        assert torch.allclose(out.cpu(), ref, atol=1e-3)
+
+    def test_gaussian_fill_statistics(self):
+        like = torch.empty(1, device="cuda")
+        out = D.gaussian_attack(
+            torch.empty(1, 1_000_000, device="cuda"), 2.0, 3.0, seed=7
+        )
+        assert out.shape == (1_000_000,)
+        assert abs(float(out.mean()) - 2.0) < 0.02
+        assert abs(float(out.std()) - 3.0) < 0.02
+        assert torch.isfinite(out).all()
+
+    def test_gaussian_fill_deterministic_per_seed(self):
+        t = torch.empty(1, 100_000, device="cuda")
+        a = D.gaussian_attack(t, 0.0, 1.0, seed=13)
+        b = D.gaussian_attack(t, 0.0, 1.0, seed=13)
+        c = D.gaussian_attack(t, 0.0, 1.0, seed=14)
+        assert torch.equal(a, b)
+        assert not torch.equal(a, c)
+
+    def test_gaussian_fill_bf16(self):
+        t = torch.empty(1, 65536, device="cuda", dtype=torch.bfloat16)
+        out = D.gaussian_attack(t, 0.0, 1.0, seed=3)
+        assert out.dtype == torch.bfloat16
+        assert abs(float(out.float().mean())) < 0.05
