@@ -20,6 +20,15 @@ def main():
     # MDA benchmark shape (n=30 d=2048 f=10)
     X = torch.randn(30, 2048, generator=g).cuda()
     print(f"mda n=30 d=2048 f=10: {timeit(lambda: D.minimum_diameter_averaging(X, 10)):.3f} ms (r01: 6.05)")
+    # host-path on the SAME box for a fair comparison
+    from byzpy_amd.hip import require
+    from byzpy_amd.ops import functional as FF
+    ext = require()
+    def host_path():
+        D2 = D.pairwise_sq_dists(X)
+        idx = ext.mda_search(D2.detach().float().cpu(), 10)
+        return D.mean_rows(X, idx.to("cuda", torch.int32))
+    print(f"mda host-DFS same box: {timeit(host_path):.3f} ms")
     Xb = X.bfloat16()
     print(f"mda bf16 same:        {timeit(lambda: D.minimum_diameter_averaging(Xb, 10)):.3f} ms")
     # SMEA shape from r01 results (n=16 f=3)
