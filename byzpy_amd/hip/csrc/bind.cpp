@@ -550,7 +550,8 @@ torch::Tensor smea_select(torch::Tensor G, torch::Tensor combos) {
 // the first found prefix (pair-lex order) holds the lex-smallest optimal
 // subset. No host sync anywhere.
 std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f,
-                                      c10::optional<torch::Tensor> ub) {
+                                      c10::optional<torch::Tensor> ub,
+                                      c10::optional<torch::Tensor> D2perm) {
   TORCH_CHECK(D2.is_cuda() && D2.dim() == 2 && D2.size(0) == D2.size(1) &&
               D2.scalar_type() == torch::kFloat32 && D2.is_contiguous());
   const int n = (int)D2.size(0);
@@ -593,8 +594,19 @@ std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f,
   auto subsets = torch::zeros({npairs, m}, D2.options().dtype(torch::kInt32));
   auto found = torch::zeros({npairs}, D2.options().dtype(torch::kInt32));
   auto* bp = reinterpret_cast<unsigned int*>(best.data_ptr<int>());
-  launch_mda_pass1(D2.data_ptr<float>(), pairs.data_ptr<int>(), n, m, npairs,
-                   bp, cur_stream());
+  // pass 1 may run on a CENTRALITY-PERMUTED copy of D2 (central rows
+  // first tightens the bound early, like the host B&B's candidate
+  // order); the bound is permutation-invariant, and pass 2 searches the
+  // ORIGINAL matrix so the lex-smallest semantics are unchanged
+  const float* d2p = D2perm.has_value() ? D2perm->data_ptr<float>()
+                                        : D2.data_ptr<float>();
+  if (D2perm.has_value()) {
+    TORCH_CHECK(D2perm->is_cuda() && D2perm->is_contiguous() &&
+                D2perm->sizes() == D2.sizes() &&
+                D2perm->scalar_type() == torch::kFloat32);
+  }
+  launch_mda_pass1(d2p, pairs.data_ptr<int>(), n, m, npairs, bp,
+                   cur_stream());
   launch_mda_pass2(D2.data_ptr<float>(), pairs.data_ptr<int>(), n, m, npairs,
                    bp, subsets.data_ptr<int>(), found.data_ptr<int>(),
                    cur_stream());
@@ -681,5 +693,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("little_fused", &little_fused, "fused Little attack (K14)");
   m.def("gaussian_fill", &gaussian_fill, "philox gaussian fill (K14)");
   m.def("mda_select", &mda_select, "device MDA two-pass B&B search (K11)",
-        py::arg("D2"), py::arg("f"), py::arg("ub") = c10::nullopt);
+        py::arg("D2"), py::arg("f"), py::arg("ub") = c10::nullopt,
+        py::arg("D2perm") = c10::nullopt);
 }
