@@ -203,16 +203,11 @@ mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
     const int need = m - 2;  // elements beyond the prefix
     int depth = 0;
     bool done = false;
-    // the shared bound lives in L2: cache it and refresh every few nodes
-    // (a volatile read per DFS node is ~500 serialized cycles)
-    float cached_best = FIND ? bound : read_best();
-    int since_read = 0;
     while (!done && depth >= 0) {
-      if (!FIND && ++since_read >= 8) {
-        cached_best = read_best();
-        since_read = 0;
-      }
-      const float cur_best = cached_best;
+      // fresh shared-bound read per node: staleness was measured FAR more
+      // expensive than the ~500-cycle L2 read (8-node caching exploded
+      // the node count 12x — collaborative pruning needs tight bounds)
+      const float cur_best = FIND ? bound : read_best();
       const int start = cand[depth];
       const int maxj = n - (need - depth - 1);
       // WAVE-PARALLEL candidate scan: n <= 64, so one ballot evaluates
@@ -254,7 +249,6 @@ mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
           dmin = fminf(dmin, __shfl_down(dmin, off, 64));
         if (lane == 0) atomicMin(best, f32_key(dmin));
         __syncthreads();
-        cached_best = fminf(cached_best, dmin);
         --depth;
         continue;
       }
