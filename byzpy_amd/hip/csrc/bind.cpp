@@ -64,10 +64,10 @@ void launch_gaussian_fill(T*, long, unsigned long long, float, float,
 // subsets.hip: device-side exact subset searches (K11)
 void launch_smea_select(const float*, const int*, int, int, int,
                         unsigned long long*, hipStream_t);
-void launch_mda_pass1(const float*, const int*, int, int, int, unsigned int*,
-                      hipStream_t);
-void launch_mda_pass2(const float*, const int*, int, int, int, unsigned int*,
-                      int*, int*, hipStream_t);
+void launch_mda_pass1(const float*, const int*, int, int, int, int,
+                      unsigned int*, hipStream_t);
+void launch_mda_pass2(const float*, const int*, int, int, int, int,
+                      unsigned int*, int*, int*, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
 void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
@@ -554,36 +554,54 @@ std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f,
                                       c10::optional<torch::Tensor> D2perm) {
   TORCH_CHECK(D2.is_cuda() && D2.dim() == 2 && D2.size(0) == D2.size(1) &&
               D2.scalar_type() == torch::kFloat32 && D2.is_contiguous());
+  (void)D2perm;  // a centrality-permuted pass-1 variant measured 15x
+                 // SLOWER (central-first lex order wanders near-optimal
+                 // subtrees B&B cannot prune); kept in the signature for
+                 // call compatibility, ignored
   const int n = (int)D2.size(0);
   const int m = n - (int)f;
   TORCH_CHECK(n <= 64, "mda_select supports n <= 64");
   TORCH_CHECK(m >= 2 && m <= n, "mda_select needs 2 <= n - f <= n");
-  // (a, b) prefix pairs in lexicographic order
-  const int npairs = n * (n - 1) / 2;
-  auto pairs_cpu = torch::empty({npairs, 2}, torch::kInt32);
-  {
-    auto acc = pairs_cpu.accessor<int, 2>();
-    int k = 0;
-    for (int a = 0; a < n; ++a)
-      for (int b = a + 1; b < n; ++b) {
-        acc[k][0] = a;
-        acc[k][1] = b;
-        ++k;
-      }
+  // P-element lexicographic prefixes: triples when the subset is deep
+  // enough (finer decomposition = shorter serial DFS tails), else pairs
+  const int P = (m >= 3 && n >= 8) ? 3 : 2;
+  long npre = 0;
+  if (P == 2) {
+    npre = (long)n * (n - 1) / 2;
+  } else {
+    npre = (long)n * (n - 1) * (n - 2) / 6;
   }
-  auto pairs = pairs_cpu.to(D2.device());
-  // init = monotone key of +inf (0xFF800000), NOT 0xFFFFFFFF: the all-ones
-  // pattern decodes to NaN and every `dj < best` comparison goes false.
-  // An optional caller-provided upper bound (a real achievable diameter)
-  // seeds the key instead: pass 1 then prunes from the start, and since
-  // the bound is achievable pass 2 still finds a subset. The key
-  // transform for non-negative floats is bits | 0x80000000, done with
-  // tensor ops so the whole path stays host-sync-free.
+  auto pre_cpu = torch::empty({npre, (long)P}, torch::kInt32);
+  {
+    auto acc = pre_cpu.accessor<int, 2>();
+    long k = 0;
+    if (P == 2) {
+      for (int a = 0; a < n; ++a)
+        for (int b = a + 1; b < n; ++b) {
+          acc[k][0] = a;
+          acc[k][1] = b;
+          ++k;
+        }
+    } else {
+      for (int a = 0; a < n; ++a)
+        for (int b = a + 1; b < n; ++b)
+          for (int c = b + 1; c < n; ++c) {
+            acc[k][0] = a;
+            acc[k][1] = b;
+            acc[k][2] = c;
+            ++k;
+          }
+    }
+  }
+  auto prefixes = pre_cpu.to(D2.device());
+  // init = monotone key of +inf (0xFF800000), NOT 0xFFFFFFFF: the
+  // all-ones pattern decodes to NaN and every comparison goes false.
+  // An optional achievable upper bound seeds the key so pass 1 prunes
+  // from the start (one ulp up so strict < can re-find an exact tie).
   torch::Tensor best;
   if (ub.has_value()) {
     TORCH_CHECK(ub->is_cuda() && ub->scalar_type() == torch::kFloat32 &&
                 ub->numel() == 1);
-    // nudge up one ulp so pass-1's strict < can re-find an exact tie
     auto bits = torch::nan_to_num(ub.value(), 0.0, 3.0e38, 0.0)
                     .view(torch::kInt32);
     best = (bits + 1).bitwise_or((int)0x80000000).contiguous();
@@ -591,25 +609,15 @@ std::vector<torch::Tensor> mda_select(torch::Tensor D2, int64_t f,
     best = torch::full({1}, (int)0xFF800000,
                        D2.options().dtype(torch::kInt32));
   }
-  auto subsets = torch::zeros({npairs, m}, D2.options().dtype(torch::kInt32));
-  auto found = torch::zeros({npairs}, D2.options().dtype(torch::kInt32));
+  auto subsets = torch::zeros({npre, (long)m},
+                              D2.options().dtype(torch::kInt32));
+  auto found = torch::zeros({npre}, D2.options().dtype(torch::kInt32));
   auto* bp = reinterpret_cast<unsigned int*>(best.data_ptr<int>());
-  // pass 1 may run on a CENTRALITY-PERMUTED copy of D2 (central rows
-  // first tightens the bound early, like the host B&B's candidate
-  // order); the bound is permutation-invariant, and pass 2 searches the
-  // ORIGINAL matrix so the lex-smallest semantics are unchanged
-  const float* d2p = D2perm.has_value() ? D2perm->data_ptr<float>()
-                                        : D2.data_ptr<float>();
-  if (D2perm.has_value()) {
-    TORCH_CHECK(D2perm->is_cuda() && D2perm->is_contiguous() &&
-                D2perm->sizes() == D2.sizes() &&
-                D2perm->scalar_type() == torch::kFloat32);
-  }
-  launch_mda_pass1(d2p, pairs.data_ptr<int>(), n, m, npairs, bp,
-                   cur_stream());
-  launch_mda_pass2(D2.data_ptr<float>(), pairs.data_ptr<int>(), n, m, npairs,
-                   bp, subsets.data_ptr<int>(), found.data_ptr<int>(),
-                   cur_stream());
+  launch_mda_pass1(D2.data_ptr<float>(), prefixes.data_ptr<int>(), P, n, m,
+                   (int)npre, bp, cur_stream());
+  launch_mda_pass2(D2.data_ptr<float>(), prefixes.data_ptr<int>(), P, n, m,
+                   (int)npre, bp, subsets.data_ptr<int>(),
+                   found.data_ptr<int>(), cur_stream());
   return {found, subsets};
 }
 

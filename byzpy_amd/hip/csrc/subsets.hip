@@ -135,52 +135,61 @@ smea_select_kernel(const float* __restrict__ G, const int* __restrict__ combos,
 // MDA
 // ---------------------------------------------------------------------------
 
-// One workgroup (one wave) per (a, b) prefix pair in lexicographic pair
-// order. LDS: D2[n][n] + maxd[m+1][n] level stack.
+// One workgroup (one wave) per lexicographic P-element prefix (P=2
+// pairs or P=3 triples — triples cut the slowest prefix's serial DFS
+// tail by ~n and put ~C(n,3) independent searchers on the chip).
+// LDS: D2[n][n] + maxd[m+1][n] level stack.
 // PASS 1 (FIND=false): branch-and-bound on the shared best diameter
-// (strict <, so tied subsets prune instantly — all-equal-distance inputs
-// finish after one completion).
+// (strict <, so tied subsets prune instantly).
 // PASS 2 (FIND=true): bounded lex-order DFS; the FIRST completion is the
-// prefix's lex-smallest subset with diameter <= bound; written out with a
-// found flag. The dispatcher picks the first found prefix in pair order,
-// which is the globally lex-smallest optimal subset (host parity:
-// bind.cpp mda_search's dfs2).
+// prefix's lex-smallest subset with diameter <= bound. The dispatcher
+// picks the first found prefix in lex order = the globally lex-smallest
+// optimal subset (host parity: bind.cpp mda_search's dfs2).
 template <bool FIND>
 __global__ void __launch_bounds__(64)
-mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
-               int n, int m, int npairs, u32* __restrict__ best,
+mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ prefixes,
+               int P, int n, int m, int npre, u32* __restrict__ best,
                int* __restrict__ out_subsets, int* __restrict__ out_found) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* D2 = smem;                  // n*n
   float* maxd = smem + n * n;        // (m+1) * n level stack
-  __shared__ int chosen[66];         // chosen[1+level] = element pushed
+  __shared__ int chosen[66];         // chosen[level] = element pushed
   __shared__ float pdiam[66];        // diameter after level pushes
   __shared__ int cand[66];           // next candidate index per level
+  __shared__ int pref[4];
   const int lane = threadIdx.x;
 
   for (int e = lane; e < n * n; e += 64) D2[e] = D2g[e];
   __syncthreads();
 
   auto read_best = [&]() -> float {
-    // pass 1 reads the live shared bound; pass 2 reads the (now final)
-    // pass-1 result as a fixed bound — no host round-trip between passes
     const u32 bk = *(volatile u32*)best;
     const u32 bb = (bk & 0x80000000u) ? (bk ^ 0x80000000u) : ~bk;
     return __uint_as_float(bb);
   };
   const float bound = FIND ? read_best() : 0.0f;
 
-  for (int pi = blockIdx.x; pi < npairs; pi += gridDim.x) {
-    const int a = pairs[pi * 2 + 0];
-    const int b = pairs[pi * 2 + 1];
+  for (int pi = blockIdx.x; pi < npre; pi += gridDim.x) {
+    if (lane < P) pref[lane] = prefixes[(long)pi * P + lane];
     if (FIND && lane == 0) out_found[pi] = 0;
-    const float diam0 = D2[a * n + b];
-    if (m == 2) {
+    __syncthreads();
+    // prefix diameter + level-0 max-distance row
+    float diam0 = 0.0f;
+    for (int i = 0; i < P; ++i)
+      for (int j = i + 1; j < P; ++j)
+        diam0 = fmaxf(diam0, D2[pref[i] * n + pref[j]]);
+    for (int k = lane; k < n; k += 64) {
+      float v = D2[pref[0] * n + k];
+      for (int i = 1; i < P; ++i) v = fmaxf(v, D2[pref[i] * n + k]);
+      maxd[0 * n + k] = v;
+    }
+    __syncthreads();
+    const int need = m - P;  // elements beyond the prefix
+    if (need == 0) {
       if (FIND) {
         if (lane == 0 && diam0 <= bound) {
           out_found[pi] = 1;
-          out_subsets[pi * 2 + 0] = a;
-          out_subsets[pi * 2 + 1] = b;
+          for (int i = 0; i < P; ++i) out_subsets[(long)pi * m + i] = pref[i];
         }
       } else if (lane == 0) {
         atomicMin(best, f32_key(diam0));
@@ -188,32 +197,26 @@ mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
       __syncthreads();
       continue;
     }
-    // prefix-level prune
     {
       const float cb = FIND ? bound : read_best();
       const bool dead = FIND ? (diam0 > cb) : (diam0 >= cb);
       if (dead) { __syncthreads(); continue; }
     }
-    for (int k = lane; k < n; k += 64)
-      maxd[0 * n + k] = fmaxf(D2[a * n + k], D2[b * n + k]);
     pdiam[0] = diam0;
-    cand[0] = b + 1;
+    cand[0] = pref[P - 1] + 1;
     __syncthreads();
 
-    const int need = m - 2;  // elements beyond the prefix
     int depth = 0;
     bool done = false;
     while (!done && depth >= 0) {
-      // fresh shared-bound read per node: staleness was measured FAR more
-      // expensive than the ~500-cycle L2 read (8-node caching exploded
-      // the node count 12x — collaborative pruning needs tight bounds)
+      // fresh shared-bound read per node: staleness was measured FAR
+      // more expensive than the L2 read (8-node caching exploded the
+      // node count 12x — collaborative pruning needs tight bounds)
       const float cur_best = FIND ? bound : read_best();
       const int start = cand[depth];
       const int maxj = n - (need - depth - 1);
       // WAVE-PARALLEL candidate scan: n <= 64, so one ballot evaluates
-      // every remaining candidate at this level at once (the serial
-      // per-candidate walk measured 9 ms at n=30 f=10; the host B&B was
-      // 6 — the scan, not the node count, was the bottleneck)
+      // every remaining candidate at this level at once
       const int j = start + lane;
       bool ok = false;
       float dj = 0.0f;
@@ -232,11 +235,11 @@ mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
           const int sel = __ffsll((long long)mask) - 1;
           if (lane == 0) {
             out_found[pi] = 1;
-            out_subsets[(long)pi * m + 0] = a;
-            out_subsets[(long)pi * m + 1] = b;
+            for (int i = 0; i < P; ++i)
+              out_subsets[(long)pi * m + i] = pref[i];
             for (int t2 = 0; t2 < need - 1; ++t2)
-              out_subsets[(long)pi * m + 2 + t2] = chosen[1 + t2];
-            out_subsets[(long)pi * m + 2 + need - 1] = start + sel;
+              out_subsets[(long)pi * m + P + t2] = chosen[t2];
+            out_subsets[(long)pi * m + P + need - 1] = start + sel;
           }
           done = true;
           break;
@@ -257,7 +260,7 @@ mda_dfs_kernel(const float* __restrict__ D2g, const int* __restrict__ pairs,
       const int jsel = start + sel;
       const float djsel = fmaxf(pdiam[depth], maxd[depth * n + jsel]);
       cand[depth] = jsel + 1;
-      chosen[1 + depth] = jsel;
+      chosen[depth] = jsel;
       pdiam[depth + 1] = djsel;
       for (int k = lane; k < n; k += 64)
         maxd[(depth + 1) * n + k] =
@@ -280,19 +283,21 @@ void launch_smea_select(const float* G, const int* combos, int n, int m,
                      combos, n, m, C, best);
 }
 
-void launch_mda_pass1(const float* D2, const int* pairs, int n, int m,
-                      int npairs, unsigned int* best, hipStream_t stream) {
-  const int grid = npairs < 8192 ? npairs : 8192;
+void launch_mda_pass1(const float* D2, const int* prefixes, int P, int n,
+                      int m, int npre, unsigned int* best,
+                      hipStream_t stream) {
+  const int grid = npre < 16384 ? npre : 16384;
   const size_t lds = (size_t)(n * n + (m + 1) * n) * sizeof(float);
   hipLaunchKernelGGL(mda_dfs_kernel<false>, dim3(grid), dim3(64), lds, stream,
-                     D2, pairs, n, m, npairs, best, nullptr, nullptr);
+                     D2, prefixes, P, n, m, npre, best, nullptr, nullptr);
 }
 
-void launch_mda_pass2(const float* D2, const int* pairs, int n, int m,
-                      int npairs, unsigned int* best, int* out_subsets,
+void launch_mda_pass2(const float* D2, const int* prefixes, int P, int n,
+                      int m, int npre, unsigned int* best, int* out_subsets,
                       int* out_found, hipStream_t stream) {
-  const int grid = npairs < 8192 ? npairs : 8192;
+  const int grid = npre < 16384 ? npre : 16384;
   const size_t lds = (size_t)(n * n + (m + 1) * n) * sizeof(float);
   hipLaunchKernelGGL(mda_dfs_kernel<true>, dim3(grid), dim3(64), lds, stream,
-                     D2, pairs, n, m, npairs, best, out_subsets, out_found);
+                     D2, prefixes, P, n, m, npre, best, out_subsets,
+                     out_found);
 }

@@ -327,10 +327,7 @@ def minimum_diameter_averaging(X: torch.Tensor, f: int) -> torch.Tensor:
             idxn = torch.topk(D2, k=m, dim=1, largest=False).indices
             sub = D2[idxn[:, :, None], idxn[:, None, :]]  # (n, m, m)
             ub = sub.amax(dim=(1, 2)).min()
-            # centrality permutation for pass 1 (host-B&B-style order)
-            perm = torch.argsort(D2.sum(dim=1))
-            D2p = D2[perm][:, perm].contiguous()
-            found, subsets = _hip.require().mda_select(D2, int(f), ub, D2p)
+            found, subsets = _hip.require().mda_select(D2, int(f), ub)
             if not bool(found.any()):  # degenerate (e.g. all-inf D2)
                 subset = F.mda_subset(D2, f)
                 idx = torch.tensor(subset, device=X.device, dtype=torch.long)
