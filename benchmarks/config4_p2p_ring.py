@@ -37,6 +37,9 @@ def main() -> None:
                    help="one HIP stream per node update (overlap)")
     p.add_argument("--fixed-iters", type=int, default=16,
                    help="poll-free Weiszfeld iterations (0 = tol/poll mode)")
+    p.add_argument("--warm-start", action="store_true",
+                   help="carry each node's center across rounds (the "
+                        "fixed point barely moves per gossip round)")
     p.add_argument("--graph", action="store_true",
                    help="capture the whole gossip round in one hipGraph")
     p.add_argument("--device", default="cuda" if torch.cuda.is_available() else "cpu")
@@ -59,12 +62,20 @@ def main() -> None:
     )
 
     fixed = args.fixed_iters if args.fixed_iters > 0 else None
+    warm: dict = {}
 
-    def geomed(mixed):
-        if fixed is not None:
+    def geomed(mixed, i):
+        if args.warm_start and i in warm:
+            out = D.geometric_median(mixed, fixed_iters=fixed or 4,
+                                     init_z=warm[i])
+        elif fixed is not None:
             # poll-free: no host sync, so per-node streams truly overlap
-            return D.geometric_median(mixed, fixed_iters=fixed)
-        return D.geometric_median(mixed, tol=1e-6, max_iter=32)
+            out = D.geometric_median(mixed, fixed_iters=fixed)
+        else:
+            out = D.geometric_median(mixed, tol=1e-6, max_iter=32)
+        if args.warm_start:
+            warm[i] = out.float()
+        return out
 
     # index tensors hoisted to device once: a python-list fancy-index does
     # a pageable H2D copy per call, which is also capture-UNSAFE (the
@@ -80,7 +91,7 @@ def main() -> None:
             for i in range(args.nodes):
                 X = theta[groups[i]]  # (1+2k, d) gather
                 mixed = D.nnm(X, args.f)
-                new[i] = geomed(mixed)
+                new[i] = geomed(mixed, i)
         else:
             # node updates are independent: one HIP stream each, so the
             # launch/sync-bound small-kernel chains overlap on the chip
@@ -91,7 +102,7 @@ def main() -> None:
                 with torch.cuda.stream(streams[i]):
                     X = theta[groups[i]]
                     mixed = D.nnm(X, args.f)
-                    new[i] = geomed(mixed)
+                    new[i] = geomed(mixed, i)
             for s in streams:
                 cur.wait_stream(s)
         theta.copy_(new)

@@ -181,14 +181,31 @@ def geometric_median(
     eps: float = 1e-12,
     init: str = "median",
     fixed_iters: Optional[int] = None,
+    init_z: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Weiszfeld fixed point. ``fixed_iters`` runs exactly that many
     iterations with NO convergence polls — fully async (no host sync), so
     per-node streams overlap and the whole call is hipGraph-capture-safe
-    (the polls serialized config-4 gossip in round 1)."""
+    (the polls serialized config-4 gossip in round 1). ``init_z`` warm-
+    starts from a caller-provided center (e.g. the previous gossip
+    round's output — the fixed point barely moves between rounds, so a
+    handful of iterations replaces a cold median init + long descent)."""
     if not _gpu(X) or X.shape[0] > 1024:
         # extension's weiszfeld_iter TORCH_CHECKs n <= 1024; the torch
         # functional path runs fine on-device for larger n
+        if init_z is not None:
+            Xf = X.float()
+            z = init_z.detach().to(device=X.device, dtype=torch.float32)
+            iters = int(fixed_iters) if fixed_iters is not None else max_iter
+            for _ in range(iters):
+                dist = (Xf - z[None, :]).norm(dim=1).clamp_(min=eps)
+                w = 1.0 / dist
+                z_new = (w[:, None] * Xf).sum(dim=0) / w.sum()
+                if fixed_iters is None and float((z_new - z).norm()) <= tol:
+                    z = z_new
+                    break
+                z = z_new
+            return z.to(X.dtype)
         if fixed_iters is not None:
             return F.geometric_median(
                 X, tol=0.0, max_iter=int(fixed_iters), eps=eps, init=init
@@ -196,7 +213,10 @@ def geometric_median(
         return F.geometric_median(X, tol=tol, max_iter=max_iter, eps=eps, init=init)
     ext = _hip.require()
     Xc = X.contiguous()
-    z = (median(Xc) if init == "median" else Xc.float().mean(dim=0)).float()
+    if init_z is not None:
+        z = init_z.detach().to(device=X.device, dtype=torch.float32)
+    else:
+        z = (median(Xc) if init == "median" else Xc.float().mean(dim=0)).float()
     shift = torch.zeros((), device=X.device, dtype=torch.float32)
     if fixed_iters is not None:
         for _ in range(int(fixed_iters)):

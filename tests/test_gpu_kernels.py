@@ -686,3 +686,17 @@ class TestAttackKernels:
         out = D.gaussian_attack(t, 0.0, 1.0, seed=3)
         assert out.dtype == torch.bfloat16
         assert abs(float(out.float().mean())) < 0.05
+
+
+def test_geomed_warm_start_init_z():
+    """init_z warm start: starting at the converged center stays there."""
+    X = _rand(12, 4096, torch.bfloat16, seed=91)
+    z0 = D.geometric_median(X, tol=1e-8, max_iter=300)
+    z1 = D.geometric_median(X, fixed_iters=3, init_z=z0.float())
+    assert torch.allclose(z0.float(), z1.float(), atol=1e-2, rtol=1e-2)
+    # and from an arbitrary center it still converges toward the oracle
+    z2 = D.geometric_median(
+        X, fixed_iters=100, init_z=torch.zeros(4096, device="cuda")
+    )
+    ref = F.geometric_median(X.float().cpu(), tol=1e-8, max_iter=300)
+    assert (z2.float().cpu() - ref).norm() < 0.05 * max(1.0, float(ref.norm()))
