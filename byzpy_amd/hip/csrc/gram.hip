@@ -742,6 +742,92 @@ gram_f32_small32_kernel(const float* __restrict__ X, float* __restrict__ G,
     atomicAdd(&G[idx], cbuf[idx]);
 }
 
+// f32, 32 < n <= 64: SYMMETRIC 3-tile wave-slab. Two 32-row fragments
+// (a0 = rows 0..31, a1 = rows 32..63) feed THREE v_mfma_f32_32x32x2_f32
+// per K-step — C00 = a0*a0^T, C01 = a0*a1^T, C11 = a1*a1^T — and C10 is
+// C01 transposed by Gram symmetry, so the matrix pipe does 3/4 of the
+// naive work and every element is loaded once. (The LDS 64-tile f32
+// kernel measured 2.4 TB/s here: it is f32-MFMA-issue-bound at 4 full
+// tiles; this variant's ceiling is ~6 TB/s.)
+__global__ void __launch_bounds__(SMALL_WAVES * 64, 4)
+gram_f32_small64_kernel(const float* __restrict__ X, float* __restrict__ G,
+                        int n, long d, long kpw) {
+  __shared__ float cbuf[64 * 64];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  for (int i = t; i < 64 * 64; i += SMALL_WAVES * 64) cbuf[i] = 0.0f;
+
+  const long W = (long)blockIdx.x * SMALL_WAVES + wave;
+  const long base = W * kpw;
+
+  const int frow = lane & 31;
+  const int i0 = min(frow, n - 1);
+  const int i1 = min(32 + frow, n - 1);
+  const long lane_k = lane >> 5;  // k = k0 + lane_k, K-step 2
+  const float* src0 = X + (long)i0 * d + base + lane_k;
+  const float* src1 = X + (long)i1 * d + base + lane_k;
+
+  f32x16 acc00 = {};
+  f32x16 acc01 = {};
+  f32x16 acc11 = {};
+  if (base + kpw <= d) {
+    long k = 0;
+    for (; k + 16 <= kpw; k += 16) {
+      float a0[8], a1[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        a0[u] = src0[k + u * 2];
+        a1[u] = src1[k + u * 2];
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0[u], a0[u], acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0[u], a1[u], acc01, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1[u], a1[u], acc11, 0, 0, 0);
+      }
+    }
+    for (; k + 2 <= kpw; k += 2) {
+      const float a0 = src0[k];
+      const float a1 = src1[k];
+      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, a0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, a1, acc01, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, a1, acc11, 0, 0, 0);
+    }
+  } else if (base < d) {
+    const long hi = d - base;
+    for (long k = 0; k < kpw && k < hi; k += 2) {
+      const long kk = k + lane_k;
+      const float a0 = (kk < hi) ? src0[k] : 0.0f;
+      const float a1 = (kk < hi) ? src1[k] : 0.0f;
+      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, a0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, a1, acc01, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, a1, acc11, 0, 0, 0);
+    }
+  }
+
+  __syncthreads();
+  // C/D map for 32x32: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int cc = lane & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rr = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    // C00 -> (rr, cc); C01 -> (rr, 32+cc) and its mirror; C11 -> (+32, +32)
+    if (rr < n && cc < n) atomicAdd(&cbuf[rr * 64 + cc], acc00[r]);
+    if (rr < n && 32 + cc < n) {
+      atomicAdd(&cbuf[rr * 64 + 32 + cc], acc01[r]);
+      atomicAdd(&cbuf[(32 + cc) * 64 + rr], acc01[r]);
+    }
+    if (32 + rr < n && 32 + cc < n)
+      atomicAdd(&cbuf[(32 + rr) * 64 + 32 + cc], acc11[r]);
+  }
+  __syncthreads();
+  for (int idx = t; idx < n * n; idx += SMALL_WAVES * 64) {
+    const int r2 = idx / n, c2 = idx % n;
+    atomicAdd(&G[idx], cbuf[r2 * 64 + c2]);
+  }
+}
+
 // kpw chosen so each wave has >= ~2048 elements and the grid still fills
 // the chip (8 XCDs x 32 CUs want >= ~512 blocks when d allows).
 inline void small_geometry(long d, int align, long& kpw, long& blocks) {
@@ -816,16 +902,20 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
 
 void launch_gram_f32(const float* X, float* G, int n, long d,
                      hipStream_t stream) {
-  if (n <= 32 && d >= 64) {
+  if (n <= 64 && d >= 64) {
     long kpw, blocks;
     if (n <= 16) {
       const int P = 16 / n;
       small_geometry(d, P * 4, kpw, blocks);
       hipLaunchKernelGGL(gram_f32_small16_kernel, dim3((unsigned)blocks),
                          dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw, P);
-    } else {
+    } else if (n <= 32) {
       small_geometry(d, 16, kpw, blocks);
       hipLaunchKernelGGL(gram_f32_small32_kernel, dim3((unsigned)blocks),
+                         dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw);
+    } else {
+      small_geometry(d, 16, kpw, blocks);
+      hipLaunchKernelGGL(gram_f32_small64_kernel, dim3((unsigned)blocks),
                          dim3(SMALL_WAVES * 64), 0, stream, X, G, n, d, kpw);
     }
     return;

@@ -700,3 +700,31 @@ def test_geomed_warm_start_init_z():
     )
     ref = F.geometric_median(X.float().cpu(), tol=1e-8, max_iter=300)
     assert (z2.float().cpu() - ref).norm() < 0.05 * max(1.0, float(ref.norm()))
+
+
+class TestGramF32Sym64:
+    """Symmetric 3-tile f32 Gram for 32 < n <= 64."""
+
+    @pytest.mark.parametrize("n", [33, 40, 48, 63, 64])
+    def test_parity(self, n):
+        X = _rand(n, 8192, torch.float32, seed=n)
+        G = D.gram(X)
+        ref = X @ X.T
+        assert torch.allclose(G, ref, atol=1e-2, rtol=1e-4), f"n={n}"
+
+    def test_asymmetric_structure(self):
+        n, d = 50, 2048
+        X = torch.zeros(n, d)
+        for i in range(n):
+            X[i, (i * 17) % d] = i + 1.0
+            X[i, (i * 5 + 3) % d] = -0.25 * i
+        X = X.cuda()
+        G = D.gram(X)
+        ref = X @ X.T
+        assert torch.allclose(G, ref, atol=1e-3)
+
+    def test_tail_d(self):
+        X = _rand(64, 100001, torch.float32, seed=3)
+        G = D.gram(X)
+        ref = X @ X.T
+        assert torch.allclose(G, ref, atol=0.05, rtol=1e-4)
