@@ -45,3 +45,9 @@ def test_rccl_ps_example_single_process():
     r = _run(["examples/rccl_multigpu_ps.py"])
     assert r.returncode == 0, r.stderr[-800:]
     assert "RCCL PS" in r.stdout
+
+
+def test_gossip_serving_example():
+    r = _run(["examples/gossip_serving.py", "--rounds", "2", "--d", "20000"])
+    assert r.returncode == 0, r.stderr
+    assert "gossip serving" in r.stdout
