@@ -728,3 +728,39 @@ class TestGramF32Sym64:
         G = D.gram(X)
         ref = X @ X.T
         assert torch.allclose(G, ref, atol=0.05, rtol=1e-4)
+
+
+class TestRadixCapEdges:
+    """Boundary shapes for the radix engine: just past the LDS cap and at
+    the 65535-row ceiling."""
+
+    def test_n_513(self):
+        X = _rand(513, 2000, torch.bfloat16, seed=513)
+        assert torch.equal(
+            D.median(X).cpu(), F.median(X.float().cpu()).bfloat16()
+        )
+        assert torch.allclose(
+            D.trimmed_mean(X, 100).float().cpu(),
+            F.trimmed_mean(X.float().cpu(), 100),
+            atol=2e-2, rtol=2e-2,
+        )
+
+    def test_n_cap_65535(self):
+        X = _rand(65535, 257, torch.bfloat16, seed=1)
+        assert torch.equal(
+            D.median(X).cpu(), F.median(X.float().cpu()).bfloat16()
+        )
+
+    def test_extreme_f(self):
+        n = 401
+        X = _rand(n, 2048, torch.float32, seed=7)
+        f = (n - 1) // 2  # keeps exactly one coordinate
+        out = D.trimmed_mean(X, f)
+        ref = F.trimmed_mean(X.cpu(), f)
+        assert torch.allclose(out.cpu(), ref, atol=1e-4)
+
+    def test_above_cap_falls_back_to_torch(self):
+        X = _rand(65600, 64, torch.bfloat16, seed=2)
+        out = D.median(X)  # dispatch cap -> F.median on device
+        ref = F.median(X.float().cpu()).bfloat16()
+        assert torch.equal(out.cpu(), ref)
