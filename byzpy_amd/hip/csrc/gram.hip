@@ -324,6 +324,203 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
+// FUSED Gram + coordinate-wise median (bench config 2's two ops share
+// their input): the DIAG single-tile kernel above already stages every
+// (64 row x 256 k) chunk of X through LDS for the MFMAs — each chunk is
+// 256 COMPLETE columns, so the block's last two waves sort the chunk's
+// 128 column pairs straight out of LDS and emit medians while staging
+// dominates. X is read from HBM ONCE for BOTH aggregates (separately:
+// 16 GB twice = 6.95 ms at 64x125M; the sort VALU hides entirely under
+// the ~2.6 us/chunk staging budget).
+//
+// Register economics force the sort into two 32-row halves: at
+// launch_bounds(1024, 2) the VGPR budget is 64/lane, so the 64-row
+// column pair is sorted as two packed 32-value bitonic networks
+// (32 u32 regs), written back IN PLACE into the chunk's LDS (dead after
+// the MFMA phase), and the median comes from a two-pointer merge walk
+// of the sorted halves. Pads (rows >= n) get key 0xFFFF at read time.
+// ---------------------------------------------------------------------------
+#include "pksort.h"
+
+template <int BK = 256>
+__global__ void __launch_bounds__(1024, 2)
+gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
+                        float* __restrict__ G,
+                        __hip_bfloat16* __restrict__ med, int n, long d,
+                        long k_per_block) {
+  constexpr int SLOTS = BK / 8;
+  constexpr int SUB = SLOTS / 16;
+  constexpr int CHUNK_BYTES = TILE * BK * 2;
+  __shared__ char smem[2 * CHUNK_BYTES];
+  char* const bufA0 = smem;
+  char* const bufA1 = smem + CHUNK_BYTES;
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const long k_lo = (long)blockIdx.x * k_per_block;
+  const long k_hi = min(d, k_lo + k_per_block);
+  const long nchunks = (k_hi - k_lo + BK - 1) / BK;
+
+  const int st_row = t >> 4;
+  const int a_rows = min(TILE, n);
+  int st_slot[SUB], src_slot[SUB];
+#pragma unroll
+  for (int u = 0; u < SUB; ++u) {
+    st_slot[u] = (t & 15) + u * 16;
+    src_slot[u] = st_slot[u] ^ (st_row & 15);
+  }
+  const __hip_bfloat16* a_row_src = X + (long)min(st_row, a_rows - 1) * d;
+
+  bf16x8 ra[SUB];
+  auto stage_load = [&](long c) {
+    const long k0 = k_lo + c * BK;
+    const bool full = (k0 + BK <= k_hi);
+#pragma unroll
+    for (int u = 0; u < SUB; ++u) {
+      const long off = k0 + (long)src_slot[u] * 8;
+      if (full && a_rows == TILE) {
+        ra[u] = *reinterpret_cast<const bf16x8*>(a_row_src + off);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ra[u][j] = (st_row < a_rows && off + j < k_hi)
+                         ? *reinterpret_cast<const __bf16*>(a_row_src + off + j)
+                         : (__bf16)0.0f;
+      }
+    }
+  };
+  auto stage_write = [&](int which) {
+    char* a = which ? bufA1 : bufA0;
+#pragma unroll
+    for (int u = 0; u < SUB; ++u)
+      *reinterpret_cast<bf16x8*>(a + st_row * (SLOTS * 16) + st_slot[u] * 16) =
+          ra[u];
+  };
+
+  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  const int rowA = wr * 16 + (lane & 15);
+  const int rowB = wc * 16 + (lane & 15);
+  const int grp = lane >> 4;
+  // sort-wave geometry: threads 896..1023 own column pair (2s, 2s+1)
+  const int s_id = t - 896;
+  const int col0 = 2 * s_id;          // column within the chunk
+  const int nv = pk_vecify(n);
+  const int plo = pk_vecify((n - 1) >> 1), phi = pk_vecify(n >> 1);
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  for (long c = 0; c < nchunks; ++c) {
+    if (c + 1 < nchunks) stage_load(c + 1);
+    char* A = (c & 1) ? bufA1 : bufA0;
+#pragma unroll
+    for (int step = 0; step < BK / 32; ++step) {
+      const int q = step * 4 + grp;
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          A + rowA * (SLOTS * 16) + ((q ^ (rowA & 15)) * 16));
+      const bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          A + rowB * (SLOTS * 16) + ((q ^ (rowB & 15)) * 16));
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    // median epilogue over the SAME chunk (buffer is dead to the MFMA
+    // waves now; it is overwritten no earlier than chunk c+2's
+    // stage_write, two barriers away)
+    if (s_id >= 0 && k_lo + c * BK + col0 < k_hi) {
+      // LDS address of the packed column pair in row r
+      auto addr = [&](int r) {
+        return A + r * (SLOTS * 16) + (((col0 >> 3) ^ (r & 15)) * 16) +
+               (col0 & 7) * 2;
+      };
+      pk_u32 v[32];
+      // half 1: rows 0..31 -> sorted back in place
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        const pk_u32 w = *reinterpret_cast<const pk_u32*>(addr(r));
+        v[r] = (r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
+      }
+      bitonic_sort_pk<32>(v);
+#pragma unroll
+      for (int r = 0; r < 32; ++r)
+        *reinterpret_cast<pk_u32*>(addr(r)) = v[r];
+      // half 2: rows 32..63
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        const pk_u32 w = *reinterpret_cast<const pk_u32*>(addr(32 + r));
+        v[r] = (32 + r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
+      }
+      bitonic_sort_pk<32>(v);
+#pragma unroll
+      for (int r = 0; r < 32; ++r)
+        *reinterpret_cast<pk_u32*>(addr(32 + r)) = v[r];
+      // two-pointer merge walk to ranks (n-1)/2 and n/2, both halves of
+      // the packed pair at once (per-half u16 compares)
+      pk_u32 m0 = 0, m1 = 0;  // packed keys at rank plo / phi
+      {
+        int li0 = 0, hi0 = 0, li1 = 0, hi1 = 0;
+        pk_u32 lo_k0 = 0, lo_k1 = 0;
+        for (int step2 = 0; step2 <= phi; ++step2) {
+          // clamped reads: once a pointer is exhausted its value is
+          // never selected, but the load must stay in bounds
+          const pk_u32 lw = *reinterpret_cast<const pk_u32*>(addr(min(li0, 31)));
+          const pk_u32 hw =
+              *reinterpret_cast<const pk_u32*>(addr(32 + min(hi0, 31)));
+          const unsigned short l0 = (unsigned short)lw, h0 = (unsigned short)hw;
+          const pk_u32 lw1 = *reinterpret_cast<const pk_u32*>(addr(min(li1, 31)));
+          const pk_u32 hw1 =
+              *reinterpret_cast<const pk_u32*>(addr(32 + min(hi1, 31)));
+          const unsigned short l1 = (unsigned short)(lw1 >> 16);
+          const unsigned short h1 = (unsigned short)(hw1 >> 16);
+          unsigned short cur0, cur1;
+          if (li0 < 32 && (hi0 >= 32 || l0 <= h0)) { cur0 = l0; ++li0; }
+          else { cur0 = h0; ++hi0; }
+          if (li1 < 32 && (hi1 >= 32 || l1 <= h1)) { cur1 = l1; ++li1; }
+          else { cur1 = h1; ++hi1; }
+          if (step2 == plo) { lo_k0 = cur0; lo_k1 = cur1; }
+          if (step2 == phi) {
+            m0 = lo_k0 | ((pk_u32)cur0 << 16);
+            m1 = lo_k1 | ((pk_u32)cur1 << 16);
+          }
+        }
+      }
+      // m0/m1: low 16 = rank-lo key, high 16 = rank-hi key
+      const long gcol = k_lo + c * BK + col0;
+      const float med0 =
+          0.5f * (pk_key_to_float(m0 & 0xFFFFu) + pk_key_to_float(m0 >> 16));
+      const float med1 =
+          0.5f * (pk_key_to_float(m1 & 0xFFFFu) + pk_key_to_float(m1 >> 16));
+      union { unsigned short s2[2]; pk_u32 w2; } o;
+      union { unsigned short s3; __hip_bfloat16 h3; } c0, c1;
+      c0.h3 = __float2bfloat16(med0);
+      c1.h3 = __float2bfloat16(med1);
+      o.s2[0] = c0.s3;
+      o.s2[1] = c1.s3;
+      if (gcol + 1 < k_hi) {
+        *reinterpret_cast<pk_u32*>(
+            reinterpret_cast<unsigned short*>(med) + gcol) = o.w2;
+      } else {
+        reinterpret_cast<unsigned short*>(med)[gcol] = o.s2[0];
+      }
+    }
+    __syncthreads();
+    if (c + 1 < nchunks) {
+      stage_write((c + 1) & 1);
+      __syncthreads();
+    }
+  }
+
+  const int out_col = wc * 16 + (lane & 15);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int out_row = wr * 16 + (lane >> 4) * 4 + r;
+    if (out_row < n && out_col < n)
+      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // LDS-staged f32 Gram (fast path for d % 4 == 0): same structure as the
 // bf16 v3 kernel with v_mfma_f32_16x16x4_f32 fragments (exact f32 at the
 // vector rate, guide §3). Chunk = 64 rows x 64 k f32 = 16 KB. The MFMA
@@ -905,6 +1102,15 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
     hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
                        stream, X, G, n, d, kpb);
   }
+}
+
+void launch_gram_median_bf16(const __hip_bfloat16* X, float* G,
+                             __hip_bfloat16* med, int n, long d,
+                             hipStream_t stream) {
+  int splitk; long kpb;
+  split_geometry(n, d, splitk, kpb);  // kpb is 256-aligned by construction
+  hipLaunchKernelGGL((gram_median_bf16_kernel<256>), dim3(splitk),
+                     dim3(1024), 0, stream, X, G, med, n, d, kpb);
 }
 
 void launch_gram_f32(const float* X, float* G, int n, long d,

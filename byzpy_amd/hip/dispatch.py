@@ -62,6 +62,18 @@ def gram(X: torch.Tensor) -> torch.Tensor:
     return Xf @ Xf.T
 
 
+def median_and_gram(X: torch.Tensor):
+    """Both of the flagship aggregates from ONE HBM pass (the fused
+    kernel's sort waves ride the Gram's LDS staging). Returns
+    (median, gram). Falls back to the two separate kernels off the
+    fused kernel's shape envelope."""
+    n, d = X.shape
+    if _gpu(X) and X.dtype == torch.bfloat16 and n <= 64 and d >= 1024:
+        G, med = _hip.require().gram_median(X.contiguous())
+        return med, G
+    return median(X), gram(X)
+
+
 def pairwise_sq_dists(X: torch.Tensor) -> torch.Tensor:
     if _gpu(X):
         G = gram(X)

@@ -42,6 +42,30 @@ def _global_gram(X_local: torch.Tensor) -> torch.Tensor:
     return all_reduce_(G)
 
 
+def median_and_multi_krum(X_local: torch.Tensor, f: int, q: int):
+    """Both flagship aggregates from ONE pass over the shard: the fused
+    gram+median kernel computes the local median shard and the partial
+    Gram together; one (n, n) all-reduce then yields global Krum
+    selection identical to multi_krum()."""
+    from byzpy_amd.hip import dispatch as _D
+
+    n = X_local.shape[0]
+    med, G = _D.median_and_gram(X_local)
+    G = all_reduce_(G)
+    if X_local.is_cuda and n <= 512:
+        from byzpy_amd import hip as _h
+
+        winners = _h.require().krum_select(G, int(f), int(q))
+        return med, D.mean_rows(X_local, winners)
+    norms = torch.diagonal(G)
+    D2 = (norms[:, None] + norms[None, :] - 2.0 * G).clamp_(min=0.0)
+    D2 = D2 + torch.diag(torch.full((n,), float("inf"), device=X_local.device))
+    k = n - f - 1
+    scores = torch.topk(D2, k=k, dim=1, largest=False).values.sum(dim=1)
+    winners = torch.topk(scores, k=q, largest=False).indices
+    return med, D.mean_rows(X_local, winners.to(torch.int32))
+
+
 def multi_krum(X_local: torch.Tensor, f: int, q: int) -> torch.Tensor:
     n = X_local.shape[0]
     G = _global_gram(X_local)

@@ -764,3 +764,41 @@ class TestRadixCapEdges:
         out = D.median(X)  # dispatch cap -> F.median on device
         ref = F.median(X.float().cpu()).bfloat16()
         assert torch.equal(out.cpu(), ref)
+
+
+class TestFusedGramMedian:
+    """One-pass gram+median (the flagship bench fusion)."""
+
+    @pytest.mark.parametrize("n", [8, 16, 33, 48, 64])
+    @pytest.mark.parametrize("d", [2048, 4097, 100000])
+    def test_parity(self, n, d):
+        X = _rand(n, d, torch.bfloat16, seed=n + d)
+        med, G = D.median_and_gram(X)
+        ref_med = F.median(X.float().cpu())
+        Xf = X.float()
+        ref_G = Xf @ Xf.T
+        assert torch.allclose(med.float().cpu(), ref_med, atol=5e-2, rtol=5e-2)
+        assert (G - ref_G).abs().max().item() < 0.3 * max(
+            1.0, float(ref_G.abs().max()) / 100
+        )
+
+    def test_median_exact_vs_separate_kernel(self):
+        # same keys, same averaging -> fused median must EQUAL D.median
+        X = _rand(64, 250000, torch.bfloat16, seed=9)
+        med, _ = D.median_and_gram(X)
+        assert torch.equal(med, D.median(X))
+
+    def test_with_inf_rows(self):
+        X = _rand(64, 8192, torch.bfloat16, seed=11)
+        X[3] = float("inf")
+        X[5] = float("-inf")
+        med, G = D.median_and_gram(X)
+        assert torch.equal(med, D.median(X))
+
+    def test_sharded_fused_matches_separate(self):
+        from byzpy_amd.parallel import sharded
+
+        X = _rand(64, 100000, torch.bfloat16, seed=13)
+        med, krum_out = sharded.median_and_multi_krum(X, 16, 12)
+        assert torch.equal(med, sharded.median(X))
+        assert torch.equal(krum_out, sharded.multi_krum(X, 16, 12))
