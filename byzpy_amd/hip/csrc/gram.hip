@@ -425,11 +425,14 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
           A + rowB * (SLOTS * 16) + ((q ^ (rowB & 15)) * 16));
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
-    // median epilogue over the SAME chunk (buffer is dead to the MFMA
-    // waves now; it is overwritten no earlier than chunk c+2's
-    // stage_write, two barriers away)
+    // barrier FIRST: the sort mutates buf c in place, so every wave's
+    // MFMA reads must be complete. The sort then runs concurrently with
+    // the other waves' stage_write into the OTHER buffer (sort waves do
+    // their stage_write slice first, which also kills the prefetch
+    // registers before the 32-register sort arrays go live).
+    __syncthreads();
+    if (c + 1 < nchunks) stage_write((c + 1) & 1);
     if (s_id >= 0 && k_lo + c * BK + col0 < k_hi) {
-      // LDS address of the packed column pair in row r
       auto addr = [&](int r) {
         return A + r * (SLOTS * 16) + (((col0 >> 3) ^ (r & 15)) * 16) +
                (col0 & 7) * 2;
@@ -455,15 +458,13 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
 #pragma unroll
       for (int r = 0; r < 32; ++r)
         *reinterpret_cast<pk_u32*>(addr(32 + r)) = v[r];
-      // two-pointer merge walk to ranks (n-1)/2 and n/2, both halves of
-      // the packed pair at once (per-half u16 compares)
-      pk_u32 m0 = 0, m1 = 0;  // packed keys at rank plo / phi
+      // two-pointer merge walk to ranks (n-1)/2 and n/2, both packed
+      // columns at once (per-half u16 compares)
+      pk_u32 m0 = 0, m1 = 0;
       {
         int li0 = 0, hi0 = 0, li1 = 0, hi1 = 0;
         pk_u32 lo_k0 = 0, lo_k1 = 0;
         for (int step2 = 0; step2 <= phi; ++step2) {
-          // clamped reads: once a pointer is exhausted its value is
-          // never selected, but the load must stay in bounds
           const pk_u32 lw = *reinterpret_cast<const pk_u32*>(addr(min(li0, 31)));
           const pk_u32 hw =
               *reinterpret_cast<const pk_u32*>(addr(32 + min(hi0, 31)));
@@ -485,7 +486,6 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
           }
         }
       }
-      // m0/m1: low 16 = rank-lo key, high 16 = rank-hi key
       const long gcol = k_lo + c * BK + col0;
       const float med0 =
           0.5f * (pk_key_to_float(m0 & 0xFFFFu) + pk_key_to_float(m0 >> 16));
@@ -505,10 +505,6 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
       }
     }
     __syncthreads();
-    if (c + 1 < nchunks) {
-      stage_write((c + 1) & 1);
-      __syncthreads();
-    }
   }
 
   const int out_col = wc * 16 + (lane & 15);
