@@ -342,8 +342,11 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
 // ---------------------------------------------------------------------------
 #include "pksort.h"
 
+// NOTE: on AMD the second __launch_bounds__ argument is MIN WAVES PER
+// EU: 8 forces the 64-VGPR budget that lets 2 blocks (32 waves)
+// co-reside like the unfused gram kernel
 template <int BK = 256>
-__global__ void __launch_bounds__(1024, 2)
+__global__ void __launch_bounds__(1024, 8)
 gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
                         float* __restrict__ G,
                         __hip_bfloat16* __restrict__ med, int n, long d,
@@ -404,9 +407,13 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
   const int rowA = wr * 16 + (lane & 15);
   const int rowB = wc * 16 + (lane & 15);
   const int grp = lane >> 4;
-  // sort-wave geometry: threads 896..1023 own column pair (2s, 2s+1)
-  const int s_id = t - 896;
-  const int col0 = 2 * s_id;          // column within the chunk
+  // sort geometry: threads 768..1023 (4 waves); adjacent thread PAIRS
+  // share one column pair — the even thread sorts rows 0..31 as two
+  // 16-row runs, the odd thread rows 32..63, and the even thread merges
+  // all four runs (same wave -> in-order DS visibility, no barrier)
+  const int s_id = (t - 768) >> 1;     // column-pair index 0..127
+  const int s_half = (t - 768) & 1;    // which 32-row half this thread sorts
+  const int col0 = 2 * s_id;           // column within the chunk
   const int nv = pk_vecify(n);
   const int plo = pk_vecify((n - 1) >> 1), phi = pk_vecify(n >> 1);
 
@@ -432,76 +439,92 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
     // registers before the 32-register sort arrays go live).
     __syncthreads();
     if (c + 1 < nchunks) stage_write((c + 1) & 1);
-    if (s_id >= 0 && k_lo + c * BK + col0 < k_hi) {
-      auto addr = [&](int r) {
-        return A + r * (SLOTS * 16) + (((col0 >> 3) ^ (r & 15)) * 16) +
-               (col0 & 7) * 2;
-      };
-      pk_u32 v[32];
-      // half 1: rows 0..31 -> sorted back in place
+    if (t >= 768 && k_lo + c * BK + col0 < k_hi) {
+      // The sort reads its column pair from GLOBAL memory: the chunk was
+      // staged one phase ago, so these lines are L2-resident — no HBM
+      // re-read, and the addresses are simple strided u32s that pipeline
+      // (reading back through the XOR-swizzled LDS image cost a
+      // ~60-cycle dependent chain per element and measured 5x slower
+      // than the unfused pair). The dead chunk buffer provides each
+      // sort thread a private LINEAR 256 B strip for the merge phase.
+      const long gcol = k_lo + c * BK + col0;
+      const unsigned short* xc =
+          reinterpret_cast<const unsigned short*>(X) + gcol;
+      char* strip = A + s_id * 256;
+      // two sorted 16-row runs per thread (v[16] keeps the register peak
+      // under the 64-VGPR / 2-blocks-per-CU budget)
+      pk_u32 v[16];
+#pragma unroll 1
+      for (int quarter = 0; quarter < 2; ++quarter) {
+        const int r0 = s_half * 32 + quarter * 16;
+#pragma unroll 8
+        for (int r = 0; r < 16; ++r) {
+          const pk_u32 w =
+              *reinterpret_cast<const pk_u32*>(xc + (long)(r0 + r) * d);
+          v[r] = (r0 + r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
+        }
+        bitonic_sort_pk<16>(v);
 #pragma unroll
-      for (int r = 0; r < 32; ++r) {
-        const pk_u32 w = *reinterpret_cast<const pk_u32*>(addr(r));
-        v[r] = (r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
+        for (int r = 0; r < 16; ++r)
+          *reinterpret_cast<pk_u32*>(strip + (r0 + r) * 4) = v[r];
       }
-      bitonic_sort_pk<32>(v);
-#pragma unroll
-      for (int r = 0; r < 32; ++r)
-        *reinterpret_cast<pk_u32*>(addr(r)) = v[r];
-      // half 2: rows 32..63
-#pragma unroll
-      for (int r = 0; r < 32; ++r) {
-        const pk_u32 w = *reinterpret_cast<const pk_u32*>(addr(32 + r));
-        v[r] = (32 + r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
-      }
-      bitonic_sort_pk<32>(v);
-#pragma unroll
-      for (int r = 0; r < 32; ++r)
-        *reinterpret_cast<pk_u32*>(addr(32 + r)) = v[r];
-      // two-pointer merge walk to ranks (n-1)/2 and n/2, both packed
-      // columns at once (per-half u16 compares)
+      // the partner lane's ds_writes are visible in program order within
+      // the wave; pin the ordering against compiler motion
+      __builtin_amdgcn_sched_barrier(0);
       pk_u32 m0 = 0, m1 = 0;
-      {
-        int li0 = 0, hi0 = 0, li1 = 0, hi1 = 0;
-        pk_u32 lo_k0 = 0, lo_k1 = 0;
-        for (int step2 = 0; step2 <= phi; ++step2) {
-          const pk_u32 lw = *reinterpret_cast<const pk_u32*>(addr(min(li0, 31)));
-          const pk_u32 hw =
-              *reinterpret_cast<const pk_u32*>(addr(32 + min(hi0, 31)));
-          const unsigned short l0 = (unsigned short)lw, h0 = (unsigned short)hw;
-          const pk_u32 lw1 = *reinterpret_cast<const pk_u32*>(addr(min(li1, 31)));
-          const pk_u32 hw1 =
-              *reinterpret_cast<const pk_u32*>(addr(32 + min(hi1, 31)));
-          const unsigned short l1 = (unsigned short)(lw1 >> 16);
-          const unsigned short h1 = (unsigned short)(hw1 >> 16);
-          unsigned short cur0, cur1;
-          if (li0 < 32 && (hi0 >= 32 || l0 <= h0)) { cur0 = l0; ++li0; }
-          else { cur0 = h0; ++hi0; }
-          if (li1 < 32 && (hi1 >= 32 || l1 <= h1)) { cur1 = l1; ++li1; }
-          else { cur1 = h1; ++hi1; }
-          if (step2 == plo) { lo_k0 = cur0; lo_k1 = cur1; }
-          if (step2 == phi) {
-            m0 = lo_k0 | ((pk_u32)cur0 << 16);
-            m1 = lo_k1 | ((pk_u32)cur1 << 16);
+      if (s_half == 0) {
+        // 4-pointer merge walk to ranks (n-1)/2 and n/2, one packed
+        // column half at a time
+#pragma unroll 1
+        for (int half = 0; half < 2; ++half) {
+          const int sh = half * 16;
+          int p0q = 0, p1q = 0, p2q = 0, p3q = 0;
+          pk_u32 lo_k = 0, m = 0;
+          for (int step2 = 0; step2 <= phi; ++step2) {
+            const unsigned short c0v = (unsigned short)(
+                *reinterpret_cast<const pk_u32*>(strip + min(p0q, 15) * 4) >> sh);
+            const unsigned short c1v = (unsigned short)(
+                *reinterpret_cast<const pk_u32*>(strip + 64 + min(p1q, 15) * 4) >> sh);
+            const unsigned short c2v = (unsigned short)(
+                *reinterpret_cast<const pk_u32*>(strip + 128 + min(p2q, 15) * 4) >> sh);
+            const unsigned short c3v = (unsigned short)(
+                *reinterpret_cast<const pk_u32*>(strip + 192 + min(p3q, 15) * 4) >> sh);
+            const unsigned int k0v = p0q < 16 ? c0v : 0x10000u;
+            const unsigned int k1v = p1q < 16 ? c1v : 0x10000u;
+            const unsigned int k2v = p2q < 16 ? c2v : 0x10000u;
+            const unsigned int k3v = p3q < 16 ? c3v : 0x10000u;
+            unsigned int cur = k0v;
+            int pick = 0;
+            if (k1v < cur) { cur = k1v; pick = 1; }
+            if (k2v < cur) { cur = k2v; pick = 2; }
+            if (k3v < cur) { cur = k3v; pick = 3; }
+            p0q += (pick == 0);
+            p1q += (pick == 1);
+            p2q += (pick == 2);
+            p3q += (pick == 3);
+            if (step2 == plo) lo_k = (pk_u32)(cur & 0xFFFFu);
+            if (step2 == phi) m = lo_k | ((pk_u32)(cur & 0xFFFFu) << 16);
           }
+          if (half == 0) m0 = m; else m1 = m;
         }
       }
-      const long gcol = k_lo + c * BK + col0;
       const float med0 =
           0.5f * (pk_key_to_float(m0 & 0xFFFFu) + pk_key_to_float(m0 >> 16));
       const float med1 =
           0.5f * (pk_key_to_float(m1 & 0xFFFFu) + pk_key_to_float(m1 >> 16));
-      union { unsigned short s2[2]; pk_u32 w2; } o;
-      union { unsigned short s3; __hip_bfloat16 h3; } c0, c1;
-      c0.h3 = __float2bfloat16(med0);
-      c1.h3 = __float2bfloat16(med1);
-      o.s2[0] = c0.s3;
-      o.s2[1] = c1.s3;
-      if (gcol + 1 < k_hi) {
-        *reinterpret_cast<pk_u32*>(
-            reinterpret_cast<unsigned short*>(med) + gcol) = o.w2;
-      } else {
-        reinterpret_cast<unsigned short*>(med)[gcol] = o.s2[0];
+      if (s_half == 0) {
+        union { unsigned short s2[2]; pk_u32 w2; } o;
+        union { unsigned short s3; __hip_bfloat16 h3; } c0, c1;
+        c0.h3 = __float2bfloat16(med0);
+        c1.h3 = __float2bfloat16(med1);
+        o.s2[0] = c0.s3;
+        o.s2[1] = c1.s3;
+        if (gcol + 1 < k_hi) {
+          *reinterpret_cast<pk_u32*>(
+              reinterpret_cast<unsigned short*>(med) + gcol) = o.w2;
+        } else {
+          reinterpret_cast<unsigned short*>(med)[gcol] = o.s2[0];
+        }
       }
     }
     __syncthreads();
