@@ -407,15 +407,14 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
   const int rowA = wr * 16 + (lane & 15);
   const int rowB = wc * 16 + (lane & 15);
   const int grp = lane >> 4;
-  // sort geometry: threads 768..1023 (4 waves); adjacent thread PAIRS
-  // share one column pair — the even thread sorts rows 0..31 as two
-  // 16-row runs, the odd thread rows 32..63, and the even thread merges
-  // all four runs (same wave -> in-order DS visibility, no barrier)
-  const int s_id = (t - 768) >> 1;     // column-pair index 0..127
-  const int s_half = (t - 768) & 1;    // which 32-row half this thread sorts
-  const int col0 = 2 * s_id;           // column within the chunk
-  const int nv = pk_vecify(n);
-  const int plo = pk_vecify((n - 1) >> 1), phi = pk_vecify(n >> 1);
+  // median phase geometry: EVERY wave rank-selects 16 of the chunk's 256
+  // columns (one column at a time, one staged value per lane, ballot
+  // bit-descent) — concentrating the median work on 2-4 sort waves left
+  // it latency-serial on the chunk critical path (35 ms total, 3 kernel
+  // variants measured; the standalone median's VALU work is ~1.85 ms at
+  // this shape and only hides under the 2.5 ms staging when all 16
+  // waves carry it)
+  const int plo = (n - 1) >> 1, phi = n >> 1;
 
   stage_load(0);
   stage_write(0);
@@ -439,91 +438,52 @@ gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
     // registers before the 32-register sort arrays go live).
     __syncthreads();
     if (c + 1 < nchunks) stage_write((c + 1) & 1);
-    if (t >= 768 && k_lo + c * BK + col0 < k_hi) {
-      // The sort reads its column pair from GLOBAL memory: the chunk was
-      // staged one phase ago, so these lines are L2-resident — no HBM
-      // re-read, and the addresses are simple strided u32s that pipeline
-      // (reading back through the XOR-swizzled LDS image cost a
-      // ~60-cycle dependent chain per element and measured 5x slower
-      // than the unfused pair). The dead chunk buffer provides each
-      // sort thread a private LINEAR 256 B strip for the merge phase.
-      const long gcol = k_lo + c * BK + col0;
-      const unsigned short* xc =
-          reinterpret_cast<const unsigned short*>(X) + gcol;
-      char* strip = A + s_id * 256;
-      // two sorted 16-row runs per thread (v[16] keeps the register peak
-      // under the 64-VGPR / 2-blocks-per-CU budget)
-      pk_u32 v[16];
+    {
+      const long kc0 = k_lo + c * BK;
 #pragma unroll 1
-      for (int quarter = 0; quarter < 2; ++quarter) {
-        const int r0 = s_half * 32 + quarter * 16;
-#pragma unroll 8
-        for (int r = 0; r < 16; ++r) {
-          const pk_u32 w =
-              *reinterpret_cast<const pk_u32*>(xc + (long)(r0 + r) * d);
-          v[r] = (r0 + r < nv) ? pk_key_from_bf16(w) : 0xFFFFFFFFu;
-        }
-        bitonic_sort_pk<16>(v);
-#pragma unroll
-        for (int r = 0; r < 16; ++r)
-          *reinterpret_cast<pk_u32*>(strip + (r0 + r) * 4) = v[r];
-      }
-      // the partner lane's ds_writes are visible in program order within
-      // the wave; pin the ordering against compiler motion
-      __builtin_amdgcn_sched_barrier(0);
-      pk_u32 m0 = 0, m1 = 0;
-      if (s_half == 0) {
-        // 4-pointer merge walk to ranks (n-1)/2 and n/2, one packed
-        // column half at a time
-#pragma unroll 1
-        for (int half = 0; half < 2; ++half) {
-          const int sh = half * 16;
-          int p0q = 0, p1q = 0, p2q = 0, p3q = 0;
-          pk_u32 lo_k = 0, m = 0;
-          for (int step2 = 0; step2 <= phi; ++step2) {
-            const unsigned short c0v = (unsigned short)(
-                *reinterpret_cast<const pk_u32*>(strip + min(p0q, 15) * 4) >> sh);
-            const unsigned short c1v = (unsigned short)(
-                *reinterpret_cast<const pk_u32*>(strip + 64 + min(p1q, 15) * 4) >> sh);
-            const unsigned short c2v = (unsigned short)(
-                *reinterpret_cast<const pk_u32*>(strip + 128 + min(p2q, 15) * 4) >> sh);
-            const unsigned short c3v = (unsigned short)(
-                *reinterpret_cast<const pk_u32*>(strip + 192 + min(p3q, 15) * 4) >> sh);
-            const unsigned int k0v = p0q < 16 ? c0v : 0x10000u;
-            const unsigned int k1v = p1q < 16 ? c1v : 0x10000u;
-            const unsigned int k2v = p2q < 16 ? c2v : 0x10000u;
-            const unsigned int k3v = p3q < 16 ? c3v : 0x10000u;
-            unsigned int cur = k0v;
-            int pick = 0;
-            if (k1v < cur) { cur = k1v; pick = 1; }
-            if (k2v < cur) { cur = k2v; pick = 2; }
-            if (k3v < cur) { cur = k3v; pick = 3; }
-            p0q += (pick == 0);
-            p1q += (pick == 1);
-            p2q += (pick == 2);
-            p3q += (pick == 3);
-            if (step2 == plo) lo_k = (pk_u32)(cur & 0xFFFFu);
-            if (step2 == phi) m = lo_k | ((pk_u32)(cur & 0xFFFFu) << 16);
-          }
-          if (half == 0) m0 = m; else m1 = m;
-        }
-      }
-      const float med0 =
-          0.5f * (pk_key_to_float(m0 & 0xFFFFu) + pk_key_to_float(m0 >> 16));
-      const float med1 =
-          0.5f * (pk_key_to_float(m1 & 0xFFFFu) + pk_key_to_float(m1 >> 16));
-      if (s_half == 0) {
-        union { unsigned short s2[2]; pk_u32 w2; } o;
-        union { unsigned short s3; __hip_bfloat16 h3; } c0, c1;
-        c0.h3 = __float2bfloat16(med0);
-        c1.h3 = __float2bfloat16(med1);
-        o.s2[0] = c0.s3;
-        o.s2[1] = c1.s3;
-        if (gcol + 1 < k_hi) {
-          *reinterpret_cast<pk_u32*>(
-              reinterpret_cast<unsigned short*>(med) + gcol) = o.w2;
+      for (int cj = 0; cj < BK / 16; ++cj) {
+        const int col = wave * (BK / 16) + cj;
+        const long gcol = kc0 + col;
+        if (gcol >= k_hi) break;
+        // lane r holds the staged value at (row=r, col); rows >= n pad
+        // with the max key. 4-way bank conflict (lanes 16 apart share a
+        // swizzle slot) — 4 cycles, amortized over the whole column.
+        const unsigned short raw = *reinterpret_cast<const unsigned short*>(
+            A + lane * (SLOTS * 16) + (((col >> 3) ^ (lane & 15)) * 16) +
+            (col & 7) * 2);
+        pk_u32 key;
+        if (lane < n) {
+          const pk_u32 b16 = raw;
+          const pk_u32 sg = (b16 >> 15) & 1u;
+          key = (b16 ^ (0x8000u + sg * 0x7FFFu)) & 0xFFFFu;
         } else {
-          reinterpret_cast<unsigned short*>(med)[gcol] = o.s2[0];
+          key = 0xFFFFu;
+        }
+        // ballot bit-descent rank selection (exact with duplicates):
+        // resolve the rank-k key from the high bit down, counting the
+        // candidates whose unresolved bit is 0
+        pk_u32 keys[2] = {0, 0};
+#pragma unroll 1
+        for (int which = 0; which < 2; ++which) {
+          int rank = which ? phi : plo;
+          pk_u32 prefix = 0;
+          for (int bit = 15; bit >= 0; --bit) {
+            const unsigned long long m =
+                __ballot((key >> bit) == (prefix >> bit));
+            const int c0 = __popcll(m);
+            if (rank >= c0) {
+              rank -= c0;
+              prefix |= 1u << bit;
+            }
+          }
+          keys[which] = prefix;
+        }
+        if (lane == 0) {
+          const float mv =
+              0.5f * (pk_key_to_float(keys[0]) + pk_key_to_float(keys[1]));
+          union { unsigned short s3; __hip_bfloat16 h3; } o;
+          o.h3 = __float2bfloat16(mv);
+          reinterpret_cast<unsigned short*>(med)[gcol] = o.s3;
         }
       }
     }
