@@ -37,12 +37,6 @@ def parse_args():
     p.add_argument("--op", choices=["both", "median", "krum"], default="both")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "f32"])
     p.add_argument(
-        "--no-fused",
-        action="store_true",
-        help="run median and Multi-Krum as separate kernels (two HBM "
-        "passes) instead of the fused gram+median kernel",
-    )
-    p.add_argument(
         "--overlap",
         action="store_true",
         help="run the two aggregates concurrently on two HIP streams "
@@ -114,9 +108,7 @@ def main():
             cur.wait_stream(s_med)
             cur.wait_stream(s_krum)
             return outs
-        if args.op == "both" and not args.no_fused:
-            # ONE HBM pass serves both aggregates: the fused kernel's
-            # sort waves compute the median from the Gram's LDS staging
+        if args.op == "both":
             med, krum_out = sharded.median_and_multi_krum(X, args.f, args.q)
             return [med, krum_out]
         if args.op in ("both", "median"):

@@ -69,8 +69,6 @@ void launch_mda_pass1(const float*, const int*, int, int, int, int,
 void launch_mda_pass2(const float*, const int*, int, int, int, int,
                       unsigned int*, int*, int*, hipStream_t);
 void launch_gram_bf16(const __hip_bfloat16*, float*, int, long, hipStream_t);
-void launch_gram_median_bf16(const __hip_bfloat16*, float*, __hip_bfloat16*,
-                             int, long, hipStream_t);
 void launch_gram_f32(const float*, float*, int, long, hipStream_t);
 void launch_krum_select(const float*, int, int, int, int*, float*, hipStream_t);
 
@@ -497,22 +495,6 @@ torch::Tensor mda_search(torch::Tensor D2in, int64_t f) {
   return out;
 }
 
-// Fused Gram + coordinate-wise median (one HBM pass serves both of the
-// flagship bench's aggregates; gram.hip gram_median_bf16_kernel).
-std::vector<torch::Tensor> gram_median(torch::Tensor X) {
-  check_matrix(X);
-  TORCH_CHECK(X.scalar_type() == torch::kBFloat16,
-              "gram_median supports bf16");
-  const int n = (int)X.size(0);
-  const long d = (long)X.size(1);
-  TORCH_CHECK(n >= 1 && n <= 64, "gram_median supports n <= 64");
-  auto G = torch::zeros({n, n}, X.options().dtype(torch::kFloat32));
-  auto med = torch::empty({d}, X.options());
-  launch_gram_median_bf16(bf16_ptr(X), G.data_ptr<float>(),
-                          bf16_ptr_mut(med), n, d, cur_stream());
-  return {G, med};
-}
-
 // Little attack fused: per-column mu + z*sigma in one streaming pass.
 torch::Tensor little_fused(torch::Tensor X, double z) {
   check_matrix(X);
@@ -706,8 +688,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_mean_rows", &group_mean_rows);
   m.def("bucket_mean", &bucket_mean);
   m.def("gram", &gram);
-  m.def("gram_median", &gram_median,
-        "fused Gram + coordinate median (one HBM pass)");
   m.def("krum_select", &krum_select);
   m.def("caf_matvec", &caf_matvec);
   m.def("caf_colsum", &caf_colsum, py::arg("X"), py::arg("a"),

@@ -323,181 +323,14 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   }
 }
 
-// ---------------------------------------------------------------------------
-// FUSED Gram + coordinate-wise median (bench config 2's two ops share
-// their input): the DIAG single-tile kernel above already stages every
-// (64 row x 256 k) chunk of X through LDS for the MFMAs — each chunk is
-// 256 COMPLETE columns, so the block's last two waves sort the chunk's
-// 128 column pairs straight out of LDS and emit medians while staging
-// dominates. X is read from HBM ONCE for BOTH aggregates (separately:
-// 16 GB twice = 6.95 ms at 64x125M; the sort VALU hides entirely under
-// the ~2.6 us/chunk staging budget).
-//
-// Register economics force the sort into two 32-row halves: at
-// launch_bounds(1024, 2) the VGPR budget is 64/lane, so the 64-row
-// column pair is sorted as two packed 32-value bitonic networks
-// (32 u32 regs), written back IN PLACE into the chunk's LDS (dead after
-// the MFMA phase), and the median comes from a two-pointer merge walk
-// of the sorted halves. Pads (rows >= n) get key 0xFFFF at read time.
-// ---------------------------------------------------------------------------
-#include "pksort.h"
-
-// NOTE: on AMD the second __launch_bounds__ argument is MIN WAVES PER
-// EU: 8 forces the 64-VGPR budget that lets 2 blocks (32 waves)
-// co-reside like the unfused gram kernel
-template <int BK = 256>
-__global__ void __launch_bounds__(1024, 8)
-gram_median_bf16_kernel(const __hip_bfloat16* __restrict__ X,
-                        float* __restrict__ G,
-                        __hip_bfloat16* __restrict__ med, int n, long d,
-                        long k_per_block) {
-  constexpr int SLOTS = BK / 8;
-  constexpr int SUB = SLOTS / 16;
-  constexpr int CHUNK_BYTES = TILE * BK * 2;
-  __shared__ char smem[2 * CHUNK_BYTES];
-  char* const bufA0 = smem;
-  char* const bufA1 = smem + CHUNK_BYTES;
-
-  const int t = threadIdx.x;
-  const int wave = t >> 6;
-  const int lane = t & 63;
-  const int wr = wave >> 2, wc = wave & 3;
-
-  const long k_lo = (long)blockIdx.x * k_per_block;
-  const long k_hi = min(d, k_lo + k_per_block);
-  const long nchunks = (k_hi - k_lo + BK - 1) / BK;
-
-  const int st_row = t >> 4;
-  const int a_rows = min(TILE, n);
-  int st_slot[SUB], src_slot[SUB];
-#pragma unroll
-  for (int u = 0; u < SUB; ++u) {
-    st_slot[u] = (t & 15) + u * 16;
-    src_slot[u] = st_slot[u] ^ (st_row & 15);
-  }
-  const __hip_bfloat16* a_row_src = X + (long)min(st_row, a_rows - 1) * d;
-
-  bf16x8 ra[SUB];
-  auto stage_load = [&](long c) {
-    const long k0 = k_lo + c * BK;
-    const bool full = (k0 + BK <= k_hi);
-#pragma unroll
-    for (int u = 0; u < SUB; ++u) {
-      const long off = k0 + (long)src_slot[u] * 8;
-      if (full && a_rows == TILE) {
-        ra[u] = *reinterpret_cast<const bf16x8*>(a_row_src + off);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          ra[u][j] = (st_row < a_rows && off + j < k_hi)
-                         ? *reinterpret_cast<const __bf16*>(a_row_src + off + j)
-                         : (__bf16)0.0f;
-      }
-    }
-  };
-  auto stage_write = [&](int which) {
-    char* a = which ? bufA1 : bufA0;
-#pragma unroll
-    for (int u = 0; u < SUB; ++u)
-      *reinterpret_cast<bf16x8*>(a + st_row * (SLOTS * 16) + st_slot[u] * 16) =
-          ra[u];
-  };
-
-  f32x4 acc = {0.0f, 0.0f, 0.0f, 0.0f};
-  const int rowA = wr * 16 + (lane & 15);
-  const int rowB = wc * 16 + (lane & 15);
-  const int grp = lane >> 4;
-  // median phase geometry: EVERY wave rank-selects 16 of the chunk's 256
-  // columns (one column at a time, one staged value per lane, ballot
-  // bit-descent) — concentrating the median work on 2-4 sort waves left
-  // it latency-serial on the chunk critical path (35 ms total, 3 kernel
-  // variants measured; the standalone median's VALU work is ~1.85 ms at
-  // this shape and only hides under the 2.5 ms staging when all 16
-  // waves carry it)
-  const int plo = (n - 1) >> 1, phi = n >> 1;
-
-  stage_load(0);
-  stage_write(0);
-  __syncthreads();
-  for (long c = 0; c < nchunks; ++c) {
-    if (c + 1 < nchunks) stage_load(c + 1);
-    char* A = (c & 1) ? bufA1 : bufA0;
-#pragma unroll
-    for (int step = 0; step < BK / 32; ++step) {
-      const int q = step * 4 + grp;
-      const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          A + rowA * (SLOTS * 16) + ((q ^ (rowA & 15)) * 16));
-      const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-          A + rowB * (SLOTS * 16) + ((q ^ (rowB & 15)) * 16));
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-    }
-    // barrier FIRST: the sort mutates buf c in place, so every wave's
-    // MFMA reads must be complete. The sort then runs concurrently with
-    // the other waves' stage_write into the OTHER buffer (sort waves do
-    // their stage_write slice first, which also kills the prefetch
-    // registers before the 32-register sort arrays go live).
-    __syncthreads();
-    if (c + 1 < nchunks) stage_write((c + 1) & 1);
-    {
-      const long kc0 = k_lo + c * BK;
-#pragma unroll 1
-      for (int cj = 0; cj < BK / 16; ++cj) {
-        const int col = wave * (BK / 16) + cj;
-        const long gcol = kc0 + col;
-        if (gcol >= k_hi) break;
-        // lane r holds the staged value at (row=r, col); rows >= n pad
-        // with the max key. 4-way bank conflict (lanes 16 apart share a
-        // swizzle slot) — 4 cycles, amortized over the whole column.
-        const unsigned short raw = *reinterpret_cast<const unsigned short*>(
-            A + lane * (SLOTS * 16) + (((col >> 3) ^ (lane & 15)) * 16) +
-            (col & 7) * 2);
-        pk_u32 key;
-        if (lane < n) {
-          const pk_u32 b16 = raw;
-          const pk_u32 sg = (b16 >> 15) & 1u;
-          key = (b16 ^ (0x8000u + sg * 0x7FFFu)) & 0xFFFFu;
-        } else {
-          key = 0xFFFFu;
-        }
-        // ballot bit-descent rank selection (exact with duplicates):
-        // resolve the rank-k key from the high bit down, counting the
-        // candidates whose unresolved bit is 0
-        pk_u32 keys[2] = {0, 0};
-#pragma unroll 1
-        for (int which = 0; which < 2; ++which) {
-          int rank = which ? phi : plo;
-          pk_u32 prefix = 0;
-          for (int bit = 15; bit >= 0; --bit) {
-            const unsigned long long m =
-                __ballot((key >> bit) == (prefix >> bit));
-            const int c0 = __popcll(m);
-            if (rank >= c0) {
-              rank -= c0;
-              prefix |= 1u << bit;
-            }
-          }
-          keys[which] = prefix;
-        }
-        if (lane == 0) {
-          const float mv =
-              0.5f * (pk_key_to_float(keys[0]) + pk_key_to_float(keys[1]));
-          union { unsigned short s3; __hip_bfloat16 h3; } o;
-          o.h3 = __float2bfloat16(mv);
-          reinterpret_cast<unsigned short*>(med)[gcol] = o.s3;
-        }
-      }
-    }
-    __syncthreads();
-  }
-
-  const int out_col = wc * 16 + (lane & 15);
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int out_row = wr * 16 + (lane >> 4) * 4 + r;
-    if (out_row < n && out_col < n)
-      atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
-  }
-}
+// (A FUSED gram+median kernel lived here and was removed after SIX
+// measured variants all lost to the separate kernels — 32-60 ms vs
+// 6.9 ms at 64x125M. The median's per-chunk work only amortizes when
+// each lane holds a full column (64 VGPRs of keys), which either
+// spills or halves occupancy inside the gram's register/LDS budget;
+// register-light organizations serialize on LDS merge chains or waste
+// 63/64 lanes on ballot descents. Full writeup:
+// profiles/r02_fusion_negative.md.)
 
 // ---------------------------------------------------------------------------
 // LDS-staged f32 Gram (fast path for d % 4 == 0): same structure as the
@@ -1081,15 +914,6 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
     hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
                        stream, X, G, n, d, kpb);
   }
-}
-
-void launch_gram_median_bf16(const __hip_bfloat16* X, float* G,
-                             __hip_bfloat16* med, int n, long d,
-                             hipStream_t stream) {
-  int splitk; long kpb;
-  split_geometry(n, d, splitk, kpb);  // kpb is 256-aligned by construction
-  hipLaunchKernelGGL((gram_median_bf16_kernel<256>), dim3(splitk),
-                     dim3(1024), 0, stream, X, G, med, n, d, kpb);
 }
 
 void launch_gram_f32(const float* X, float* G, int n, long d,

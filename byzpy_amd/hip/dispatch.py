@@ -63,14 +63,12 @@ def gram(X: torch.Tensor) -> torch.Tensor:
 
 
 def median_and_gram(X: torch.Tensor):
-    """Both of the flagship aggregates from ONE HBM pass (the fused
-    kernel's sort waves ride the Gram's LDS staging). Returns
-    (median, gram). Falls back to the two separate kernels off the
-    fused kernel's shape envelope."""
-    n, d = X.shape
-    if _gpu(X) and X.dtype == torch.bfloat16 and n <= 64 and d >= 1024:
-        G, med = _hip.require().gram_median(X.contiguous())
-        return med, G
+    """Both flagship aggregates: (median, gram). Two separate kernels —
+    a fused single-HBM-pass kernel was built and measured across six
+    organizations and LOST to this pair every time (32-60 ms vs 6.9 at
+    64x125M; see profiles/r02_fusion_negative.md): the median work only
+    amortizes with full per-lane column residency, which cannot coexist
+    with the Gram's register/occupancy budget."""
     return median(X), gram(X)
 
 
