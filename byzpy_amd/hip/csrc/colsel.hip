@@ -1,6 +1,9 @@
 // Column-wise order-statistic kernels: per-coordinate
 // median / trimmed mean / mean-of-medians over the n-axis of an (n, d)
-// matrix, one column per thread.
+// matrix, one column per thread. Large n (> 64 at d >= 32K, always past
+// 512) is served by the generic streaming radix-select engine in
+// rsel.hip; this file keeps the small-n register kernels and the
+// mid-n/small-d LDS sort.
 //
 // Two variants:
 //  - register kernel (n <= 64): the column lives in a register array; the
