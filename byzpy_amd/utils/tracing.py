@@ -18,7 +18,11 @@ def _roctx() -> Optional[ctypes.CDLL]:
     global _lib, _tried
     if not _tried:
         _tried = True
-        for name in ("libroctx64.so", "libroctx64.so.4", "librocprofiler-sdk-roctx.so"):
+        # the rocprofiler-sdk roctx FIRST: rocprofv3 --marker-trace only
+        # captures SDK markers (the legacy roctracer libroctx64 loads
+        # fine but its ranges never reach the rocprofv3 timeline)
+        for name in ("librocprofiler-sdk-roctx.so", "libroctx64.so",
+                     "libroctx64.so.4"):
             try:
                 _lib = ctypes.CDLL(name)
                 break
