@@ -42,6 +42,9 @@ def main() -> None:
                         "fixed point barely moves per gossip round)")
     p.add_argument("--graph", action="store_true",
                    help="capture the whole gossip round in one hipGraph")
+    p.add_argument("--grouped", action="store_true",
+                   help="batch ALL node updates into grouped kernels "
+                        "(one launch pair per Weiszfeld iteration)")
     p.add_argument("--device", default="cuda" if torch.cuda.is_available() else "cpu")
     args = p.parse_args()
     dev = torch.device(args.device)
@@ -85,7 +88,26 @@ def main() -> None:
         for i in range(args.nodes)
     ]
 
+    group_mat = torch.stack(groups)  # (nodes, 1+2k)
+    warm_grouped: dict = {}
+
+    def round_once_grouped():
+        Xg = theta[group_mat]  # (G, m, d)
+        mixed = D.nnm_grouped(Xg, args.f)
+        Z = D.geometric_median_grouped(
+            mixed,
+            iters=(fixed or 4) if args.warm_start and "Z" in warm_grouped
+            else (fixed or 8),
+            init_z=warm_grouped.get("Z") if args.warm_start else None,
+        )
+        if args.warm_start:
+            warm_grouped["Z"] = Z
+        theta.copy_(Z.to(theta.dtype))
+
     def round_once():
+        if args.grouped:
+            round_once_grouped()
+            return
         new = torch.empty_like(theta)
         if streams is None:
             for i in range(args.nodes):

@@ -33,6 +33,9 @@ template <typename T>
 void launch_weiszfeld_update(const T*, const float*, const float*, float*,
                              float*, int, long, float, hipStream_t);
 template <typename T>
+void launch_grouped_weiszfeld(const T*, const float*, float*, float*, int,
+                              int, long, float, hipStream_t);
+template <typename T>
 void launch_cc_update(const T*, const float*, const float*, float*, int, long,
                       float, float, hipStream_t);
 // rsel.hip: generic multi-pass radix-select for the other large-n modes
@@ -313,6 +316,37 @@ torch::Tensor krum_select(torch::Tensor G, int64_t f, int64_t q) {
   launch_krum_select(G.data_ptr<float>(), n, (int)f, (int)q,
                      idx.data_ptr<int>(), nullptr, cur_stream());
   return idx;
+}
+
+// Grouped Weiszfeld iteration: G independent (m, d) problems, one kernel
+// pair per iteration for ALL groups (gossip rounds: every node's geomed
+// advances together — no per-node streams, no launch storms).
+torch::Tensor weiszfeld_iter_grouped(torch::Tensor X3, torch::Tensor Z,
+                                     double eps) {
+  TORCH_CHECK(X3.is_cuda() && X3.dim() == 3 && X3.is_contiguous() &&
+              (X3.scalar_type() == torch::kFloat32 ||
+               X3.scalar_type() == torch::kBFloat16),
+              "expected contiguous (G, m, d) f32/bf16");
+  const int G = (int)X3.size(0);
+  const int m = (int)X3.size(1);
+  const long d = (long)X3.size(2);
+  TORCH_CHECK(m >= 1 && m <= 32, "grouped weiszfeld supports m <= 32");
+  TORCH_CHECK(Z.is_cuda() && Z.scalar_type() == torch::kFloat32 &&
+              Z.is_contiguous() && Z.dim() == 2 && Z.size(0) == G &&
+              Z.size(1) == d);
+  auto dist2 = torch::zeros({G, (long)m},
+                            X3.options().dtype(torch::kFloat32));
+  auto Z_new = torch::empty_like(Z);
+  if (X3.scalar_type() == torch::kFloat32)
+    launch_grouped_weiszfeld<float>(X3.data_ptr<float>(), Z.data_ptr<float>(),
+                                    dist2.data_ptr<float>(),
+                                    Z_new.data_ptr<float>(), G, m, d,
+                                    (float)eps, cur_stream());
+  else
+    launch_grouped_weiszfeld<__hip_bfloat16>(
+        bf16_ptr(X3), Z.data_ptr<float>(), dist2.data_ptr<float>(),
+        Z_new.data_ptr<float>(), G, m, d, (float)eps, cur_stream());
+  return Z_new;
 }
 
 // Sharded form: apply the Weiszfeld update with externally-reduced global
@@ -693,6 +727,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("caf_colsum", &caf_colsum, py::arg("X"), py::arg("a"),
         py::arg("mu") = c10::nullopt, py::arg("scale") = c10::nullopt);
   m.def("weiszfeld_iter", &weiszfeld_iter);
+  m.def("weiszfeld_iter_grouped", &weiszfeld_iter_grouped,
+        "one Weiszfeld iteration for G independent (m, d) groups");
   m.def("weiszfeld_apply", &weiszfeld_apply);
   m.def("cc_iter", &cc_iter);
   m.def("cc_apply", &cc_apply);

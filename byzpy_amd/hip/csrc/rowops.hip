@@ -205,6 +205,110 @@ __global__ void center_sqdists_group_kernel(const T* __restrict__ X,
   }
 }
 
+// -- grouped center distances + Weiszfeld update ---------------------------
+// G independent small aggregation problems per launch (gossip: every
+// node's geomed iteration in ONE kernel pair instead of 2G launches on
+// G streams). blockIdx.y = group; each group has its own center.
+
+template <typename T, bool VEC>
+__global__ void center_sqdists_grouped_kernel(const T* __restrict__ X,
+                                              const float* __restrict__ Z,
+                                              float* __restrict__ out, int m,
+                                              long d) {
+  __shared__ float lds[DIST_GROUP][16];
+  constexpr int V = VecTraits<T>::V;
+  const int g = blockIdx.y;
+  const T* Xg = X + (long)g * m * d;
+  const float* z = Z + (long)g * d;
+  float acc[DIST_GROUP] = {0};
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float zv[V];
+      {
+        float tmp[4];
+#pragma unroll
+        for (int q4 = 0; q4 < V / 4; ++q4) {
+          VecTraits<float>::load(z + jv * V + q4 * 4, tmp);
+#pragma unroll
+          for (int c = 0; c < 4; ++c) zv[q4 * 4 + c] = tmp[c];
+        }
+      }
+      for (int i = 0; i < m; ++i) {
+        float x[V];
+        VecTraits<T>::load(Xg + (long)i * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) {
+          const float dd = x[c] - zv[c];
+          acc[i] += dd * dd;
+        }
+      }
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      const float zj = z[j];
+      for (int i = 0; i < m; ++i) {
+        const float dd = to_f<T>(Xg[(long)i * d + j]) - zj;
+        acc[i] += dd * dd;
+      }
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < DIST_GROUP; ++i) {
+    if (i < m) {
+      const float sred = block_reduce_sum(acc[i], lds[i]);
+      if (threadIdx.x == 0) atomicAdd(&out[(long)g * m + i], sred);
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, bool VEC>
+__global__ void weiszfeld_update_grouped_kernel(
+    const T* __restrict__ X, const float* __restrict__ Z,
+    const float* __restrict__ dist2, float* __restrict__ Z_new, int m, long d,
+    float eps) {
+  __shared__ float w_lds[DIST_GROUP];
+  constexpr int V = VecTraits<T>::V;
+  const int g = blockIdx.y;
+  const T* Xg = X + (long)g * m * d;
+  const float* z = Z + (long)g * d;
+  float* z_new = Z_new + (long)g * d;
+  for (int i = threadIdx.x; i < m; i += blockDim.x)
+    w_lds[i] = 1.0f / fmaxf(sqrtf(dist2[(long)g * m + i]), eps);
+  __syncthreads();
+  float den = 0.0f;
+  for (int i = 0; i < m; ++i) den += w_lds[i];
+  const float inv_den = 1.0f / den;
+
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long dv = d / V;
+    for (long jv = start; jv < dv; jv += stride) {
+      float num[V] = {0};
+      for (int i = 0; i < m; ++i) {
+        const float w = w_lds[i];
+        float x[V];
+        VecTraits<T>::load(Xg + (long)i * d + jv * V, x);
+#pragma unroll
+        for (int c = 0; c < V; ++c) num[c] += w * x[c];
+      }
+#pragma unroll
+      for (int c = 0; c < V; ++c) z_new[jv * V + c] = num[c] * inv_den;
+    }
+  } else {
+    for (long j = start; j < d; j += stride) {
+      float num = 0.0f;
+      for (int i = 0; i < m; ++i)
+        num += w_lds[i] * to_f<T>(Xg[(long)i * d + j]);
+      z_new[j] = num * inv_den;
+    }
+  }
+}
+
 // -- row scaling ------------------------------------------------------------
 
 template <typename T, bool VEC>
@@ -644,6 +748,31 @@ void launch_row_center_sqdists(const T* X, const float* z, float* out, int n,
     hipLaunchKernelGGL((center_sqdists_group_kernel<T, false>), grid,
                        dim3(block), 0, stream, X, z, out, n, d);
 }
+
+template <typename T>
+void launch_grouped_weiszfeld(const T* X, const float* Z, float* dist2,
+                              float* Z_new, int G, int m, long d, float eps,
+                              hipStream_t stream) {
+  const int block = 256;
+  dim3 grid(slab_grid(d, G, block), G);
+  if (vec_ok<T>(d)) {
+    hipLaunchKernelGGL((center_sqdists_grouped_kernel<T, true>), grid,
+                       dim3(block), 0, stream, X, Z, dist2, m, d);
+    hipLaunchKernelGGL((weiszfeld_update_grouped_kernel<T, true>), grid,
+                       dim3(block), 0, stream, X, Z, dist2, Z_new, m, d, eps);
+  } else {
+    hipLaunchKernelGGL((center_sqdists_grouped_kernel<T, false>), grid,
+                       dim3(block), 0, stream, X, Z, dist2, m, d);
+    hipLaunchKernelGGL((weiszfeld_update_grouped_kernel<T, false>), grid,
+                       dim3(block), 0, stream, X, Z, dist2, Z_new, m, d, eps);
+  }
+}
+template void launch_grouped_weiszfeld<float>(const float*, const float*,
+                                              float*, float*, int, int, long,
+                                              float, hipStream_t);
+template void launch_grouped_weiszfeld<__hip_bfloat16>(
+    const __hip_bfloat16*, const float*, float*, float*, int, int, long,
+    float, hipStream_t);
 
 template <typename T>
 void launch_row_scale(const T* X, const float* s, T* out, int n, long d,

@@ -249,6 +249,55 @@ def geometric_median(
     return z.to(X.dtype)
 
 
+def geometric_median_grouped(
+    X3: torch.Tensor,
+    *,
+    iters: int = 8,
+    eps: float = 1e-12,
+    init_z: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Geometric medians of G independent (m, d) groups in one kernel
+    pair per iteration (gossip rounds: all nodes advance together —
+    replaces per-node streams + 2*G*iters launches). Fixed iteration
+    count (poll-free); warm-start with ``init_z`` (G, d)."""
+    G, m, d = X3.shape
+    if not _gpu(X3) or m > 32:
+        Z = []
+        for g in range(G):
+            z0 = init_z[g] if init_z is not None else None
+            Z.append(
+                geometric_median(
+                    X3[g], fixed_iters=int(iters), eps=eps, init_z=z0
+                ).float()
+            )
+        return torch.stack(Z)
+    ext = _hip.require()
+    Xc = X3.contiguous()
+    if init_z is not None:
+        Z = init_z.detach().to(dtype=torch.float32).contiguous()
+    else:
+        Z = Xc.float().median(dim=1).values.contiguous()
+    for _ in range(int(iters)):
+        Z = ext.weiszfeld_iter_grouped(Xc, Z, float(eps))
+    return Z
+
+
+def nnm_grouped(X3: torch.Tensor, f: int) -> torch.Tensor:
+    """NNM for G independent (m, d) groups via strided-batched GEMMs:
+    bmm distance matrices + a (1/k)-mask bmm mixes neighbors without
+    materializing gathered copies."""
+    G, m, d = X3.shape
+    k = m - f
+    Xf = X3.float()
+    Gm = torch.bmm(Xf, Xf.transpose(1, 2))  # (G, m, m)
+    norms = torch.diagonal(Gm, dim1=1, dim2=2)
+    D2 = (norms[:, :, None] + norms[:, None, :] - 2.0 * Gm).clamp_(min=0.0)
+    idx = torch.topk(D2, k=k, dim=2, largest=False).indices
+    mask = torch.zeros_like(Gm)
+    mask.scatter_(2, idx, 1.0 / k)
+    return torch.bmm(mask, Xf).to(X3.dtype)
+
+
 def centered_clipping(
     X: torch.Tensor,
     *,

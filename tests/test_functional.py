@@ -283,3 +283,27 @@ class TestRound2Seams:
 
         with trace_range("x"):
             pass  # must not raise whether or not a roctx lib exists
+
+    def test_grouped_geomed_cpu(self):
+        import torch
+
+        from byzpy_amd.hip import dispatch as D
+
+        g = torch.Generator().manual_seed(24)
+        X3 = torch.randn(3, 5, 257, generator=g)
+        Z = D.geometric_median_grouped(X3, iters=30)
+        for i in range(3):
+            zi = D.geometric_median(X3[i], fixed_iters=30)
+            assert torch.allclose(Z[i], zi.float(), atol=1e-4), i
+
+    def test_grouped_nnm_cpu(self):
+        import torch
+
+        from byzpy_amd.hip import dispatch as D
+        from byzpy_amd.ops import functional as F
+
+        g = torch.Generator().manual_seed(25)
+        X3 = torch.randn(4, 6, 128, generator=g)
+        out = D.nnm_grouped(X3, 2)
+        for i in range(4):
+            assert torch.allclose(out[i], F.nnm(X3[i], 2), atol=1e-4), i

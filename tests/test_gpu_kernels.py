@@ -802,3 +802,41 @@ class TestFusedGramMedian:
         med, krum_out = sharded.median_and_multi_krum(X, 16, 12)
         assert torch.equal(med, sharded.median(X))
         assert torch.equal(krum_out, sharded.multi_krum(X, 16, 12))
+
+
+class TestGroupedOps:
+    """Grouped gossip kernels: all nodes' aggregates per launch."""
+
+    @pytest.mark.parametrize("dtype", DTYPES)
+    def test_grouped_weiszfeld_matches_per_group(self, dtype):
+        g = torch.Generator().manual_seed(21)
+        X3 = torch.randn(6, 5, 4096, generator=g).to("cuda", dtype)
+        Z = D.geometric_median_grouped(X3, iters=12)
+        for i in range(6):
+            zi = D.geometric_median(X3[i], fixed_iters=12)
+            assert torch.allclose(Z[i], zi.float(), atol=1e-2, rtol=1e-2), i
+
+    def test_grouped_weiszfeld_warm_start(self):
+        g = torch.Generator().manual_seed(22)
+        X3 = torch.randn(4, 7, 2048, generator=g).cuda()
+        Z0 = D.geometric_median_grouped(X3, iters=100)
+        Z1 = D.geometric_median_grouped(X3, iters=2, init_z=Z0)
+        assert torch.allclose(Z0, Z1, atol=1e-3)
+
+    def test_nnm_grouped_matches_per_group(self):
+        g = torch.Generator().manual_seed(23)
+        X3 = torch.randn(5, 6, 1024, generator=g).to("cuda", torch.bfloat16)
+        out = D.nnm_grouped(X3, 2)
+        for i in range(5):
+            ref = D.nnm(X3[i], 2)
+            assert torch.allclose(
+                out[i].float(), ref.float(), atol=5e-2, rtol=5e-2
+            ), i
+
+    def test_grouped_cpu_fallback(self):
+        g = torch.Generator().manual_seed(24)
+        X3 = torch.randn(3, 5, 257, generator=g)
+        Z = D.geometric_median_grouped(X3, iters=30)
+        for i in range(3):
+            zi = D.geometric_median(X3[i], fixed_iters=30)
+            assert torch.allclose(Z[i], zi.float(), atol=1e-4), i
