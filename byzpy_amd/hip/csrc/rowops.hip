@@ -236,7 +236,24 @@ __global__ void center_sqdists_grouped_kernel(const T* __restrict__ X,
           for (int c = 0; c < 4; ++c) zv[q4 * 4 + c] = tmp[c];
         }
       }
-      for (int i = 0; i < m; ++i) {
+      int i = 0;
+      for (; i + 4 <= m; i += 4) {  // 4 row loads in flight
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(Xg + (long)(i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(Xg + (long)(i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(Xg + (long)(i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(Xg + (long)(i + 3) * d + jv * V, x3);
+#pragma unroll
+        for (int c = 0; c < V; ++c) {
+          const float d0 = x0[c] - zv[c], d1 = x1[c] - zv[c];
+          const float d2 = x2[c] - zv[c], d3 = x3[c] - zv[c];
+          acc[i + 0] += d0 * d0;
+          acc[i + 1] += d1 * d1;
+          acc[i + 2] += d2 * d2;
+          acc[i + 3] += d3 * d3;
+        }
+      }
+      for (; i < m; ++i) {
         float x[V];
         VecTraits<T>::load(Xg + (long)i * d + jv * V, x);
 #pragma unroll
@@ -289,7 +306,20 @@ __global__ void weiszfeld_update_grouped_kernel(
     const long dv = d / V;
     for (long jv = start; jv < dv; jv += stride) {
       float num[V] = {0};
-      for (int i = 0; i < m; ++i) {
+      int i = 0;
+      for (; i + 4 <= m; i += 4) {  // 4 row loads in flight
+        float x0[V], x1[V], x2[V], x3[V];
+        VecTraits<T>::load(Xg + (long)(i + 0) * d + jv * V, x0);
+        VecTraits<T>::load(Xg + (long)(i + 1) * d + jv * V, x1);
+        VecTraits<T>::load(Xg + (long)(i + 2) * d + jv * V, x2);
+        VecTraits<T>::load(Xg + (long)(i + 3) * d + jv * V, x3);
+        const float w0 = w_lds[i], w1 = w_lds[i + 1], w2 = w_lds[i + 2],
+                    w3 = w_lds[i + 3];
+#pragma unroll
+        for (int c = 0; c < V; ++c)
+          num[c] += (w0 * x0[c] + w1 * x1[c]) + (w2 * x2[c] + w3 * x3[c]);
+      }
+      for (; i < m; ++i) {
         const float w = w_lds[i];
         float x[V];
         VecTraits<T>::load(Xg + (long)i * d + jv * V, x);
