@@ -324,8 +324,11 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
-    if (out_row < n && out_col < n)
+    if (out_row < n && out_col < n) {
       atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
+      if (MIRROR && blockIdx.z != blockIdx.y)
+        atomicAdd(&G[(long)out_col * n + out_row], acc[r]);
+    }
   }
 }
 
