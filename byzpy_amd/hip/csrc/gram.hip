@@ -210,7 +210,7 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   // — halves the staged HBM traffic (the full grid re-read every row
   // band once per PAIRED tile: n=128 measured 2.45 TB/s vs the ~6.3
   // single-tile ceiling purely from that amplification).
-  if (MIRROR && blockIdx.z < blockIdx.y) return;
+  if (MIRROR && blockIdx.z <= blockIdx.y) return;  // diag tiles run DIAG=true
   constexpr int SLOTS = BK / 8;                 // 16-B slots per row
   constexpr int SUB = SLOTS / 16;               // stage passes per thread
   constexpr int CHUNK_BYTES = TILE * BK * 2;    // 16 KB at BK=128
@@ -227,7 +227,9 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
   const int lane = t & 63;
   const int wr = wave >> 2, wc = wave & 3;
   const int row_base = blockIdx.y * TILE;
-  const int col_base = blockIdx.z * TILE;
+  // DIAG grids launch with gridDim.z == 1 and enumerate diagonal tiles
+  // along y (single shared A==B image per tile)
+  const int col_base = DIAG ? row_base : blockIdx.z * TILE;
 
   const long k_lo = (long)blockIdx.x * k_per_block;
   const long k_hi = min(d, k_lo + k_per_block);
@@ -326,7 +328,7 @@ gram_bf16_lds_kernel(const __hip_bfloat16* __restrict__ X,
     const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
     if (out_row < n && out_col < n) {
       atomicAdd(&G[(long)out_row * n + out_col], acc[r]);
-      if (MIRROR && blockIdx.z != blockIdx.y)
+      if (MIRROR)  // strictly-upper tiles only
         atomicAdd(&G[(long)out_col * n + out_row], acc[r]);
     }
   }
@@ -358,7 +360,7 @@ template <bool DIAG, int BKF = 64, bool MIRROR = false>
 __global__ void __launch_bounds__(1024, 2)
 gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
                     int n, long d, long k_per_block) {
-  if (MIRROR && blockIdx.z < blockIdx.y) return;  // symmetric-tile skip
+  if (MIRROR && blockIdx.z <= blockIdx.y) return;  // diag tiles run DIAG=true
   constexpr int CHUNK_BYTES = TILE * BKF * 4;    // 16 KB
   __shared__ char smem[(DIAG ? 2 : 4) * CHUNK_BYTES];
   char* const bufA0 = smem;
@@ -371,7 +373,7 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
   const int lane = t & 63;
   const int wr = wave >> 2, wc = wave & 3;
   const int row_base = blockIdx.y * TILE;
-  const int col_base = blockIdx.z * TILE;
+  const int col_base = DIAG ? row_base : blockIdx.z * TILE;
 
   const long k_lo = (long)blockIdx.x * k_per_block;
   const long k_hi = min(d, k_lo + k_per_block);
@@ -476,7 +478,7 @@ gram_f32_lds_kernel(const float* __restrict__ X, float* __restrict__ G,
     const int out_row = row_base + wr * 16 + (lane >> 4) * 4 + r;
     if (out_row < n && out_col < n) {
       atomicAdd(&G[(long)out_row * n + out_col], acc0[r] + acc1[r]);
-      if (MIRROR && blockIdx.z != blockIdx.y)
+      if (MIRROR)  // strictly-upper tiles only
         atomicAdd(&G[(long)out_col * n + out_row], acc0[r] + acc1[r]);
     }
   }
@@ -911,19 +913,27 @@ void launch_gram_bf16(const __hip_bfloat16* X, float* G, int n, long d,
       // 2 x 32 KB still fits 2 blocks/CU with the single A image)
       hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 256>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
-    else
-      // symmetric tiling: lower-triangle blocks exit, upper blocks write
-      // both halves (diag blocks of the MIRROR grid stay DIAG=false —
-      // they stage A==B twice, a tiles-vs-tiles^2/2 minority cost)
+    else {
+      // symmetric tiling: diagonal tiles on the single-image DIAG
+      // staging (grid z=1, tiles along y), strictly-upper tiles mirror
+      // into both halves, lower blocks exit immediately
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 128>),
+                         dim3(splitk, tiles, 1), dim3(WAVES * 64), 0, stream,
+                         X, G, n, d, kpb);
       hipLaunchKernelGGL((gram_bf16_lds_kernel<false, 128, true>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+    }
   } else if ((d % 8) == 0 && (kpb % 128) == 0) {
     if (tiles == 1)
       hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 128>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
-    else
+    else {
+      hipLaunchKernelGGL((gram_bf16_lds_kernel<true, 128>),
+                         dim3(splitk, tiles, 1), dim3(WAVES * 64), 0, stream,
+                         X, G, n, d, kpb);
       hipLaunchKernelGGL((gram_bf16_lds_kernel<false, 128, true>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+    }
   } else {
     hipLaunchKernelGGL((gram_bf16_kernel<false>), grid, dim3(WAVES * 64), 0,
                        stream, X, G, n, d, kpb);
@@ -962,12 +972,16 @@ void launch_gram_f32(const float* X, float* G, int n, long d,
     return;
   }
   if ((d % 4) == 0 && (kpb % 64) == 0) {
-    if (tiles == 1)
+    if (tiles == 1) {
       hipLaunchKernelGGL((gram_f32_lds_kernel<true, 64>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
-    else
+    } else {
+      hipLaunchKernelGGL((gram_f32_lds_kernel<true, 64>),
+                         dim3(splitk, tiles, 1), dim3(WAVES * 64), 0, stream,
+                         X, G, n, d, kpb);
       hipLaunchKernelGGL((gram_f32_lds_kernel<false, 64, true>), grid,
                          dim3(WAVES * 64), 0, stream, X, G, n, d, kpb);
+    }
     return;
   }
   hipLaunchKernelGGL(gram_f32_kernel, grid, dim3(WAVES * 64), 0, stream, X, G,
