@@ -347,6 +347,30 @@ def test_config5_sharded_path_and_checkpoint(tmp_path):
     _run_workers_n(_body_config5_path, 2, str(tmp_path))
 
 
+def _body_bench_path_ws8(rank):
+    """The exact bench step (sharded median_and_multi_krum) at world
+    size 8 — the driver's largest SCALE point — against the full-matrix
+    single-process reference."""
+    from byzpy_amd.hip import dispatch as D
+    from byzpy_amd.parallel import sharded
+    from byzpy_amd.parallel.dist import column_shard
+
+    g = torch.Generator().manual_seed(42)
+    n, d = 16, 320
+    X = torch.randn(n, d, generator=g)  # identical on all ranks
+    Xs = column_shard(X, rank)
+    med, krum = sharded.median_and_multi_krum(Xs, 3, 4)
+    lo = rank * 40
+    ref_med, ref_G = D.median_and_gram(X)
+    assert torch.allclose(med, ref_med[lo : lo + 40], atol=1e-5)
+    ref_krum = D.multi_krum(X, 3, 4)
+    assert torch.allclose(krum, ref_krum[lo : lo + 40], atol=1e-4)
+
+
+def test_bench_path_world8():
+    _run_workers_n(_body_bench_path_ws8, 8)
+
+
 def test_ring_p2p_byzantine_context():
     _run_workers_n(_body_ring_p2p, 4)
 
