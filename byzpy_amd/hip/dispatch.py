@@ -440,6 +440,7 @@ def monna(X: torch.Tensor, f: int, reference_index: int = 0) -> torch.Tensor:
 
 
 _SMEA_COMBOS: dict = {}
+_SMEA_COMBOS_MAX_CACHED = 4  # FIFO cap: a combos tensor can be ~33 MB
 _SMEA_MAX_COMBOS = 1 << 17
 
 
@@ -462,6 +463,8 @@ def smea(X: torch.Tensor, f: int) -> torch.Tensor:
         key = (n, m, X.device.index)
         combos = _SMEA_COMBOS.get(key)
         if combos is None:
+            if len(_SMEA_COMBOS) >= _SMEA_COMBOS_MAX_CACHED:
+                _SMEA_COMBOS.pop(next(iter(_SMEA_COMBOS)))
             combos = torch.tensor(
                 list(itertools.combinations(range(n), m)),
                 dtype=torch.int32,
@@ -574,6 +577,7 @@ class _CafGraphBlock:
 
 
 _CAF_GRAPHS: dict = {}
+_CAF_GRAPHS_MAX = 4  # FIFO cap: each block pins X + seeds device buffers
 
 # graph path bounds: seeds are staged per replay (R*d f32 H2D), so cap d;
 # n caps the round count (and kernel TORCH_CHECKs n <= 1024)
@@ -613,6 +617,8 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
         key = (n, d, X.dtype, int(f), int(power_iters), R, X.device.index)
         blk = _CAF_GRAPHS.get(key)
         if blk is None:
+            if len(_CAF_GRAPHS) >= _CAF_GRAPHS_MAX:
+                _CAF_GRAPHS.pop(next(iter(_CAF_GRAPHS)))
             blk = _CafGraphBlock(n, d, X.dtype, int(f), power_iters, R, X.device)
             _CAF_GRAPHS[key] = blk
         return blk.run(X.contiguous())
