@@ -60,6 +60,12 @@ class RcclPeerToPeer:
             raise ValueError(
                 "this rank has an attack but is not in byzantine_ranks"
             )
+        if attack is None and pdist.get_rank() in self.byzantine_ranks:
+            # a rank the OTHERS treat as byzantine must send an attack
+            # vector in phase 2 — otherwise their posted recvs deadlock
+            raise ValueError(
+                "this rank is in byzantine_ranks but has no attack"
+            )
         if transport not in {"auto", "all_gather", "p2p"}:
             raise ValueError(f"bad transport {transport!r}")
         if transport == "auto":

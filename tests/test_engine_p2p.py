@@ -192,3 +192,21 @@ def test_p2p_base_grad_attack():
             assert torch.isfinite(h.p2p_flat_params()).all()
 
     asyncio.run(gossip())
+
+
+def test_rccl_p2p_rejects_inconsistent_byzantine_set():
+    """A rank listed in byzantine_ranks without an attack would deadlock
+    its peers' phase-2 recvs — must be rejected at construction."""
+    import pytest
+    import torch
+
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+    from byzpy_amd.engine.peer_to_peer.rccl import RcclPeerToPeer
+
+    with pytest.raises(ValueError):
+        RcclPeerToPeer(
+            lambda: torch.ones(4),
+            lambda v: None,
+            CoordinateWiseMedian(),
+            byzantine_ranks=[0],  # this (only) rank has no attack
+        )
