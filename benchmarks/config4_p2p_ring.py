@@ -132,6 +132,8 @@ def main() -> None:
     if args.graph and dev.type == "cuda":
         if fixed is None:
             raise SystemExit("--graph requires --fixed-iters > 0 (capture-safe)")
+        if args.grouped:
+            raise SystemExit("--graph and --grouped are mutually exclusive")
         # capture one full gossip round (all node updates + the theta swap)
         # and replay it per round: zero Python dispatch in the loop
         s0 = torch.cuda.Stream()
