@@ -152,10 +152,16 @@ def geometric_median(
     max_iter: int = 256,
     eps: float = 1e-12,
     init: str = "median",
+    fixed_iters: "int | None" = None,
 ) -> torch.Tensor:
+    """``fixed_iters`` runs exactly that many iterations with NO
+    convergence polls — drops the per-poll shift all-reduce AND the host
+    sync, leaving one (n,) all-reduce per iteration (the minimum for the
+    d-sharded layout)."""
     if not X_local.is_cuda:
         return _geometric_median_cpu(
-            X_local, tol=tol, max_iter=max_iter, eps=eps, init=init
+            X_local, tol=tol, max_iter=max_iter, eps=eps, init=init,
+            fixed_iters=fixed_iters,
         )
     from byzpy_amd.hip import require
 
@@ -163,6 +169,11 @@ def geometric_median(
     Xc = X_local.contiguous()
     z = (D.median(Xc) if init == "median" else Xc.float().mean(dim=0)).float()
     shift = torch.zeros((), device=Xc.device, dtype=torch.float32)
+    if fixed_iters is not None:
+        for _ in range(int(fixed_iters)):
+            dist2 = all_reduce_(ext.row_center_sqdists(Xc, z))
+            z = ext.weiszfeld_apply(Xc, z, dist2, float(eps), shift)
+        return z.to(X_local.dtype)
     poll = 4
     it = 0
     while it < max_iter:
@@ -176,18 +187,22 @@ def geometric_median(
     return z.to(X_local.dtype)
 
 
-def _geometric_median_cpu(X, *, tol, max_iter, eps, init):
+def _geometric_median_cpu(X, *, tol, max_iter, eps, init, fixed_iters=None):
     Xf = X.float()
     z = (F.median(Xf) if init == "median" else Xf.mean(dim=0)).float()
-    for _ in range(max_iter):
+    iters = int(fixed_iters) if fixed_iters is not None else max_iter
+    for _ in range(iters):
         dist2 = all_reduce_(((Xf - z[None, :]) ** 2).sum(dim=1))
         d = dist2.sqrt().clamp_(min=eps)
         w = 1.0 / d
         z_new = (w[:, None] * Xf).sum(dim=0) / w.sum()
-        shift2 = all_reduce_(((z_new - z) ** 2).sum())
-        z = z_new
-        if float(shift2) <= tol * tol:
-            break
+        if fixed_iters is None:
+            shift2 = all_reduce_(((z_new - z) ** 2).sum())
+            z = z_new
+            if float(shift2) <= tol * tol:
+                break
+        else:
+            z = z_new
     return z.to(X.dtype)
 
 

@@ -58,6 +58,11 @@ def _body_sharded_ops(rank):
     med = sharded.median(Xs)
     assert torch.allclose(med, F.median(X)[lo:hi], atol=1e-5)
 
+    # fixed-iteration sharded geomed (poll-free: no shift all-reduce)
+    gm = sharded.geometric_median(Xs, fixed_iters=40)
+    gm_ref = F.geometric_median(X, tol=1e-30, max_iter=40)
+    assert torch.allclose(gm, gm_ref[lo:hi], atol=1e-4)
+
     mk = sharded.multi_krum(Xs, 2, 3)
     assert torch.allclose(mk, F.multi_krum(X, 2, 3)[lo:hi], atol=1e-4)
 
