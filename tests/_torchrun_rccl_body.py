@@ -42,6 +42,10 @@ def main() -> int:
         b = D.centered_clipping(X, c_tau=0.7, M=5)
         assert torch.equal(a, b), "sharded cc != dispatch cc"
 
+        a = sharded.geometric_median(X, fixed_iters=12)
+        b = D.geometric_median(X, fixed_iters=12)
+        assert torch.equal(a, b), "sharded fixed_iters geomed != dispatch"
+
     # neighbor_exchange at ws=1 is a no-op but must not deadlock
     got = pdist.neighbor_exchange([], [], torch.ones(4, device="cuda"))
     assert got == {}
