@@ -28,8 +28,14 @@ def main():
         out = D.caf(X, f)
         ref = F.caf(X.cpu(), f)
         err = (out.cpu() - ref).abs().max().item()
+        import os
+        os.environ["BYZPY_CAF_SYNC"] = "1"
+        ms_sync = t(lambda: D.caf(X, f), iters=10)
+        out_sync = D.caf(X, f)
+        del os.environ["BYZPY_CAF_SYNC"]
         ms = t(lambda: D.caf(X, f), iters=10)
-        print(f"caf n={n} d={d} f={f}: {ms:.3f} ms  max|err|={err:.2e}")
+        assert torch.equal(out, out_sync), "pipelined != sync loop"
+        print(f"caf n={n} d={d} f={f}: pipe {ms:.3f} ms  sync {ms_sync:.3f} ms  max|err|={err:.2e}")
         assert err < 1e-3, "parity broke"
     print("CAF_PROBE_OK")
 

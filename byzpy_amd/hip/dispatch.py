@@ -607,6 +607,7 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     functional oracle. Steady-state shapes run the hipGraph block path
     (_CafGraphBlock); others fall back to the eager loop below."""
     import math
+    import os
 
     n = X.shape[0]
     if not (_gpu(X) and n <= 1024 and n - 2 * f > 0):
@@ -640,7 +641,11 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
     # (speculative rounds launched past a break point are discarded —
     # they only ever consumed RNG stream, never affected the output),
     # and seed blocks stage through double-buffered pinned memory.
-    pin_seeds = block_rows * d * 4 <= (64 << 20)  # skip pinning huge-d blocks
+    sync_mode = os.environ.get("BYZPY_CAF_SYNC", "0") == "1"  # A/B probe aid
+    nopin = os.environ.get("BYZPY_CAF_NOPIN", "0") == "1"  # A/B probe aid
+    pin_seeds = (
+        not sync_mode and not nopin and block_rows * d * 4 <= (64 << 20)
+    )
     if pin_seeds:
         seed_pin = [
             torch.empty(block_rows, d, pin_memory=True) for _ in range(2)
@@ -696,7 +701,7 @@ def caf(X: torch.Tensor, f: int, *, power_iters: int = 3) -> torch.Tensor:
         stat_ev[slot].record()
         pending.append((mu, slot))
         w = w_next
-        if len(pending) == 2 and consume_oldest():
+        if len(pending) == (1 if sync_mode else 2) and consume_oldest():
             broke = True  # the still-pending round is speculative: discard
             break
     while not broke and pending:
