@@ -43,10 +43,14 @@ enum Mode { MEDIAN = 0, TRIMMED = 1, MEAMED = 2 };
 // register variant, n <= P, P in {8, 16, 32, 64}
 // ---------------------------------------------------------------------------
 
-template <int P>
+// KMAX < P stops the network after phase KMAX: with KMAX = P/2 the two
+// halves come out sorted ascending/descending — exactly the input the
+// median SELECTION epilogue needs (see median_select_reg), at the cost
+// of the full sort minus its entire last merge phase (log2(P) substages).
+template <int P, int KMAX = P>
 DEV void bitonic_sort_reg(float (&v)[P]) {
 #pragma unroll
-  for (int k = 2; k <= P; k <<= 1) {
+  for (int k = 2; k <= KMAX; k <<= 1) {
 #pragma unroll
     for (int j = k >> 1; j > 0; j >>= 1) {
 #pragma unroll
@@ -85,6 +89,31 @@ DEV void bitonic_sort_kv_reg(float (&key)[P], float (&val)[P]) {
       }
     }
   }
+}
+
+// Median ranks P/2-1 and P/2 of [asc32 | desc32] (a bitonic sequence):
+// one elementwise min/max pass splits it into the smallest and largest
+// P/2 multisets, then rank P/2-1 = max(lower), rank P/2 = min(upper) —
+// 3P/2 - 2 ops replacing the last merge phase (P/2*log2(P) CEs) PLUS the
+// two O(P) predicated rank extractions. The caller pre-pads so the
+// median lands exactly on these ranks (see the MEDIAN pad split below).
+template <int P>
+DEV void median_select_reg(float (&v)[P], float& mlo, float& mhi) {
+#pragma unroll
+  for (int i = 0; i < P / 2; ++i) {
+    const float a = v[i], b = v[i + P / 2];
+    v[i] = fminf(a, b);
+    v[i + P / 2] = fmaxf(a, b);
+  }
+#pragma unroll
+  for (int s = P / 4; s >= 1; s >>= 1)
+#pragma unroll
+    for (int i = 0; i < s; ++i) {
+      v[i] = fmaxf(v[i], v[i + s]);
+      v[P / 2 + i] = fminf(v[P / 2 + i], v[P / 2 + s + i]);
+    }
+  mlo = v[0];
+  mhi = v[P / 2];
 }
 
 template <int P>
@@ -130,21 +159,26 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
 #pragma unroll
       for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
       const int nv = vecify(n);
+      // MEDIAN pads split: L low pads (-inf) shift the median ranks to
+      // exactly P/2-1 / P/2 for the selection epilogue; a -inf pad tying
+      // a -inf data value is value-identical, so the statistic is exact.
+      // Other modes keep all-high pads (the sorted prefix is the data).
+      const int n_lo = vecify(
+          MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
 #pragma unroll
       for (int i = 0; i < P; ++i)
-        if (i >= nv) v[i] = PAD;
+        if (i >= nv)
+          v[i] = (MODE == MEDIAN && i < n_lo) ? -PAD : PAD;
     }
-    bitonic_sort_reg<P>(v);  // in place: meamed only needs sorted order
-
-    const int pos_lo = vecify((n - 1) >> 1), pos_hi = vecify(n >> 1);
-    const float med_lo = extract_at<P>(v, pos_lo);
-    const float med_hi = extract_at<P>(v, pos_hi);
-    const float med = 0.5f * (med_lo + med_hi);
 
     float result;
     if (MODE == MEDIAN) {
-      result = med;
+      bitonic_sort_reg<P, P / 2>(v);
+      float mlo, mhi;
+      median_select_reg<P>(v, mlo, mhi);
+      result = (n & 1) ? mlo : 0.5f * (mlo + mhi);
     } else if (MODE == TRIMMED) {
+      bitonic_sort_reg<P>(v);
       const int fv = vecify(f), nfv = vecify(n - f);
       float s = 0.0f;
 #pragma unroll
@@ -152,6 +186,10 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
         if (i >= fv && i < nfv) s += v[i];
       result = s / (float)(n - 2 * f);
     } else {  // MEAMED: mean of the n-f values closest to the median
+      bitonic_sort_reg<P>(v);
+      const int pos_lo = vecify((n - 1) >> 1), pos_hi = vecify(n >> 1);
+      const float med =
+          0.5f * (extract_at<P>(v, pos_lo) + extract_at<P>(v, pos_hi));
       const int nv2 = vecify(n), nfv = vecify(n - f);
       float dev[P];
 #pragma unroll
@@ -207,10 +245,10 @@ DEV float key_to_float(u32 key16) {
   return __bfloat162float(c.h);
 }
 
-template <int P>
+template <int P, int KMAX = P>
 DEV void bitonic_sort_pk(u32 (&v)[P]) {
 #pragma unroll
-  for (int k = 2; k <= P; k <<= 1) {
+  for (int k = 2; k <= KMAX; k <<= 1) {
 #pragma unroll
     for (int j = k >> 1; j > 0; j >>= 1) {
 #pragma unroll
@@ -226,6 +264,28 @@ DEV void bitonic_sort_pk(u32 (&v)[P]) {
       }
     }
   }
+}
+
+// packed twin of median_select_reg (see its comment): split the bitonic
+// [asc P/2 | desc P/2] keys, then max-reduce the lower / min-reduce the
+// upper half — both columns of the pair in one packed op throughout
+template <int P>
+DEV void pk_median_select(u32 (&v)[P], u32& mlo, u32& mhi) {
+#pragma unroll
+  for (int i = 0; i < P / 2; ++i) {
+    const u32 a = v[i], b = v[i + P / 2];
+    v[i] = pk_min_u16(a, b);
+    v[i + P / 2] = pk_max_u16(a, b);
+  }
+#pragma unroll
+  for (int s = P / 4; s >= 1; s >>= 1)
+#pragma unroll
+    for (int i = 0; i < s; ++i) {
+      v[i] = pk_max_u16(v[i], v[i + s]);
+      v[P / 2 + i] = pk_min_u16(v[P / 2 + i], v[P / 2 + s + i]);
+    }
+  mlo = v[0];
+  mhi = v[P / 2];
 }
 
 template <int P>
@@ -282,14 +342,35 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
       v[i] = pk_key_from_bf16(v[i]);
       if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
     }
+    // MEDIAN pads split low/high so the selection epilogue's fixed ranks
+    // P/2-1 / P/2 hit the true median (low-pad key 0 only ties a
+    // negative-NaN data key — NaN order statistics are unspecified in
+    // the full-sort path too)
+    const int n_lo = vecify(
+        MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
 #pragma unroll
     for (int i = 0; i < P; ++i)
       if (i >= nv) {
-        v[i] = 0xFFFFFFFFu;  // largest key: pads sort last
-        if (QUADS) v2[i] = 0xFFFFFFFFu;
+        const u32 pad = (MODE == MEDIAN && i < n_lo) ? 0u : 0xFFFFFFFFu;
+        v[i] = pad;
+        if (QUADS) v2[i] = pad;
       }
-    bitonic_sort_pk<P>(v);
-    if (QUADS) bitonic_sort_pk<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]));
+    u32 sel_lo[2] = {0u, 0u}, sel_hi[2] = {0u, 0u};
+    if (MODE == MEDIAN) {
+      // selection network: stop the bitonic sort one phase early (halves
+      // sorted asc/desc), then split + reduce — drops P/2*log2(P) CEs
+      // AND both O(P) predicated rank extractions from the hot path
+      bitonic_sort_pk<P, P / 2>(v);
+      pk_median_select<P>(v, sel_lo[0], sel_hi[0]);
+      if (QUADS) {
+        u32(&vv2)[P] = *reinterpret_cast<u32(*)[P]>(&v2[0]);
+        bitonic_sort_pk<P, P / 2>(vv2);
+        pk_median_select<P>(vv2, sel_lo[1], sel_hi[1]);
+      }
+    } else {
+      bitonic_sort_pk<P>(v);
+      if (QUADS) bitonic_sort_pk<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]));
+    }
     const int plo = vecify((n - 1) >> 1), phi = vecify(n >> 1);
     const int fv = vecify(f), nfv = vecify(n - f);
     u32* outw = reinterpret_cast<u32*>(out);
@@ -298,12 +379,8 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
       const u32* arr = half ? v2 : v;
       float m0, m1;
       if (MODE == MEDIAN) {
-        u32 lo = 0, hi = 0;
-#pragma unroll
-        for (int i = 0; i < P; ++i) {
-          if (i == plo) lo = arr[i];
-          if (i == phi) hi = arr[i];
-        }
+        const u32 lo = sel_lo[half];
+        const u32 hi = (n & 1) ? lo : sel_hi[half];
         m0 = 0.5f * (key_to_float(lo & 0xFFFFu) + key_to_float(hi & 0xFFFFu));
         m1 = 0.5f * (key_to_float(lo >> 16) + key_to_float(hi >> 16));
       } else if (MODE == TRIMMED) {
