@@ -228,12 +228,19 @@ DEV u32 pk_max_u16(u32 a, u32 b) {
   return r;
 }
 
-// bf16 bits -> sortable u16 key, two lanes at once (32-bit arithmetic:
-// per half key = bits ^ (sign ? 0xFFFF : 0x8000); the multiply cannot
-// carry across halves)
+// bf16 bits -> sortable u16 key, two lanes at once: per half
+// key = bits ^ (sign ? 0xFFFF : 0x8000). The packed arithmetic shift
+// broadcasts each half's sign to a 0x0000/0xFFFF mask in one op, so the
+// transform is 3 VALU (ashr, or, xor) for both columns — it runs once
+// per element on a VALU-bound kernel, so width matters.
 DEV u32 pk_key_from_bf16(u32 bits) {
-  const u32 s = (bits >> 15) & 0x00010001u;
-  return bits ^ (0x80008000u + s * 0x7FFFu);
+  u32 t;
+  // shift amount lives in a VGPR as {15, 15}: a bare inline constant
+  // would shift only the LOW half (VOP3P literals do not replicate)
+  asm("v_pk_ashrrev_i16 %0, %2, %1"
+      : "=v"(t)
+      : "v"(bits), "v"(0x000F000Fu));
+  return bits ^ (t | 0x80008000u);
 }
 
 DEV float key_to_float(u32 key16) {
