@@ -158,17 +158,20 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
       }
 #pragma unroll
       for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
-      const int nv = vecify(n);
-      // MEDIAN pads split: L low pads (-inf) shift the median ranks to
-      // exactly P/2-1 / P/2 for the selection epilogue; a -inf pad tying
-      // a -inf data value is value-identical, so the statistic is exact.
-      // Other modes keep all-high pads (the sorted prefix is the data).
-      const int n_lo = vecify(
-          MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
+      if (n < P) {  // wave-uniform: the full tile needs no pads at all
+        // MEDIAN pads split: L low pads (-inf) shift the median ranks to
+        // exactly P/2-1 / P/2 for the selection epilogue; a -inf pad
+        // tying a -inf data value is value-identical, so the statistic
+        // is exact. Other modes keep all-high pads (the sorted prefix is
+        // the data).
+        const int nv = vecify(n);
+        const int n_lo = vecify(
+            MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
 #pragma unroll
-      for (int i = 0; i < P; ++i)
-        if (i >= nv)
-          v[i] = (MODE == MEDIAN && i < n_lo) ? -PAD : PAD;
+        for (int i = 0; i < P; ++i)
+          if (i >= nv)
+            v[i] = (MODE == MEDIAN && i < n_lo) ? -PAD : PAD;
+      }
     }
 
     float result;
@@ -329,39 +332,58 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
     // phase for why (one live base, no SGPR spill storm). The walk
     // condition is vecified too — 64 uniform selects otherwise become
     // batched SGPR mask pairs and spill at QUADS register pressure.
-    const int n_walk = vecify(n);
     const u32* p = reinterpret_cast<const u32*>(X) + pair;
+    if (n == P) {
+      // full tile (the flagship n=64 shape): unconditional walk, no pads
+      // — drops ~190 predication VALU ops from a VALU-bound kernel
 #pragma unroll
-    for (int i = 0; i < P; ++i) {
-      if (QUADS) {
-        const uint2 w = *reinterpret_cast<const uint2*>(p);
-        v[i] = w.x;
-        v2[i] = w.y;
-      } else {
-        v[i] = *p;
+      for (int i = 0; i < P; ++i) {
+        if (QUADS) {
+          const uint2 w = *reinterpret_cast<const uint2*>(p);
+          v[i] = w.x;
+          v2[i] = w.y;
+        } else {
+          v[i] = *p;
+        }
+        if (i + 1 < P) p += rowstride;
+        __builtin_amdgcn_sched_barrier(0);
       }
-      p += (i + 1 < n_walk) ? rowstride : 0;
-      __builtin_amdgcn_sched_barrier(0);
+    } else {
+      const int n_walk = vecify(n);
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        if (QUADS) {
+          const uint2 w = *reinterpret_cast<const uint2*>(p);
+          v[i] = w.x;
+          v2[i] = w.y;
+        } else {
+          v[i] = *p;
+        }
+        p += (i + 1 < n_walk) ? rowstride : 0;
+        __builtin_amdgcn_sched_barrier(0);
+      }
     }
-    const int nv = vecify(n);
 #pragma unroll
     for (int i = 0; i < P; ++i) {
       v[i] = pk_key_from_bf16(v[i]);
       if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
     }
-    // MEDIAN pads split low/high so the selection epilogue's fixed ranks
-    // P/2-1 / P/2 hit the true median (low-pad key 0 only ties a
-    // negative-NaN data key — NaN order statistics are unspecified in
-    // the full-sort path too)
-    const int n_lo = vecify(
-        MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
+    if (n < P) {
+      // MEDIAN pads split low/high so the selection epilogue's fixed
+      // ranks P/2-1 / P/2 hit the true median (low-pad key 0 only ties a
+      // negative-NaN data key — NaN order statistics are unspecified in
+      // the full-sort path too)
+      const int nv = vecify(n);
+      const int n_lo = vecify(
+          MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
 #pragma unroll
-    for (int i = 0; i < P; ++i)
-      if (i >= nv) {
-        const u32 pad = (MODE == MEDIAN && i < n_lo) ? 0u : 0xFFFFFFFFu;
-        v[i] = pad;
-        if (QUADS) v2[i] = pad;
-      }
+      for (int i = 0; i < P; ++i)
+        if (i >= nv) {
+          const u32 pad = (MODE == MEDIAN && i < n_lo) ? 0u : 0xFFFFFFFFu;
+          v[i] = pad;
+          if (QUADS) v2[i] = pad;
+        }
+    }
     u32 sel_lo[2] = {0u, 0u}, sel_hi[2] = {0u, 0u};
     if (MODE == MEDIAN) {
       // selection network: stop the bitonic sort one phase early (halves
