@@ -368,12 +368,12 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
       v[i] = pk_key_from_bf16(v[i]);
       if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
     }
+    const int nv = vecify(n);  // MEAMED epilogue + pad loop both use it
     if (n < P) {
       // MEDIAN pads split low/high so the selection epilogue's fixed
       // ranks P/2-1 / P/2 hit the true median (low-pad key 0 only ties a
       // negative-NaN data key — NaN order statistics are unspecified in
       // the full-sort path too)
-      const int nv = vecify(n);
       const int n_lo = vecify(
           MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
 #pragma unroll
