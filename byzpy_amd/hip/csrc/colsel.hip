@@ -316,7 +316,9 @@ template <int P, bool QUADS, int MODE = MEDIAN>
 __global__ void
 // QUADS at P=64 holds two 64-u32 arrays (~150 VGPRs): ask for 3 waves/SIMD
 // so the allocator doesn't cap at 128 and spill
-__launch_bounds__(256, (QUADS && P >= 64) ? 3 : 4)
+__launch_bounds__(256, (QUADS && P >= 64) ? 3
+                       : (MODE == MEDIAN && P >= 64) ? 5
+                                                     : 4)
 colsel_pk_median_bf16(const unsigned short* __restrict__ X,
                                       unsigned short* __restrict__ out, int n,
                                       long d, int f) {

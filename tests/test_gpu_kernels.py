@@ -840,3 +840,29 @@ class TestGroupedOps:
         for i in range(3):
             zi = D.geometric_median(X3[i], fixed_iters=30)
             assert torch.allclose(Z[i], zi.float(), atol=1e-4), i
+
+
+@pytest.mark.gpu
+class TestMedianSelectionAllN:
+    """Exhaustive n sweep for the selection-network median (the pad-split
+    epilogue maps every n <= P onto fixed ranks P/2-1 / P/2 — each n
+    exercises a different pad count L, so cover them all)."""
+
+    def test_every_n_to_64(self):
+        g = torch.Generator().manual_seed(123)
+        for dtype in (torch.bfloat16, torch.float32):
+            for n in range(1, 65):
+                X = torch.randn(n, 4096, generator=g).to("cuda", dtype)
+                out = D.median(X)
+                ref = F.median(X.float().cpu()).to(dtype)
+                assert torch.equal(out.cpu(), ref), (dtype, n)
+
+    def test_every_n_with_infs(self):
+        g = torch.Generator().manual_seed(7)
+        for n in range(2, 65, 5):
+            X = torch.randn(n, 2048, generator=g).to("cuda", torch.bfloat16)
+            X[0] = float("inf")
+            X[n - 1] = float("-inf")
+            out = D.median(X)
+            ref = F.median(X.float().cpu()).to(torch.bfloat16)
+            assert torch.equal(out.cpu(), ref), n
