@@ -869,3 +869,22 @@ class TestMedianSelectionAllN:
             assert torch.allclose(
                 out.float().cpu(), ref.float(), atol=0, rtol=0, equal_nan=True
             ), n
+
+
+
+@pytest.mark.gpu
+class TestCafPipelineParity:
+    """The CAF eager loop pipelines its break-condition readback one round
+    deep and stages seeds through pinned memory; BYZPY_CAF_SYNC=1 recovers
+    the fully synchronous loop. The two must be BITWISE equal — the
+    pipeline only moves host waits, never changes a computed value."""
+
+    def test_sync_equals_pipelined(self, monkeypatch):
+        g = torch.Generator().manual_seed(99)
+        for (n, d, f) in [(64, 65536, 16), (33, 10000, 9), (128, 4096, 40)]:
+            X = torch.randn(n, d, generator=g).to("cuda")
+            monkeypatch.setenv("BYZPY_CAF_SYNC", "1")
+            ref = D.caf(X, f)
+            monkeypatch.delenv("BYZPY_CAF_SYNC")
+            out = D.caf(X, f)
+            assert torch.equal(out, ref), (n, d, f)
