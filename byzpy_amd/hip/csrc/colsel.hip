@@ -116,6 +116,29 @@ DEV void median_select_reg(float (&v)[P], float& mlo, float& mhi) {
   mhi = v[P / 2];
 }
 
+// standalone half-sort (see bitonic_sort_pk_half for why source order
+// matters): UP=true ascending, UP=false the mirrored descending network
+template <int P2, bool UP>
+DEV void bitonic_sort_reg_half(float (&v)[P2]) {
+#pragma unroll
+  for (int k = 2; k <= P2; k <<= 1) {
+#pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+#pragma unroll
+      for (int i = 0; i < P2; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = ((i & k) == 0) == UP;
+          const float a = v[i], b = v[l];
+          const float lo = fminf(a, b), hi = fmaxf(a, b);
+          v[i] = asc ? lo : hi;
+          v[l] = asc ? hi : lo;
+        }
+      }
+    }
+  }
+}
+
 template <int P>
 DEV float extract_at(const float (&v)[P], int pos) {
   float r = 0.0f;
@@ -139,6 +162,7 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long col = col0; col < d; col += stride) {
     float v[P];
+    T raw[P];
     // Load phase. Two rules keep the address walk at ONE live base:
     // (1) raw loads only — any data-dependent convert inside the walk
     //     would force a wait per element; conversion happens after;
@@ -148,7 +172,6 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
     // Once a load ISSUES its address registers are free; the loads stay
     // in flight and the converts below wait on counted vmcnt.
     {
-      T raw[P];
       const T* p = X + col;
 #pragma unroll
       for (int i = 0; i < P; ++i) {
@@ -156,31 +179,45 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
         if (i + 1 < n) p += d;
         __builtin_amdgcn_sched_barrier(0);
       }
-#pragma unroll
-      for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
-      if (n < P) {  // wave-uniform: the full tile needs no pads at all
-        // MEDIAN pads split: L low pads (-inf) shift the median ranks to
-        // exactly P/2-1 / P/2 for the selection epilogue; a -inf pad
-        // tying a -inf data value is value-identical, so the statistic
-        // is exact. Other modes keep all-high pads (the sorted prefix is
-        // the data).
-        const int nv = vecify(n);
-        const int n_lo = vecify(
-            MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
-#pragma unroll
-        for (int i = 0; i < P; ++i)
-          if (i >= nv)
-            v[i] = (MODE == MEDIAN && i < n_lo) ? -PAD : PAD;
-      }
     }
+    const int nv = vecify(n);
 
     float result;
     if (MODE == MEDIAN) {
-      bitonic_sort_reg<P, P / 2>(v);
+      // selection network, half at a time IN SOURCE ORDER (see the pk
+      // kernel): half A's conversion+sort consume only the first P/2
+      // loads and overlap the tail loads. MEDIAN pads split: L low pads
+      // (-inf) shift the median ranks to exactly P/2-1 / P/2; a -inf
+      // pad tying a -inf data value is value-identical.
+      const int n_lo = vecify(n + (P / 2 - 1 - ((n - 1) >> 1)));
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int base = h * (P / 2);
+#pragma unroll
+        for (int i = base; i < base + P / 2; ++i) v[i] = to_f<T>(raw[i]);
+        if (n < P) {
+#pragma unroll
+          for (int i = base; i < base + P / 2; ++i)
+            if (i >= nv) v[i] = (i < n_lo) ? -PAD : PAD;
+        }
+        if (h == 0)
+          bitonic_sort_reg_half<P / 2, true>(
+              *reinterpret_cast<float(*)[P / 2]>(&v[0]));
+        else
+          bitonic_sort_reg_half<P / 2, false>(
+              *reinterpret_cast<float(*)[P / 2]>(&v[P / 2]));
+      }
       float mlo, mhi;
       median_select_reg<P>(v, mlo, mhi);
       result = (n & 1) ? mlo : 0.5f * (mlo + mhi);
     } else if (MODE == TRIMMED) {
+#pragma unroll
+      for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
+      if (n < P) {
+#pragma unroll
+        for (int i = 0; i < P; ++i)
+          if (i >= nv) v[i] = PAD;
+      }
       bitonic_sort_reg<P>(v);
       const int fv = vecify(f), nfv = vecify(n - f);
       float s = 0.0f;
@@ -189,6 +226,13 @@ colsel_reg_kernel(const T* __restrict__ X, T* __restrict__ out,
         if (i >= fv && i < nfv) s += v[i];
       result = s / (float)(n - 2 * f);
     } else {  // MEAMED: mean of the n-f values closest to the median
+#pragma unroll
+      for (int i = 0; i < P; ++i) v[i] = to_f<T>(raw[i]);
+      if (n < P) {
+#pragma unroll
+        for (int i = 0; i < P; ++i)
+          if (i >= nv) v[i] = PAD;
+      }
       bitonic_sort_reg<P>(v);
       const int pos_lo = vecify((n - 1) >> 1), pos_hi = vecify(n >> 1);
       const float med =
@@ -266,6 +310,33 @@ DEV void bitonic_sort_pk(u32 (&v)[P]) {
         const int l = i ^ j;
         if (l > i) {
           const bool asc = (i & k) == 0;
+          const u32 a = v[i], b = v[l];
+          const u32 lo = pk_min_u16(a, b), hi = pk_max_u16(a, b);
+          v[i] = asc ? lo : hi;
+          v[l] = asc ? hi : lo;
+        }
+      }
+    }
+  }
+}
+
+// standalone half-sort for the median selection path: UP=true ascending,
+// UP=false the mirrored (descending) network. Sorting a half touches only
+// its own P2 registers, so in source order the FIRST half's 240 CEs issue
+// while the second half's row loads are still in flight (the shared
+// <P, P/2> network's k=2 phase touches all P loads up front and forces a
+// full vmcnt(0) wait before any compute).
+template <int P2, bool UP>
+DEV void bitonic_sort_pk_half(u32 (&v)[P2]) {
+#pragma unroll
+  for (int k = 2; k <= P2; k <<= 1) {
+#pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+#pragma unroll
+      for (int i = 0; i < P2; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool asc = ((i & k) == 0) == UP;
           const u32 a = v[i], b = v[l];
           const u32 lo = pk_min_u16(a, b), hi = pk_max_u16(a, b);
           v[i] = asc ? lo : hi;
@@ -365,40 +436,67 @@ colsel_pk_median_bf16(const unsigned short* __restrict__ X,
         __builtin_amdgcn_sched_barrier(0);
       }
     }
-#pragma unroll
-    for (int i = 0; i < P; ++i) {
-      v[i] = pk_key_from_bf16(v[i]);
-      if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
-    }
-    const int nv = vecify(n);  // MEAMED epilogue + pad loop both use it
-    if (n < P) {
-      // MEDIAN pads split low/high so the selection epilogue's fixed
-      // ranks P/2-1 / P/2 hit the true median (low-pad key 0 only ties a
-      // negative-NaN data key — NaN order statistics are unspecified in
-      // the full-sort path too)
-      const int n_lo = vecify(
-          MODE == MEDIAN ? n + (P / 2 - 1 - ((n - 1) >> 1)) : 0);
-#pragma unroll
-      for (int i = 0; i < P; ++i)
-        if (i >= nv) {
-          const u32 pad = (MODE == MEDIAN && i < n_lo) ? 0u : 0xFFFFFFFFu;
-          v[i] = pad;
-          if (QUADS) v2[i] = pad;
-        }
-    }
+    const int nv = vecify(n);  // conversion/pad loops + MEAMED epilogue
     u32 sel_lo[2] = {0u, 0u}, sel_hi[2] = {0u, 0u};
     if (MODE == MEDIAN) {
-      // selection network: stop the bitonic sort one phase early (halves
-      // sorted asc/desc), then split + reduce — drops P/2*log2(P) CEs
-      // AND both O(P) predicated rank extractions from the hot path
-      bitonic_sort_pk<P, P / 2>(v);
-      pk_median_select<P>(v, sel_lo[0], sel_hi[0]);
-      if (QUADS) {
-        u32(&vv2)[P] = *reinterpret_cast<u32(*)[P]>(&v2[0]);
-        bitonic_sort_pk<P, P / 2>(vv2);
-        pk_median_select<P>(vv2, sel_lo[1], sel_hi[1]);
+      // selection network, half at a time IN SOURCE ORDER: converting and
+      // sorting [0,P/2) consumes only the first P/2 row loads, so those
+      // 240 CEs overlap the in-flight tail loads; then the (mirrored
+      // descending) upper half, one split pass and two reductions.
+      // MEDIAN pads split low/high so the fixed ranks P/2-1 / P/2 hit
+      // the true median (low-pad key 0 only ties a negative-NaN data
+      // key — NaN order statistics are unspecified in the full-sort
+      // path too).
+      const int n_lo = vecify(n + (P / 2 - 1 - ((n - 1) >> 1)));
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int base = h * (P / 2);
+#pragma unroll
+        for (int i = base; i < base + P / 2; ++i) {
+          v[i] = pk_key_from_bf16(v[i]);
+          if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
+        }
+        if (n < P) {
+#pragma unroll
+          for (int i = base; i < base + P / 2; ++i)
+            if (i >= nv) {
+              const u32 pad = (i < n_lo) ? 0u : 0xFFFFFFFFu;
+              v[i] = pad;
+              if (QUADS) v2[i] = pad;
+            }
+        }
+        if (h == 0) {
+          bitonic_sort_pk_half<P / 2, true>(
+              *reinterpret_cast<u32(*)[P / 2]>(&v[0]));
+          if (QUADS)
+            bitonic_sort_pk_half<P / 2, true>(
+                *reinterpret_cast<u32(*)[P / 2]>(&v2[0]));
+        } else {
+          bitonic_sort_pk_half<P / 2, false>(
+              *reinterpret_cast<u32(*)[P / 2]>(&v[P / 2]));
+          if (QUADS)
+            bitonic_sort_pk_half<P / 2, false>(
+                *reinterpret_cast<u32(*)[P / 2]>(&v2[P / 2]));
+        }
       }
+      pk_median_select<P>(v, sel_lo[0], sel_hi[0]);
+      if (QUADS)
+        pk_median_select<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]),
+                            sel_lo[1], sel_hi[1]);
     } else {
+#pragma unroll
+      for (int i = 0; i < P; ++i) {
+        v[i] = pk_key_from_bf16(v[i]);
+        if (QUADS) v2[i] = pk_key_from_bf16(v2[i]);
+      }
+      if (n < P) {
+#pragma unroll
+        for (int i = 0; i < P; ++i)
+          if (i >= nv) {
+            v[i] = 0xFFFFFFFFu;
+            if (QUADS) v2[i] = 0xFFFFFFFFu;
+          }
+      }
       bitonic_sort_pk<P>(v);
       if (QUADS) bitonic_sort_pk<P>(*reinterpret_cast<u32(*)[P]>(&v2[0]));
     }
