@@ -274,3 +274,104 @@ def test_rccl_p2p_fixed_iters_geomed_on_device():
     out = p2p.round()
     assert out.is_cuda
     assert torch.allclose(out.float(), torch.full((30_000,), 3.0, device="cuda"), atol=1e-2)
+
+
+class GpuRandHonest:
+    def __init__(self, seed: int, d: int = 8192):
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        self.grad = (
+            torch.randn(d, generator=g).to("cuda", torch.bfloat16)
+            + float(seed % 3)
+        )
+
+    def honest_gradient_for_next_batch(self):
+        return self.grad
+
+    def apply_server_gradient(self, g):
+        pass
+
+
+class GpuRandByz:
+    def __init__(self, attack, d: int = 8192):
+        self.attack = attack
+        self.d = d
+
+    def byzantine_gradient_for_next_batch(self, honest_grads=None):
+        kwargs = {}
+        if getattr(self.attack, "uses_honest_grads", False):
+            kwargs["honest_grads"] = honest_grads
+        if getattr(self.attack, "uses_base_grad", False):
+            kwargs["base_grad"] = honest_grads[0]
+        return self.attack.apply(**kwargs).to("cuda", torch.bfloat16)
+
+    def apply_server_gradient(self, g):
+        pass
+
+
+def test_ps_engine_fuzz_on_device():
+    """Seeded engine-level fuzz fully on device: random aggregator x
+    attack combinations through the real ParameterServer + stream actors.
+    Output must be finite and bounded by the honest scale (the robust
+    aggregators' whole contract) for every pairing."""
+    from byzpy_amd.aggregators import (
+        CAF,
+        CoordinateWiseMedian,
+        ComparativeGradientElimination,
+        GeometricMedian,
+        MeanOfMedians,
+    )
+    from byzpy_amd.attacks import (
+        EmpireAttack,
+        GaussianAttack,
+        InfAttack,
+        LittleAttack,
+        SignFlipAttack,
+    )
+
+    aggs = [
+        CoordinateWiseMedian(),
+        CoordinateWiseTrimmedMean(f=2),
+        MultiKrum(f=2, q=3),
+        MeanOfMedians(f=2),
+        ComparativeGradientElimination(f=2),
+        GeometricMedian(max_iter=32),
+        CAF(f=2),
+    ]
+    attacks = [
+        SignFlipAttack(),
+        EmpireAttack(scale=-8.0),
+        LittleAttack(f=2, N=9),
+        GaussianAttack(0.0, 30.0),
+        InfAttack(),
+    ]
+
+    async def one(agg, atk, seed):
+        honest = [
+            await HonestNodeActor.spawn(GpuRandHonest, seed * 31 + i, backend="stream:0")
+            for i in range(7)
+        ]
+        byz = [
+            await ByzantineNodeActor.spawn(GpuRandByz, atk, backend="stream:0")
+            for _ in range(2)
+        ]
+        ps = ParameterServer(honest, byz, agg)
+        update = await ps.round()
+        assert update.is_cuda, (agg, atk)
+        uf = update.float()
+        if not isinstance(atk, InfAttack) or not isinstance(
+            agg, GeometricMedian
+        ):
+            assert torch.isfinite(uf).all(), (agg, atk)
+            assert float(uf.abs().max()) < 50.0, (agg, atk)
+        for a in honest + byz:
+            await a.close()
+
+    async def main():
+        for i, agg in enumerate(aggs):
+            atk = attacks[i % len(attacks)]
+            await one(agg, atk, i)
+        # and the historically nastiest pairing: inf rows into order stats
+        await one(CoordinateWiseTrimmedMean(f=2), InfAttack(), 99)
+        await one(CoordinateWiseMedian(), InfAttack(), 100)
+
+    asyncio.run(main())
