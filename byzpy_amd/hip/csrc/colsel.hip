@@ -18,6 +18,7 @@
 // Loads are coalesced: adjacent lanes read adjacent columns, so each
 // row-iteration is one 256 B (f32) / 128 B (bf16) wave transaction.
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -735,14 +736,27 @@ void launch_colsel_bf16(const __hip_bfloat16* X, __hip_bfloat16* out, int n,
                         long d, int mode, int f, hipStream_t stream) {
   if (n <= 64 && (d % 2) == 0) {
     const int block = 256;
-    // A/B'd on MI355X: QUADS (8 B/lane) ties the single-pair variant
-    // (3.9 vs 3.8 ms at 64 x 125M) — the occupancy drop offsets the wider
-    // bursts. Keep single-pair as the shipping path.
+    // A/B'd on MI355X: QUADS (8 B/lane) TIED the old full-sort variant
+    // (3.9 vs 3.8 ms at 64 x 125M) and LOSES 3x to the selection-network
+    // pair variant (9.09 vs 3.09 ms — two 64-reg key arrays force the
+    // 3-waves/SIMD bound while the pair kernel runs at 5). Single-pair
+    // is the shipping path; BYZPY_PK_QUADS=1 re-runs the comparison.
     const long units = d >> 1;
     const long want = (units + block - 1) / block;
     const int grid = (int)(want < 8192 ? (want > 0 ? want : 1) : 8192);
     const unsigned short* Xu = reinterpret_cast<const unsigned short*>(X);
     unsigned short* Ou = reinterpret_cast<unsigned short*>(out);
+    // A/B aid: BYZPY_PK_QUADS=1 re-measures the 4-column (8 B/lane)
+    // MEDIAN variant (tied pre-selection-network; see the comment above)
+    if (mode == MEDIAN && n > 32 && (d % 4) == 0 &&
+        std::getenv("BYZPY_PK_QUADS") != nullptr) {
+      const long qunits = d >> 2;
+      const long qwant = (qunits + block - 1) / block;
+      const int qgrid = (int)(qwant < 8192 ? (qwant > 0 ? qwant : 1) : 8192);
+      hipLaunchKernelGGL((colsel_pk_median_bf16<64, true, MEDIAN>),
+                         dim3(qgrid), dim3(block), 0, stream, Xu, Ou, n, d, f);
+      return;
+    }
 #define PK_LAUNCH(P)                                                          \
   do {                                                                        \
     if (mode == MEDIAN)                                                       \
