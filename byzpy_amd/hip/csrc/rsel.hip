@@ -95,6 +95,18 @@ DEV float val_from_key(u32 key) {
   return (KK == VAL_BF16) ? bf16val_from_key(key) : f32_from_key(key);
 }
 
+// bf16 value key in the 16-bit domain (no <<16): the level loop's
+// per-element fast path — the u32 state convention is translated once
+// per column instead (see rsel_level_kernel)
+template <typename T>
+DEV u32 make_key16(T raw) {
+  union { T t; unsigned short s; } c;
+  c.t = raw;
+  const u32 bits = (u32)c.s;
+  const u32 sg = (bits >> 15) & 1u;
+  return (bits ^ (0x8000u + sg * 0x7FFFu)) & 0xFFFFu;
+}
+
 // -- level pass -------------------------------------------------------------
 
 // TWO=true tracks two ranks with a rank-packed 16/16 histogram; TWO=false
@@ -134,6 +146,20 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
     const u32 one = TWO ? 1u : ((c & 1) ? 0x10000u : 1u);
     const int w = TWO ? c : (c >> 1);
     const T* xc = X + col;
+    // The streaming loop is VALU-bound (PMC: 17.4 VALU/element at 64
+    // bins before this rework), so per-element arithmetic is hoisted
+    // per COLUMN wherever possible:
+    //  - the prefix compare ((key ^ p) & hi_mask) == 0 becomes
+    //    (key & hi_mask) == qp with qp = p & hi_mask precomputed, and
+    //    the AND is shared between both rank trackers;
+    //  - bf16 value keys live in the 16-bit domain (the u16 key is
+    //    produced WITHOUT the <<16 normalization; shift/masks/prefixes
+    //    are moved down 16 once per column instead).
+    constexpr bool K16 = (KK == VAL_BF16);
+    const int sh = K16 ? shift - 16 : shift;
+    const u32 hm = K16 ? (hi_mask >> 16) : hi_mask;
+    const u32 q0 = (K16 ? (p0 >> 16) : p0) & hm;
+    const u32 q1 = TWO ? (((K16 ? (p1 >> 16) : p1)) & hm) : 0;
     // 8 loads in flight per slice thread; UNCONDITIONAL zero-capable
     // atomics (an `if (inc)` guard emits per-element exec save/restore)
     int row = slice;
@@ -143,29 +169,32 @@ rsel_level_kernel(const T* __restrict__ X, const float* __restrict__ med,
       for (int j = 0; j < 8; ++j) raw[j] = xc[(long)(row + SLICES * j) * d];
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        const u32 key = make_key<T, KK>(raw[q], m);
-        const bool m0 = ((key ^ p0) & hi_mask) == 0;
+        const u32 key = K16 ? make_key16<T>(raw[q]) : make_key<T, KK>(raw[q], m);
+        const u32 kh = key & hm;
+        const bool m0 = kh == q0;
         u32 inc;
         if (TWO) {
-          const bool m1 = ((key ^ p1) & hi_mask) == 0;
+          const bool m1 = kh == q1;
           inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
         } else {
           inc = m0 ? one : 0u;
         }
-        atomicAdd(&cnt[(key >> shift) & (RS_BINS - 1)][w], inc);
+        atomicAdd(&cnt[(key >> sh) & (RS_BINS - 1)][w], inc);
       }
     }
     for (; row < n; row += SLICES) {
-      const u32 key = make_key<T, KK>(xc[(long)row * d], m);
-      const bool m0 = ((key ^ p0) & hi_mask) == 0;
+      const u32 key = K16 ? make_key16<T>(xc[(long)row * d])
+                          : make_key<T, KK>(xc[(long)row * d], m);
+      const u32 kh = key & hm;
+      const bool m0 = kh == q0;
       u32 inc;
       if (TWO) {
-        const bool m1 = ((key ^ p1) & hi_mask) == 0;
+        const bool m1 = kh == q1;
         inc = (m0 ? 1u : 0u) | (m1 ? 0x10000u : 0u);
       } else {
         inc = m0 ? one : 0u;
       }
-      atomicAdd(&cnt[(key >> shift) & (RS_BINS - 1)][w], inc);
+      atomicAdd(&cnt[(key >> sh) & (RS_BINS - 1)][w], inc);
     }
   }
   __syncthreads();
