@@ -375,3 +375,41 @@ def test_ps_engine_fuzz_on_device():
         await one(CoordinateWiseMedian(), InfAttack(), 100)
 
     asyncio.run(main())
+
+
+def test_ps_long_run_no_leak_no_fade():
+    """200 rounds through the PS on stream actors: device memory must be
+    flat (no per-round allocation leak from the engine/actor plumbing)
+    and late rounds must not be slower than early ones (no queue/handle
+    accumulation)."""
+    import time
+
+    async def main():
+        honest = [
+            await HonestNodeActor.spawn(GpuRandHonest, i, backend="stream:0")
+            for i in range(6)
+        ]
+        byz = [await ByzantineNodeActor.spawn(GpuByz, 8192, backend="stream:0")]
+        ps = ParameterServer(honest, byz, CoordinateWiseTrimmedMean(f=1))
+        for _ in range(10):  # warm the caches/allocator
+            await ps.round()
+        torch.cuda.synchronize()
+        mem0 = torch.cuda.memory_allocated()
+        t0 = time.perf_counter()
+        for _ in range(95):
+            await ps.round()
+        torch.cuda.synchronize()
+        first = time.perf_counter() - t0
+        mem1 = torch.cuda.memory_allocated()
+        t0 = time.perf_counter()
+        for _ in range(95):
+            await ps.round()
+        torch.cuda.synchronize()
+        second = time.perf_counter() - t0
+        mem2 = torch.cuda.memory_allocated()
+        assert mem1 == mem0 and mem2 == mem0, (mem0, mem1, mem2)
+        assert second < first * 1.5, (first, second)
+        for a in honest + byz:
+            await a.close()
+
+    asyncio.run(main())
