@@ -30,6 +30,11 @@ class _P2PNodeShell(DecentralizedNode):
         self._round_conds: Dict[int, asyncio.Condition] = {}
         self._expected = 0
         self.register_handler("gradient", self._on_gradient)
+        # gossip shells consume every message through the handler above;
+        # the scheduler's mailbox mirror would otherwise retain one
+        # broadcast payload per sender per round (caught by the device
+        # long-run soak — see MessageAwareNodeScheduler.cache_limit)
+        self.scheduler.cache_limit = 0
 
     def set_expected(self, count: int) -> None:
         self._expected = count

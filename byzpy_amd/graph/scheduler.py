@@ -68,12 +68,23 @@ class NodeScheduler:
 
 class MessageAwareNodeScheduler(NodeScheduler):
     """Adds deliver_message/wait_for_message with a cache so messages that
-    arrive before the consumer are not lost."""
+    arrive before the consumer are not lost.
 
-    def __init__(self, graph: ComputationGraph, pool: Any = None, metadata: Optional[dict] = None) -> None:
+    The per-type cache is a BOUNDED FIFO (``cache_limit``, default 1024):
+    every incoming node message is mirrored here whether or not any graph
+    ever consumes that type, so an unbounded list is a leak — in the P2P
+    gossip engine it retained one broadcast payload per sender per round
+    forever (caught by the device long-run soak). When the cap is hit the
+    OLDEST unconsumed message is dropped, which matches the consumer's
+    pop-oldest order: a reader that far behind has already lost ordering
+    guarantees."""
+
+    def __init__(self, graph: ComputationGraph, pool: Any = None, metadata: Optional[dict] = None,
+                 cache_limit: int = 1024) -> None:
         super().__init__(graph, pool, metadata)
         self._cache: Dict[str, List[Any]] = {}
         self._waiters: Dict[str, List[asyncio.Future]] = {}
+        self.cache_limit = int(cache_limit)
 
     def deliver_message(self, message_type: str, payload: Any) -> None:
         # skip and prune futures that are already done (e.g. a wait_for
@@ -84,7 +95,10 @@ class MessageAwareNodeScheduler(NodeScheduler):
             if not fut.done():
                 fut.set_result(payload)
                 return
-        self._cache.setdefault(message_type, []).append(payload)
+        cache = self._cache.setdefault(message_type, [])
+        cache.append(payload)
+        if len(cache) > self.cache_limit:
+            del cache[0]
 
     async def wait_for_message(self, message_type: str, timeout: Optional[float] = None) -> Any:
         cached = self._cache.get(message_type)

@@ -210,3 +210,50 @@ def test_rccl_p2p_rejects_inconsistent_byzantine_set():
             CoordinateWiseMedian(),
             byzantine_ranks=[0],  # this (only) rank has no attack
         )
+
+
+def test_p2p_rounds_do_not_retain_tensors():
+    """Regression for a real leak: the scheduler's message-cache mirror
+    retained one broadcast payload per sender per round, forever (gossip
+    shells consume via handlers; the mailbox was never drained). Tensor
+    census must be flat across rounds."""
+    import gc
+
+    import torch.nn as nn
+
+    from byzpy_amd.attacks import SignFlipAttack
+
+    class H(P2PHonestMixin):
+        def __init__(self, seed):
+            torch.manual_seed(seed)
+            self.model = nn.Linear(16, 1, bias=False)
+            self.lr = 0.05
+            g = torch.Generator().manual_seed(seed)
+            self.x = torch.randn(8, 16, generator=g)
+            self.y = self.x.sum(dim=1, keepdim=True) * 0.1
+
+        def p2p_local_loss_backward(self):
+            loss = ((self.model(self.x) - self.y) ** 2).mean()
+            loss.backward()
+
+    class B(P2PByzantineMixin):
+        def __init__(self):
+            self.attack = SignFlipAttack()
+
+    def census():
+        gc.collect()
+        return sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+
+    async def main():
+        honest = [H(s) for s in (1, 2, 3)]
+        p2p = PeerToPeer(honest, [B()], CoordinateWiseMedian(), lr=0.05)
+        for _ in range(5):
+            await p2p.round()
+        c0 = census()
+        for _ in range(25):
+            await p2p.round()
+        c1 = census()
+        assert c1 <= c0, f"tensor census grew {c0} -> {c1} over 25 rounds"
+        await p2p.shutdown()
+
+    asyncio.run(main())

@@ -413,3 +413,53 @@ def test_ps_long_run_no_leak_no_fade():
             await a.close()
 
     asyncio.run(main())
+
+
+def test_p2p_long_run_no_leak():
+    """100 gossip rounds fully on device: flat memory, finite models —
+    the P2P twin of the PS soak above."""
+    import torch.nn as nn
+
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+    from byzpy_amd.engine.peer_to_peer.mixin import (
+        P2PByzantineMixin,
+        P2PHonestMixin,
+    )
+    from byzpy_amd.engine.peer_to_peer.train import PeerToPeer
+
+    class DevHonest2(P2PHonestMixin):
+        def __init__(self, seed):
+            torch.manual_seed(seed)
+            self.model = nn.Linear(64, 1, bias=False).cuda()
+            self.lr = 0.02
+            g = torch.Generator().manual_seed(seed)
+            self.x = torch.randn(32, 64, generator=g).cuda()
+            self.y = self.x.sum(dim=1, keepdim=True) * 0.1
+
+        def p2p_local_loss_backward(self):
+            loss = ((self.model(self.x) - self.y) ** 2).mean()
+            loss.backward()
+
+    class DevByz2(P2PByzantineMixin):
+        def __init__(self):
+            from byzpy_amd.attacks import SignFlipAttack
+
+            self.attack = SignFlipAttack()
+
+    async def main():
+        honest = [DevHonest2(s) for s in (1, 2, 3, 4)]
+        byz = [DevByz2()]
+        p2p = PeerToPeer(honest, byz, CoordinateWiseMedian(), lr=0.02)
+        for _ in range(10):
+            await p2p.round()
+        torch.cuda.synchronize()
+        mem0 = torch.cuda.memory_allocated()
+        for _ in range(90):
+            await p2p.round()
+        torch.cuda.synchronize()
+        assert torch.cuda.memory_allocated() == mem0
+        for h in honest:
+            assert torch.isfinite(h.model.weight).all()
+        await p2p.shutdown()
+
+    asyncio.run(main())
