@@ -150,3 +150,42 @@ def test_process_chan_get_before_put_no_deadlock():
 
     out = asyncio.run(main())
     assert out["x"] == 7
+
+
+def test_actor_spawn_close_churn_no_leak():
+    """Spawn and close many actors: thread/process handles, queues and
+    tensor references must not accumulate across lifecycles."""
+    import gc
+    import threading
+
+    import torch
+
+    from byzpy_amd.engine.node.actors import HonestNodeActor
+
+    class W:
+        def __init__(self):
+            self.g = torch.ones(128)
+
+        def honest_gradient_for_next_batch(self):
+            return self.g
+
+        def apply_server_gradient(self, g):
+            pass
+
+    async def churn(n):
+        for _ in range(n):
+            a = await HonestNodeActor.spawn(W, backend="thread")
+            out = await a.honest_gradient_for_next_batch()
+            assert out.shape == (128,)
+            await a.close()
+
+    asyncio.run(churn(5))
+    gc.collect()
+    threads0 = threading.active_count()
+    c0 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    asyncio.run(churn(25))
+    gc.collect()
+    threads1 = threading.active_count()
+    c1 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    assert threads1 <= threads0 + 1, (threads0, threads1)
+    assert c1 <= c0 + 4, (c0, c1)

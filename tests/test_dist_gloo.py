@@ -382,3 +382,32 @@ def test_ring_p2p_byzantine_context():
 
 def test_ring_p2p_honest():
     _run_workers_n(_body_ring_honest_only, 4)
+
+
+def _body_rccl_ps_census(rank):
+    """50 rounds through the RCCL PS: tensor census must be flat (the
+    collective path must not retain per-round buffers)."""
+    import gc
+
+    from byzpy_amd.engine.parameter_server.rccl import (
+        RcclParameterServer,
+        trimmed_mean_aggregate,
+    )
+
+    d = 256
+    fns = [lambda: torch.randn(d, generator=torch.Generator().manual_seed(1)),
+           lambda: torch.full((d,), float(rank))]
+    ps = RcclParameterServer(fns, trimmed_mean_aggregate(1), gather_result=True)
+    for _ in range(5):
+        ps.round()
+    gc.collect()
+    c0 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    for _ in range(45):
+        ps.round()
+    gc.collect()
+    c1 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    assert c1 <= c0 + 4, (c0, c1)
+
+
+def test_rccl_ps_no_tensor_retention():
+    _run_workers(_body_rccl_ps_census)
