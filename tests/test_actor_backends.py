@@ -189,3 +189,36 @@ def test_actor_spawn_close_churn_no_leak():
     c1 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
     assert threads1 <= threads0 + 1, (threads0, threads1)
     assert c1 <= c0 + 4, (c0, c1)
+
+
+def test_process_actor_churn_no_zombies():
+    """Spawn/close process-backend actors repeatedly: child processes
+    must actually exit (no zombie/orphan accumulation)."""
+    import multiprocessing as mp
+
+    import torch
+
+    from byzpy_amd.engine.node.actors import HonestNodeActor
+
+    class W:
+        def __init__(self):
+            self.g = torch.ones(64)
+
+        def honest_gradient_for_next_batch(self):
+            return self.g
+
+        def apply_server_gradient(self, g):
+            pass
+
+    async def churn(n):
+        for _ in range(n):
+            a = await HonestNodeActor.spawn(W, backend="process")
+            out = await a.honest_gradient_for_next_batch()
+            assert out.shape == (64,)
+            await a.close()
+
+    asyncio.run(churn(2))
+    base = len(mp.active_children())
+    asyncio.run(churn(6))
+    leftover = len(mp.active_children())
+    assert leftover <= base, (base, leftover)
