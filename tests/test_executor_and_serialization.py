@@ -185,3 +185,30 @@ class TestAttackExecutorSweep:
 
         out = asyncio.run(run())
         assert torch.allclose(out, -1.5 * base)
+
+
+def test_run_operator_repeated_no_retention():
+    """200 run_operator calls through the cached executor/pool path:
+    tensor census flat (the graph/scheduler caches must be bounded)."""
+    import gc
+
+    import torch
+
+    from byzpy_amd import run_operator
+    from byzpy_amd.aggregators import CoordinateWiseMedian
+
+    agg = CoordinateWiseMedian()
+    X = [torch.randn(4, 64) for _ in range(3)]
+
+    def once():
+        for i in range(10):
+            run_operator(agg, {"gradients": list(X[i % 3])})
+
+    once()
+    gc.collect()
+    c0 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    for _ in range(19):
+        once()
+    gc.collect()
+    c1 = sum(1 for o in gc.get_objects() if torch.is_tensor(o))
+    assert c1 <= c0 + 8, (c0, c1)
