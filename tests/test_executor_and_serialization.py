@@ -201,8 +201,12 @@ def test_run_operator_repeated_no_retention():
     X = [torch.randn(4, 64) for _ in range(3)]
 
     def once():
-        for i in range(10):
-            run_operator(agg, {"gradients": list(X[i % 3])})
+        async def batch():
+            for i in range(10):
+                out = await run_operator(agg, {"gradients": list(X[i % 3])})
+                assert out.shape == (64,)
+
+        asyncio.run(batch())
 
     once()
     gc.collect()
