@@ -118,6 +118,8 @@ class RemoteActorServer:
                 try:
                     result = await self._handle(msg)
                     await send_obj(writer, ("ok", result))
+                except asyncio.CancelledError:
+                    raise  # server shutdown: never mask cancellation
                 except BaseException as e:  # noqa: BLE001
                     await send_obj(writer, ("err", repr(e)))
         finally:

@@ -97,6 +97,8 @@ def _process_node_main(node_blob: bytes, node_id: str, cmd_q, out_q) -> None:
                         try:
                             result = await node.execute_pipeline(name, inputs)
                             out_q.put(cloudpickle.dumps(("result", (req_id, result))))
+                        except asyncio.CancelledError:
+                            raise  # loop shutdown: never mask cancellation
                         except BaseException as e:  # noqa: BLE001
                             out_q.put(cloudpickle.dumps(("error", (req_id, repr(e)))))
 
