@@ -232,21 +232,49 @@ def geometric_median(
         for _ in range(int(fixed_iters)):
             z = ext.weiszfeld_iter(Xc, z, float(eps), shift)
         return z.to(X.dtype)
-    # fused per-iteration kernel pair; convergence polled every `poll` iters
-    # to avoid a host sync per iteration (SURVEY.md §7 hard part 2)
+    # fused per-iteration kernel pair; convergence polled every `poll`
+    # iters, with the poll readback PIPELINED one block deep (same
+    # pattern as the CAF loop): block b+1's kernels launch while block
+    # b's pinned copy of `shift` is in flight, so the host never stalls
+    # on a sync — value-identical, speculative blocks past the break
+    # point are discarded (each iteration returns a fresh z tensor, so
+    # the pre-break z is simply kept by reference).
+    # `shift` holds the last iteration's SQUARED step ||dz||^2 (rowops.hip
+    # weiszfeld_iter resets + atomicAdds it), so compare against tol^2 to
+    # match the CPU oracle's ||dz|| <= tol and parallel/sharded.py.
     poll = 4
+    stat_pin = [torch.empty((), pin_memory=True) for _ in range(2)]
+    stat_ev = [torch.cuda.Event() for _ in range(2)]
+    pending: list = []  # (z_after_block, slot)
     it = 0
+    blk = 0
+    result_z = None
     while it < max_iter:
         steps = min(poll, max_iter - it)
         for _ in range(steps):
             z = ext.weiszfeld_iter(Xc, z, float(eps), shift)
         it += steps
-        # `shift` holds the accumulated SQUARED step ||dz||^2 (rowops.hip
-        # weiszfeld_iter atomicAdd), so compare against tol^2 to match the
-        # CPU oracle's ||dz|| <= tol and parallel/sharded.py
-        if float(shift) <= tol * tol:  # one sync per `poll` iterations
-            break
-    return z.to(X.dtype)
+        slot = blk & 1
+        blk += 1
+        stat_pin[slot].copy_(shift, non_blocking=True)
+        stat_ev[slot].record()
+        pending.append((z, slot))
+        if len(pending) == 2:
+            z_b, sb = pending.pop(0)
+            stat_ev[sb].synchronize()
+            if float(stat_pin[sb]) <= tol * tol:
+                result_z = z_b
+                break
+    if result_z is None:
+        while pending:
+            z_b, sb = pending.pop(0)
+            stat_ev[sb].synchronize()
+            if float(stat_pin[sb]) <= tol * tol:
+                result_z = z_b
+                break
+        if result_z is None:
+            result_z = z
+    return result_z.to(X.dtype)
 
 
 def geometric_median_grouped(
