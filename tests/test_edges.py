@@ -249,3 +249,21 @@ class TestDispatchFallbacks:
         from byzpy_amd import hip as _h
 
         assert _h.available() in (True, False)  # probe never raises
+
+
+class TestSharedStoreChurn:
+    def test_register_cleanup_leaves_no_shm_segments(self):
+        """POSIX shm churn: 50 register/open/cleanup cycles must leave no
+        /dev/shm segments behind (the leak-hunt family — shm leaks
+        exhaust the host, not the GPU)."""
+        import glob
+
+        before = set(glob.glob("/dev/shm/*"))
+        for i in range(50):
+            h = register_tensor(torch.randn(16, 16))
+            with open_tensor(h) as view:
+                assert view.shape == (16, 16)
+            cleanup_tensor(h)
+        after = set(glob.glob("/dev/shm/*"))
+        leaked = after - before
+        assert not leaked, sorted(leaked)[:5]
