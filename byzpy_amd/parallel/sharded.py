@@ -43,10 +43,12 @@ def _global_gram(X_local: torch.Tensor) -> torch.Tensor:
 
 
 def median_and_multi_krum(X_local: torch.Tensor, f: int, q: int):
-    """Both flagship aggregates from ONE pass over the shard: the fused
-    gram+median kernel computes the local median shard and the partial
-    Gram together; one (n, n) all-reduce then yields global Krum
-    selection identical to multi_krum()."""
+    """Both flagship aggregates with ONE collective: the local median
+    shard and the partial Gram (two kernels — the single-pass fusion was
+    measured slower and removed, see profiles/r02_fusion_negative.md),
+    then one (n, n) all-reduce yields a Krum selection identical on
+    every rank (RCCL all-reduce returns bitwise-identical sums on all
+    ranks, so the winner set never diverges)."""
     from byzpy_amd.hip import dispatch as _D
 
     n = X_local.shape[0]
